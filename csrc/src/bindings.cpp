@@ -1,0 +1,231 @@
+// pybind11 bindings for the stencil_amd native core.
+//
+// The native library is self-contained C++/HIP (no libtorch dependency);
+// staging buffers are exported to Python through DLPack capsules so
+// torch.from_dlpack can alias them for RCCL (torch.distributed) transport.
+#include <pybind11/pybind11.h>
+#include <pybind11/stl.h>
+
+#include <cstring>
+
+#include "stencil_amd/core.hpp"
+#include "stencil_amd/domain.hpp"
+#include "stencil_amd/engine.hpp"
+#include "stencil_amd/hip_check.hpp"
+#include "stencil_amd/ops.hpp"
+#include "stencil_amd/partition.hpp"
+#include "stencil_amd/qap.hpp"
+
+namespace py = pybind11;
+using namespace stencil_amd;
+
+namespace {
+
+//// minimal DLPack (v0.8 ABI) ////
+typedef struct {
+  int32_t device_type;
+  int32_t device_id;
+} DLDevice_;
+typedef struct {
+  uint8_t code;
+  uint8_t bits;
+  uint16_t lanes;
+} DLDataType_;
+typedef struct {
+  void *data;
+  DLDevice_ device;
+  int32_t ndim;
+  DLDataType_ dtype;
+  int64_t *shape;
+  int64_t *strides;
+  uint64_t byte_offset;
+} DLTensor_;
+struct DLManagedTensor_ {
+  DLTensor_ dl_tensor;
+  void *manager_ctx;
+  void (*deleter)(DLManagedTensor_ *);
+};
+constexpr int32_t kDLROCM = 10;
+
+struct DLHolder {
+  DLManagedTensor_ mt;
+  int64_t shape[1];
+};
+
+// 1-D uint8 view of `bytes` device bytes at `ptr` on HIP device `dev`.
+// The engine owns the memory; the capsule only owns the descriptor.
+py::capsule make_dlpack_u8(uintptr_t ptr, int64_t bytes, int dev) {
+  auto *h = new DLHolder();
+  h->shape[0] = bytes;
+  h->mt.dl_tensor.data = (void *)ptr;
+  h->mt.dl_tensor.device = {kDLROCM, dev};
+  h->mt.dl_tensor.ndim = 1;
+  h->mt.dl_tensor.dtype = {1 /*kDLUInt*/, 8, 1};
+  h->mt.dl_tensor.shape = h->shape;
+  h->mt.dl_tensor.strides = nullptr;
+  h->mt.dl_tensor.byte_offset = 0;
+  h->mt.manager_ctx = h;
+  h->mt.deleter = [](DLManagedTensor_ *mt) { delete (DLHolder *)mt->manager_ctx; };
+  return py::capsule(&h->mt, "dltensor", [](PyObject *cap) {
+    if (PyCapsule_IsValid(cap, "dltensor")) {
+      auto *mt = (DLManagedTensor_ *)PyCapsule_GetPointer(cap, "dltensor");
+      if (mt && mt->deleter) mt->deleter(mt);
+    }
+  });
+}
+
+} // namespace
+
+PYBIND11_MODULE(_C, m) {
+  m.doc() = "stencil_amd native core (MI355X / gfx950)";
+
+  py::class_<Vec3>(m, "Vec3")
+      .def(py::init<int64_t, int64_t, int64_t>())
+      .def_readwrite("x", &Vec3::x)
+      .def_readwrite("y", &Vec3::y)
+      .def_readwrite("z", &Vec3::z)
+      .def("__eq__", [](const Vec3 &a, const Vec3 &b) { return a == b; })
+      .def("__lt__", [](const Vec3 &a, const Vec3 &b) { return a < b; })
+      .def("__add__", [](const Vec3 &a, const Vec3 &b) { return a + b; })
+      .def("__sub__", [](const Vec3 &a, const Vec3 &b) { return a - b; })
+      .def("__mul__", [](const Vec3 &a, const Vec3 &b) { return a * b; })
+      .def("__mod__", [](const Vec3 &a, const Vec3 &b) { return a % b; })
+      .def("__neg__", [](const Vec3 &a) { return -a; })
+      .def("__hash__", [](const Vec3 &a) { return py::hash(py::make_tuple(a.x, a.y, a.z)); })
+      .def("wrap", &Vec3::wrap)
+      .def("flatten", &Vec3::flatten)
+      .def("tuple", [](const Vec3 &a) { return py::make_tuple(a.x, a.y, a.z); })
+      .def("__repr__", &Vec3::str);
+
+  py::class_<Rect3>(m, "Rect3")
+      .def(py::init<const Vec3 &, const Vec3 &>())
+      .def_readwrite("lo", &Rect3::lo)
+      .def_readwrite("hi", &Rect3::hi)
+      .def("extent", &Rect3::extent)
+      .def("volume", &Rect3::volume)
+      .def("contains", &Rect3::contains)
+      .def("__eq__", [](const Rect3 &a, const Rect3 &b) { return a == b; })
+      .def("__repr__", &Rect3::str);
+
+  py::class_<Radius>(m, "Radius")
+      .def(py::init<>())
+      .def_static("constant", &Radius::constant)
+      .def_static("face_edge_corner", &Radius::face_edge_corner)
+      .def("dir", [](const Radius &r, int x, int y, int z) { return r.dir(x, y, z); })
+      .def("set_dir", [](Radius &r, int x, int y, int z, int64_t v) { r.dir(x, y, z) = v; })
+      .def("x", &Radius::x)
+      .def("y", &Radius::y)
+      .def("z", &Radius::z)
+      .def("__eq__", [](const Radius &a, const Radius &b) { return a == b; });
+
+  m.def("prime_factors", &prime_factors);
+  m.def("div_ceil", &div_ceil);
+
+  py::class_<RankPartition>(m, "RankPartition")
+      .def(py::init<const Vec3 &, int64_t>())
+      .def("dim", &RankPartition::dim)
+      .def("subdomain_size", &RankPartition::subdomain_size)
+      .def("subdomain_origin", &RankPartition::subdomain_origin)
+      .def("linearize", &RankPartition::linearize)
+      .def("dimensionize", &RankPartition::dimensionize);
+
+  py::class_<NodePartition>(m, "NodePartition")
+      .def(py::init<const Vec3 &, const Radius &, int64_t, int64_t>())
+      .def("sys_dim", &NodePartition::sys_dim)
+      .def("node_dim", &NodePartition::node_dim)
+      .def("dim", &NodePartition::dim)
+      .def("subdomain_size", &NodePartition::subdomain_size)
+      .def("subdomain_origin", &NodePartition::subdomain_origin);
+
+  py::class_<SqMat>(m, "SqMat")
+      .def(py::init<int64_t, double>(), py::arg("n"), py::arg("fill") = 0.0)
+      .def("set", [](SqMat &mm, int64_t i, int64_t j, double v) { mm.at(i, j) = v; })
+      .def("get", [](const SqMat &mm, int64_t i, int64_t j) { return mm.at(i, j); });
+  m.def("qap_solve", &qap::solve, py::arg("w"), py::arg("d"), py::arg("timeout_sec") = 10.0);
+  m.def("qap_cost", &qap::cost);
+
+  py::class_<LocalDomain, std::shared_ptr<LocalDomain>>(m, "LocalDomain")
+      .def(py::init<const Vec3 &, const Vec3 &, int>())
+      .def("add_data", &LocalDomain::add_data, py::arg("elem_size"), py::arg("name") = "")
+      .def("set_radius", &LocalDomain::set_radius)
+      .def("realize", &LocalDomain::realize)
+      .def("swap", &LocalDomain::swap)
+      .def("halo_pos", (Vec3(LocalDomain::*)(const Vec3 &, bool) const) & LocalDomain::halo_pos)
+      .def("halo_extent", (Vec3(LocalDomain::*)(const Vec3 &) const) & LocalDomain::halo_extent)
+      .def("halo_coords", &LocalDomain::halo_coords)
+      .def("halo_bytes", &LocalDomain::halo_bytes)
+      .def("compute_region", &LocalDomain::compute_region)
+      .def("full_region", &LocalDomain::full_region)
+      .def("raw_size", &LocalDomain::raw_size)
+      .def("size", &LocalDomain::size)
+      .def("origin", &LocalDomain::origin)
+      .def("gpu", &LocalDomain::gpu)
+      .def("num_data", &LocalDomain::num_data)
+      .def("elem_size", &LocalDomain::elem_size)
+      .def("region_to_host",
+           [](const LocalDomain &d, const Vec3 &pos, const Vec3 &ext, int64_t qi, bool fromNext) {
+             std::string out(ext.flatten() * d.elem_size(qi), '\0');
+             d.region_to_host(out.data(), pos, ext, qi, fromNext);
+             return py::bytes(out);
+           },
+           py::arg("pos"), py::arg("ext"), py::arg("qi"), py::arg("from_next") = false)
+      .def("region_from_host",
+           [](const LocalDomain &d, py::bytes data, const Vec3 &pos, const Vec3 &ext, int64_t qi,
+              bool toNext) {
+             std::string s = data;
+             if ((int64_t)s.size() != ext.flatten() * d.elem_size(qi))
+               throw std::runtime_error("region_from_host: size mismatch");
+             d.region_from_host(s.data(), pos, ext, qi, toNext);
+           },
+           py::arg("data"), py::arg("pos"), py::arg("ext"), py::arg("qi"),
+           py::arg("to_next") = false);
+
+  // module-level geometry statics (CPU-testable without a GPU)
+  m.def("halo_pos", [](const Vec3 &dir, const Vec3 &sz, const Radius &r, bool halo) {
+    return LocalDomain::halo_pos(dir, sz, r, halo);
+  });
+  m.def("halo_extent", [](const Vec3 &dir, const Vec3 &sz, const Radius &r) {
+    return LocalDomain::halo_extent(dir, sz, r);
+  });
+
+  py::class_<ExchangeEngine>(m, "ExchangeEngine")
+      .def(py::init<std::vector<std::shared_ptr<LocalDomain>>>())
+      .def("enable_peer_all", &ExchangeEngine::enable_peer_all)
+      .def_static("can_access_peer", &ExchangeEngine::can_access_peer)
+      .def("add_translate", &ExchangeEngine::add_translate)
+      .def("create_buffer", &ExchangeEngine::create_buffer)
+      .def("add_pack", &ExchangeEngine::add_pack)
+      .def("add_unpack", &ExchangeEngine::add_unpack)
+      .def("finalize", &ExchangeEngine::finalize)
+      .def("launch_translates", &ExchangeEngine::launch_translates)
+      .def("launch_packs", &ExchangeEngine::launch_packs)
+      .def("launch_unpacks", &ExchangeEngine::launch_unpacks)
+      .def("sync_translates", &ExchangeEngine::sync_translates)
+      .def("sync_packs", &ExchangeEngine::sync_packs)
+      .def("sync_all", &ExchangeEngine::sync_all)
+      .def("sync_compute", &ExchangeEngine::sync_compute)
+      .def("compute_stream_handle", &ExchangeEngine::compute_stream_handle)
+      .def("buffer_bytes", &ExchangeEngine::buffer_bytes)
+      .def("buffer_device", &ExchangeEngine::buffer_device)
+      .def("num_domains", &ExchangeEngine::num_domains)
+      .def("buffer_dlpack", [](ExchangeEngine &e, int64_t buf) {
+        return make_dlpack_u8(e.buffer_ptr(buf), e.buffer_bytes(buf), e.buffer_device(buf));
+      });
+
+  m.def("jacobi_step", &jacobi_step);
+  m.def("fill_f32", &fill_f32);
+
+  m.def("device_count", []() {
+    int n = 0;
+    if (hipGetDeviceCount(&n) != hipSuccess) return 0;
+    return n;
+  });
+  m.def("device_synchronize_all", []() {
+    int n = 0;
+    STENCIL_HIP(hipGetDeviceCount(&n));
+    for (int i = 0; i < n; ++i) {
+      STENCIL_HIP(hipSetDevice(i));
+      STENCIL_HIP(hipDeviceSynchronize());
+    }
+  });
+}

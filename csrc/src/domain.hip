@@ -1,0 +1,143 @@
+// LocalDomain implementation (hipMalloc double buffers + device pointer
+// tables). Reference behavior: src/local_domain.cu:159-220 (realize),
+// :67-84 (swap), :86-129 (halo_pos), :131-157 (region_to_host).
+#include "stencil_amd/domain.hpp"
+#include "stencil_amd/hip_check.hpp"
+
+#include <cstring>
+#include <stdexcept>
+
+namespace stencil_amd {
+
+LocalDomain::LocalDomain(const Vec3 &sz, const Vec3 &origin, int dev)
+    : sz_(sz), origin_(origin), dev_(dev) {}
+
+LocalDomain::~LocalDomain() {
+  if (!realized_) return;
+  (void)hipSetDevice(dev_);
+  for (auto &p : curr_)
+    if (p.ptr) (void)hipFree(p.ptr);
+  for (auto &p : next_)
+    if (p.ptr) (void)hipFree(p.ptr);
+  if (devCurrRaw_) (void)hipFree(devCurrRaw_);
+  if (devNextRaw_) (void)hipFree(devNextRaw_);
+}
+
+int64_t LocalDomain::add_data(int64_t elemSize, const std::string &name) {
+  elemSize_.push_back(elemSize);
+  name_.push_back(name);
+  return (int64_t)elemSize_.size() - 1;
+}
+
+void LocalDomain::realize() {
+  if (realized_) throw std::runtime_error("LocalDomain::realize called twice");
+  STENCIL_HIP(hipSetDevice(dev_));
+  const Vec3 raw = raw_size();
+  curr_.resize(elemSize_.size());
+  next_.resize(elemSize_.size());
+  for (size_t qi = 0; qi < elemSize_.size(); ++qi) {
+    // pitch rows to 256 B so every y-row starts on an HBM-friendly boundary
+    const int64_t pitch = align_up(raw.x * elemSize_[qi], 256);
+    const int64_t bytes = pitch * raw.y * raw.z;
+    for (Pitched *buf : {&curr_[qi], &next_[qi]}) {
+      buf->pitch = pitch;
+      buf->ysize = raw.y;
+      STENCIL_HIP(hipMalloc((void **)&buf->ptr, bytes));
+      STENCIL_HIP(hipMemset(buf->ptr, 0, bytes));
+    }
+  }
+  // device pointer tables (fixed addresses; contents refreshed on swap)
+  const int64_t n = num_data();
+  STENCIL_HIP(hipMalloc((void **)&devCurrRaw_, n * sizeof(char *)));
+  STENCIL_HIP(hipMalloc((void **)&devNextRaw_, n * sizeof(char *)));
+  realized_ = true;
+  swapUpload_();
+}
+
+void LocalDomain::swap() {
+  std::swap(curr_, next_);
+  swapUpload_();
+}
+
+void LocalDomain::swapUpload_() {
+  STENCIL_HIP(hipSetDevice(dev_));
+  const int64_t n = num_data();
+  std::vector<char *> c(n), x(n);
+  for (int64_t i = 0; i < n; ++i) {
+    c[i] = curr_[i].ptr;
+    x[i] = next_[i].ptr;
+  }
+  STENCIL_HIP(hipMemcpy(devCurrRaw_, c.data(), n * sizeof(char *), hipMemcpyHostToDevice));
+  STENCIL_HIP(hipMemcpy(devNextRaw_, x.data(), n * sizeof(char *), hipMemcpyHostToDevice));
+}
+
+Vec3 LocalDomain::halo_pos(const Vec3 &dir, const Vec3 &sz, const Radius &radius, bool halo) {
+  Vec3 ret;
+  for (int i = 0; i < 3; ++i) {
+    const int d = (int)dir[i];
+    const int64_t rNeg = radius.dir(i == 0 ? -1 : 0, i == 1 ? -1 : 0, i == 2 ? -1 : 0);
+    if (1 == d) {
+      ret[i] = sz[i] + (halo ? rNeg : 0);
+    } else if (-1 == d) {
+      ret[i] = halo ? 0 : rNeg;
+    } else {
+      ret[i] = rNeg;
+    }
+  }
+  return ret;
+}
+
+Rect3 LocalDomain::halo_coords(const Vec3 &dir, bool halo) const {
+  Vec3 pos = halo_pos(dir, halo);
+  const Vec3 ext = halo_extent(dir);
+  // allocation coords -> global coords
+  pos.x -= radius_.x(-1);
+  pos.y -= radius_.y(-1);
+  pos.z -= radius_.z(-1);
+  pos += origin_;
+  return Rect3(pos, pos + ext);
+}
+
+Rect3 LocalDomain::full_region() const {
+  Vec3 lo = origin_;
+  Vec3 hi = origin_ + sz_;
+  lo.x -= radius_.x(-1);
+  lo.y -= radius_.y(-1);
+  lo.z -= radius_.z(-1);
+  hi.x += radius_.x(1);
+  hi.y += radius_.y(1);
+  hi.z += radius_.z(1);
+  return Rect3(lo, hi);
+}
+
+void LocalDomain::region_to_host(void *dst, const Vec3 &pos, const Vec3 &ext, int64_t qi,
+                                 bool fromNext) const {
+  STENCIL_HIP(hipSetDevice(dev_));
+  const Pitched &p = fromNext ? next_[qi] : curr_[qi];
+  const int64_t es = elemSize_[qi];
+  const int64_t rowBytes = ext.x * es;
+  char *d = (char *)dst;
+  for (int64_t z = 0; z < ext.z; ++z)
+    for (int64_t y = 0; y < ext.y; ++y) {
+      const char *s = p.ptr + (pos.z + z) * p.plane() + (pos.y + y) * p.pitch + pos.x * es;
+      STENCIL_HIP(hipMemcpy(d, s, rowBytes, hipMemcpyDeviceToHost));
+      d += rowBytes;
+    }
+}
+
+void LocalDomain::region_from_host(const void *src, const Vec3 &pos, const Vec3 &ext, int64_t qi,
+                                   bool toNext) const {
+  STENCIL_HIP(hipSetDevice(dev_));
+  const Pitched &p = toNext ? next_[qi] : curr_[qi];
+  const int64_t es = elemSize_[qi];
+  const int64_t rowBytes = ext.x * es;
+  const char *s = (const char *)src;
+  for (int64_t z = 0; z < ext.z; ++z)
+    for (int64_t y = 0; y < ext.y; ++y) {
+      char *d = p.ptr + (pos.z + z) * p.plane() + (pos.y + y) * p.pitch + pos.x * es;
+      STENCIL_HIP(hipMemcpy(d, s, rowBytes, hipMemcpyHostToDevice));
+      s += rowBytes;
+    }
+}
+
+} // namespace stencil_amd

@@ -1,0 +1,328 @@
+// ExchangeEngine implementation: job-table construction and the batched
+// copy kernel. See engine.hpp for the design rationale.
+#include "stencil_amd/engine.hpp"
+#include "stencil_amd/hip_check.hpp"
+
+#include <algorithm>
+#include <initializer_list>
+#include <numeric>
+#include <stdexcept>
+
+namespace stencil_amd {
+
+namespace {
+
+constexpr int kBlock = 256;
+
+struct alignas(16) W16 {
+  uint64_t a, b;
+};
+
+__global__ void copy_batch_kernel(const CopyJob *__restrict__ jobs, const int64_t *__restrict__ prefix,
+                                  int nJobs) {
+  // binary-search the job that owns this block
+  const int64_t b = blockIdx.x;
+  int lo = 0, hi = nJobs - 1;
+  while (lo < hi) {
+    const int mid = (lo + hi) / 2;
+    if (prefix[mid + 1] <= b) {
+      lo = mid + 1;
+    } else {
+      hi = mid;
+    }
+  }
+  const CopyJob j = jobs[lo];
+  const char *src = (j.srcSlot ? *j.srcSlot : j.srcDirect) + j.srcOff;
+  char *dst = (j.dstSlot ? *j.dstSlot : j.dstDirect) + j.dstOff;
+
+  const int64_t jobBlocks = prefix[lo + 1] - prefix[lo];
+  const int64_t stride = jobBlocks * blockDim.x;
+  for (int64_t w = (b - prefix[lo]) * blockDim.x + threadIdx.x; w < j.nWords; w += stride) {
+    const int32_t wx = (int32_t)(w % j.extXw);
+    const int64_t t = w / j.extXw;
+    const int32_t wy = (int32_t)(t % j.extY);
+    const int64_t wz = t / j.extY;
+    const char *s = src + wz * j.srcPlane + (int64_t)wy * j.srcPitch + (int64_t)wx * j.wordBytes;
+    char *d = dst + wz * j.dstPlane + (int64_t)wy * j.dstPitch + (int64_t)wx * j.wordBytes;
+    switch (j.wordBytes) {
+    case 16:
+      *reinterpret_cast<W16 *>(d) = *reinterpret_cast<const W16 *>(s);
+      break;
+    case 8:
+      *reinterpret_cast<uint64_t *>(d) = *reinterpret_cast<const uint64_t *>(s);
+      break;
+    case 4:
+      *reinterpret_cast<uint32_t *>(d) = *reinterpret_cast<const uint32_t *>(s);
+      break;
+    case 2:
+      *reinterpret_cast<uint16_t *>(d) = *reinterpret_cast<const uint16_t *>(s);
+      break;
+    default:
+      *d = *s;
+      break;
+    }
+  }
+}
+
+// largest word size (<=16) honoring every alignment constraint of the copy
+int pick_word(int64_t rowBytes, std::initializer_list<int64_t> alignedQuantities) {
+  for (int w : {16, 8, 4, 2}) {
+    if (rowBytes % w) continue;
+    bool ok = true;
+    for (int64_t q : alignedQuantities)
+      if (q % w) {
+        ok = false;
+        break;
+      }
+    if (ok) return w;
+  }
+  return 1;
+}
+
+} // namespace
+
+void CopyBatch::finalize_upload() {
+  prefix.assign(jobs.size() + 1, 0);
+  for (size_t i = 0; i < jobs.size(); ++i) {
+    const int64_t blocks = std::max<int64_t>(1, (jobs[i].nWords + kBlock - 1) / kBlock);
+    prefix[i + 1] = prefix[i] + blocks;
+  }
+  nBlocks = prefix.back();
+  STENCIL_HIP(hipSetDevice(dev));
+  STENCIL_HIP(hipMalloc((void **)&dJobs, jobs.size() * sizeof(CopyJob)));
+  STENCIL_HIP(hipMalloc((void **)&dPrefix, prefix.size() * sizeof(int64_t)));
+  STENCIL_HIP(hipMemcpy(dJobs, jobs.data(), jobs.size() * sizeof(CopyJob), hipMemcpyHostToDevice));
+  STENCIL_HIP(
+      hipMemcpy(dPrefix, prefix.data(), prefix.size() * sizeof(int64_t), hipMemcpyHostToDevice));
+}
+
+void CopyBatch::launch(hipStream_t stream) {
+  if (jobs.empty()) return;
+  STENCIL_HIP(hipSetDevice(dev));
+  hipLaunchKernelGGL(copy_batch_kernel, dim3((uint32_t)nBlocks), dim3(kBlock), 0, stream, dJobs,
+                     dPrefix, (int)jobs.size());
+  STENCIL_HIP(hipGetLastError());
+}
+
+void CopyBatch::destroy() {
+  if (dJobs) (void)hipFree(dJobs);
+  if (dPrefix) (void)hipFree(dPrefix);
+  dJobs = nullptr;
+  dPrefix = nullptr;
+}
+
+ExchangeEngine::ExchangeEngine(std::vector<std::shared_ptr<LocalDomain>> domains)
+    : domains_(std::move(domains)) {
+  computeStreams_.resize(domains_.size(), nullptr);
+}
+
+ExchangeEngine::~ExchangeEngine() {
+  for (auto &b : translateBatches_) b.destroy();
+  for (auto &b : packBatches_) b.destroy();
+  for (auto &b : unpackBatches_) b.destroy();
+  for (auto &kv : commStreams_) (void)hipStreamDestroy(kv.second);
+  for (auto &kv : packStreams_) (void)hipStreamDestroy(kv.second);
+  for (auto s : computeStreams_)
+    if (s) (void)hipStreamDestroy(s);
+  for (auto &b : buffers_)
+    if (b.ptr) (void)hipFree(b.ptr);
+}
+
+bool ExchangeEngine::can_access_peer(int src, int dst) {
+  if (src == dst) return true;
+  int ok = 0;
+  if (hipDeviceCanAccessPeer(&ok, src, dst) != hipSuccess) return false;
+  return ok != 0;
+}
+
+void ExchangeEngine::enable_peer_all() {
+  std::vector<int> devs;
+  for (auto &d : domains_) devs.push_back(d->gpu());
+  std::sort(devs.begin(), devs.end());
+  devs.erase(std::unique(devs.begin(), devs.end()), devs.end());
+  for (int a : devs)
+    for (int b : devs) {
+      if (a == b) continue;
+      STENCIL_HIP(hipSetDevice(a));
+      hipError_t err = hipDeviceEnablePeerAccess(b, 0);
+      if (err != hipSuccess && err != hipErrorPeerAccessAlreadyEnabled) {
+        STENCIL_HIP(err);
+      }
+      (void)hipGetLastError(); // clear sticky already-enabled
+    }
+}
+
+void ExchangeEngine::add_translate(int srcDom, int dstDom, const Vec3 &srcPos, const Vec3 &dstPos,
+                                   const Vec3 &ext) {
+  translateSpecs_.push_back({srcDom, dstDom, srcPos, dstPos, ext});
+}
+
+int64_t ExchangeEngine::create_buffer(int dom, int64_t bytes) {
+  Buffer b;
+  b.dev = domains_[dom]->gpu();
+  b.bytes = bytes;
+  STENCIL_HIP(hipSetDevice(b.dev));
+  STENCIL_HIP(hipMalloc((void **)&b.ptr, std::max<int64_t>(bytes, 16)));
+  buffers_.push_back(b);
+  return (int64_t)buffers_.size() - 1;
+}
+
+void ExchangeEngine::add_pack(int dom, int64_t buf, int64_t offset, const Vec3 &pos, const Vec3 &ext,
+                              int64_t qi) {
+  packSpecs_.push_back({dom, buf, offset, pos, ext, qi, false});
+}
+
+void ExchangeEngine::add_unpack(int dom, int64_t buf, int64_t offset, const Vec3 &pos,
+                                const Vec3 &ext, int64_t qi) {
+  packSpecs_.push_back({dom, buf, offset, pos, ext, qi, true});
+}
+
+void ExchangeEngine::finalize() {
+  if (finalized_) throw std::runtime_error("ExchangeEngine::finalize called twice");
+  build_batches_(translateSpecs_, packSpecs_);
+  for (auto &b : translateBatches_) b.finalize_upload();
+  for (auto &b : packBatches_) b.finalize_upload();
+  for (auto &b : unpackBatches_) b.finalize_upload();
+  finalized_ = true;
+}
+
+void ExchangeEngine::build_batches_(const std::vector<TranslateSpec> &ts,
+                                    const std::vector<PackSpec> &ps) {
+  std::map<int, CopyBatch> tb, pb, ub;
+
+  for (const auto &t : ts) {
+    LocalDomain &s = *domains_[t.srcDom];
+    LocalDomain &d = *domains_[t.dstDom];
+    for (int64_t qi = 0; qi < s.num_data(); ++qi) {
+      const int64_t es = s.elem_size(qi);
+      const Pitched &sp = s.curr(qi); // pitch/ysize only; base via slot
+      const Pitched &dp = d.curr(qi);
+      CopyJob j{};
+      j.srcSlot = (const char *const *)(s.dev_curr_slots() + qi);
+      j.dstSlot = (char *const *)(d.dev_curr_slots() + qi);
+      j.srcOff = t.srcPos.z * sp.plane() + t.srcPos.y * sp.pitch + t.srcPos.x * es;
+      j.dstOff = t.dstPos.z * dp.plane() + t.dstPos.y * dp.pitch + t.dstPos.x * es;
+      j.srcPitch = sp.pitch;
+      j.srcPlane = sp.plane();
+      j.dstPitch = dp.pitch;
+      j.dstPlane = dp.plane();
+      const int64_t rowBytes = t.ext.x * es;
+      const int w = pick_word(rowBytes, {j.srcOff, j.dstOff, j.srcPitch, j.dstPitch});
+      j.wordBytes = w;
+      j.extXw = (int32_t)(rowBytes / w);
+      j.extY = (int32_t)t.ext.y;
+      j.nWords = (int64_t)j.extXw * t.ext.y * t.ext.z;
+      auto &batch = tb[s.gpu()];
+      batch.dev = s.gpu();
+      batch.jobs.push_back(j);
+    }
+  }
+
+  for (const auto &p : ps) {
+    LocalDomain &dom = *domains_[p.dom];
+    const int64_t es = dom.elem_size(p.qi);
+    const Pitched &dp = dom.curr(p.qi);
+    const Buffer &buf = buffers_[p.buf];
+    const int64_t rowBytes = p.ext.x * es;
+    CopyJob j{};
+    const int64_t domOff = p.pos.z * dp.plane() + p.pos.y * dp.pitch + p.pos.x * es;
+    if (!p.unpack) { // domain -> buffer
+      j.srcSlot = (const char *const *)(dom.dev_curr_slots() + p.qi);
+      j.srcOff = domOff;
+      j.srcPitch = dp.pitch;
+      j.srcPlane = dp.plane();
+      j.dstDirect = buf.ptr;
+      j.dstOff = p.offset;
+      j.dstPitch = rowBytes;
+      j.dstPlane = rowBytes * p.ext.y;
+    } else { // buffer -> domain
+      j.srcDirect = buf.ptr;
+      j.srcOff = p.offset;
+      j.srcPitch = rowBytes;
+      j.srcPlane = rowBytes * p.ext.y;
+      j.dstSlot = (char *const *)(dom.dev_curr_slots() + p.qi);
+      j.dstOff = domOff;
+      j.dstPitch = dp.pitch;
+      j.dstPlane = dp.plane();
+    }
+    const int w = pick_word(rowBytes, {j.srcOff, j.dstOff, j.srcPitch, j.dstPitch});
+    j.wordBytes = w;
+    j.extXw = (int32_t)(rowBytes / w);
+    j.extY = (int32_t)p.ext.y;
+    j.nWords = (int64_t)j.extXw * p.ext.y * p.ext.z;
+    auto &batch = (p.unpack ? ub : pb)[dom.gpu()];
+    batch.dev = dom.gpu();
+    batch.jobs.push_back(j);
+  }
+
+  for (auto &kv : tb) translateBatches_.push_back(std::move(kv.second));
+  for (auto &kv : pb) packBatches_.push_back(std::move(kv.second));
+  for (auto &kv : ub) unpackBatches_.push_back(std::move(kv.second));
+}
+
+hipStream_t ExchangeEngine::comm_stream_(int dev) {
+  auto it = commStreams_.find(dev);
+  if (it != commStreams_.end()) return it->second;
+  STENCIL_HIP(hipSetDevice(dev));
+  hipStream_t s;
+  STENCIL_HIP(hipStreamCreateWithFlags(&s, hipStreamNonBlocking));
+  commStreams_[dev] = s;
+  return s;
+}
+
+hipStream_t ExchangeEngine::pack_stream_(int dev) {
+  auto it = packStreams_.find(dev);
+  if (it != packStreams_.end()) return it->second;
+  STENCIL_HIP(hipSetDevice(dev));
+  hipStream_t s;
+  STENCIL_HIP(hipStreamCreateWithFlags(&s, hipStreamNonBlocking));
+  packStreams_[dev] = s;
+  return s;
+}
+
+hipStream_t ExchangeEngine::compute_stream(int dom) {
+  if (!computeStreams_[dom]) {
+    STENCIL_HIP(hipSetDevice(domains_[dom]->gpu()));
+    hipStream_t s;
+    STENCIL_HIP(hipStreamCreateWithFlags(&s, hipStreamNonBlocking));
+    computeStreams_[dom] = s;
+  }
+  return computeStreams_[dom];
+}
+
+void ExchangeEngine::launch_translates() {
+  for (auto &b : translateBatches_) b.launch(comm_stream_(b.dev));
+}
+void ExchangeEngine::launch_packs() {
+  for (auto &b : packBatches_) b.launch(pack_stream_(b.dev));
+}
+void ExchangeEngine::launch_unpacks() {
+  for (auto &b : unpackBatches_) b.launch(pack_stream_(b.dev));
+}
+
+void ExchangeEngine::sync_translates() {
+  for (auto &b : translateBatches_) {
+    STENCIL_HIP(hipSetDevice(b.dev));
+    STENCIL_HIP(hipStreamSynchronize(comm_stream_(b.dev)));
+  }
+}
+void ExchangeEngine::sync_packs() {
+  for (auto &kv : packStreams_) {
+    STENCIL_HIP(hipSetDevice(kv.first));
+    STENCIL_HIP(hipStreamSynchronize(kv.second));
+  }
+}
+void ExchangeEngine::sync_all() {
+  sync_translates();
+  sync_packs();
+}
+
+void ExchangeEngine::sync_compute() {
+  for (size_t i = 0; i < computeStreams_.size(); ++i)
+    if (computeStreams_[i]) {
+      STENCIL_HIP(hipSetDevice(domains_[i]->gpu()));
+      STENCIL_HIP(hipStreamSynchronize(computeStreams_[i]));
+    }
+}
+
+} // namespace stencil_amd

@@ -1,0 +1,227 @@
+"""Subdomain -> (rank, GPU) placement strategies.
+
+MI355X-native equivalent of the reference's Placement hierarchy
+(reference: include/stencil/partition.hpp:258-831,
+src/placement_intranoderandom.cpp). Three strategies:
+
+- Trivial: round-robin linearized subdomains over the global slot list.
+- NodeAware: two-level NodePartition (nodes x GPUs); within each node a QAP
+  solve assigns node-local subdomains to GPUs against a bandwidth matrix.
+  On one MI355X node all 8 GPUs are one xGMI hop apart so the matrix is
+  uniform and the QAP degenerates to identity, but the machinery is kept
+  for multi-node and heterogeneous-link systems.
+- IntraNodeRandom: NodeAware scaffolding with a seeded random intra-node
+  assignment (the experimental control for placement studies).
+
+All ranks compute the assignment deterministically from all-gathered
+(hostname, gpus) data; no broadcast of the result is needed.
+"""
+from __future__ import annotations
+
+import enum
+import random
+from typing import Dict, List, Tuple
+
+from .. import _C
+from .comm import Comm
+
+Vec = Tuple[int, int, int]
+
+
+class PlacementStrategy(enum.Enum):
+    NodeAware = "node_aware"
+    Trivial = "trivial"
+    IntraNodeRandom = "intra_node_random"
+
+
+class Slot:
+    """one (rank, local domain index, cuda id, node) GPU slot"""
+
+    __slots__ = ("rank", "local_id", "cuda", "node")
+
+    def __init__(self, rank, local_id, cuda, node):
+        self.rank, self.local_id, self.cuda, self.node = rank, local_id, cuda, node
+
+
+def gather_slots(comm: Comm, my_gpus: List[int]) -> List[Slot]:
+    infos = comm.allgather_object({"host": comm.hostname, "gpus": list(my_gpus)})
+    nodes = comm.node_of_rank()
+    slots = []
+    for rank, info in enumerate(infos):
+        for li, cuda in enumerate(info["gpus"]):
+            slots.append(Slot(rank, li, cuda, nodes[rank]))
+    return slots
+
+
+def _vec3(t) -> "_C.Vec3":
+    return _C.Vec3(int(t[0]), int(t[1]), int(t[2]))
+
+
+class Placement:
+    """Maps every subdomain index of the partition grid to a GPU slot."""
+
+    def __init__(self, size: Vec, radius: "_C.Radius", slots: List[Slot]):
+        self.slots = slots
+        n_nodes = len({s.node for s in slots})
+        per_node = [len([s for s in slots if s.node == n]) for n in range(n_nodes)]
+        if len(set(per_node)) != 1:
+            raise ValueError("placement requires the same GPU count on every node")
+        self.part = _C.NodePartition(_vec3(size), radius, n_nodes, per_node[0])
+        self.dim_v = self.part.dim().tuple()
+        self.sys_dim = self.part.sys_dim().tuple()
+        self.node_dim = self.part.node_dim().tuple()
+        # assignment: linear gid -> slot index, filled by subclasses
+        self.assign: Dict[int, int] = {}
+        self._by_rank: Dict[Tuple[int, int], Vec] = {}
+
+    # grid helpers
+    def dim(self) -> Vec:
+        return self.dim_v
+
+    def linearize(self, idx: Vec) -> int:
+        d = self.dim_v
+        return idx[0] + idx[1] * d[0] + idx[2] * d[0] * d[1]
+
+    def dimensionize(self, gid: int) -> Vec:
+        d = self.dim_v
+        return (gid % d[0], (gid // d[0]) % d[1], gid // (d[0] * d[1]))
+
+    def _finish(self):
+        for gid, si in self.assign.items():
+            s = self.slots[si]
+            self._by_rank[(s.rank, s.local_id)] = self.dimensionize(gid)
+
+    # queries (mirror the reference's Placement interface)
+    def get_rank(self, idx: Vec) -> int:
+        return self.slots[self.assign[self.linearize(idx)]].rank
+
+    def get_subdomain_id(self, idx: Vec) -> int:
+        return self.slots[self.assign[self.linearize(idx)]].local_id
+
+    def get_cuda(self, idx: Vec) -> int:
+        return self.slots[self.assign[self.linearize(idx)]].cuda
+
+    def get_idx(self, rank: int, local_id: int) -> Vec:
+        return self._by_rank[(rank, local_id)]
+
+    def num_local(self, rank: int) -> int:
+        return len([1 for s in self.slots if s.rank == rank])
+
+    def subdomain_size(self, idx: Vec) -> Vec:
+        return self.part.subdomain_size(_vec3(idx)).tuple()
+
+    def subdomain_origin(self, idx: Vec) -> Vec:
+        return self.part.subdomain_origin(_vec3(idx)).tuple()
+
+    # per-node gid lists: node n owns the sys-block of subdomains
+    def _node_gids(self, node: int) -> List[int]:
+        sd, nd = self.sys_dim, self.node_dim
+        sx, sy, sz = node % sd[0], (node // sd[0]) % sd[1], node // (sd[0] * sd[1])
+        gids = []
+        for z in range(nd[2]):
+            for y in range(nd[1]):
+                for x in range(nd[0]):
+                    gids.append(self.linearize((sx * nd[0] + x, sy * nd[1] + y, sz * nd[2] + z)))
+        return gids
+
+
+class TrivialPlacement(Placement):
+    def __init__(self, size, radius, slots):
+        super().__init__(size, radius, slots)
+        n = self.dim_v[0] * self.dim_v[1] * self.dim_v[2]
+        if n != len(slots):
+            raise ValueError(f"{n} subdomains but {len(slots)} GPU slots")
+        for gid in range(n):
+            self.assign[gid] = gid
+        self._finish()
+
+
+def _comm_matrix(gids: List[int], placement: Placement, radius: "_C.Radius") -> "_C.SqMat":
+    """bytes-proportional comm volume between node-local subdomains
+    (halo extents with periodic wrap; reference partition.hpp:723-752)"""
+    n = len(gids)
+    w = _C.SqMat(n, 0.0)
+    dim = placement.dim()
+    pos = {gid: placement.dimensionize(gid) for gid in gids}
+    index_of = {gid: i for i, gid in enumerate(gids)}
+    for gid in gids:
+        p = pos[gid]
+        for dz in (-1, 0, 1):
+            for dy in (-1, 0, 1):
+                for dx in (-1, 0, 1):
+                    if (dx, dy, dz) == (0, 0, 0):
+                        continue
+                    if radius.dir(-dx, -dy, -dz) == 0:
+                        continue
+                    nb = tuple((p[i] + (dx, dy, dz)[i]) % dim[i] for i in range(3))
+                    ngid = placement.linearize(nb)
+                    if ngid in index_of and ngid != gid:
+                        sz = placement.subdomain_size(nb)
+                        ext = _C.halo_extent(
+                            _vec3((-dx, -dy, -dz)), _vec3(sz), radius
+                        ).tuple()
+                        vol = ext[0] * ext[1] * ext[2]
+                        i, j = index_of[gid], index_of[ngid]
+                        w.set(i, j, w.get(i, j) + vol)
+    return w
+
+
+def _bandwidth_matrix(cudas: List[int]) -> "_C.SqMat":
+    """distance = 1/bandwidth between GPU slots. Same-GPU is near-zero
+    distance; distinct MI355X GPUs are one xGMI hop (uniform)."""
+    n = len(cudas)
+    d = _C.SqMat(n, 0.0)
+    for i in range(n):
+        for j in range(n):
+            if cudas[i] == cudas[j]:
+                d.set(i, j, 0.1)
+            else:
+                peer = _C.ExchangeEngine.can_access_peer(cudas[i], cudas[j])
+                d.set(i, j, 1.0 if peer else 5.0)
+    return d
+
+
+class NodeAwarePlacement(Placement):
+    def __init__(self, size, radius, slots):
+        super().__init__(size, radius, slots)
+        n = self.dim_v[0] * self.dim_v[1] * self.dim_v[2]
+        if n != len(slots):
+            raise ValueError(f"{n} subdomains but {len(slots)} GPU slots")
+        n_nodes = len({s.node for s in slots})
+        for node in range(n_nodes):
+            gids = self._node_gids(node)
+            slot_ids = [i for i, s in enumerate(self.slots) if s.node == node]
+            w = _comm_matrix(gids, self, radius)
+            d = _bandwidth_matrix([self.slots[i].cuda for i in slot_ids])
+            f = _C.qap_solve(w, d)
+            for a, gid in enumerate(gids):
+                self.assign[gid] = slot_ids[f[a]]
+        self._finish()
+
+
+class IntraNodeRandomPlacement(Placement):
+    def __init__(self, size, radius, slots, seed: int = 0):
+        super().__init__(size, radius, slots)
+        n = self.dim_v[0] * self.dim_v[1] * self.dim_v[2]
+        if n != len(slots):
+            raise ValueError(f"{n} subdomains but {len(slots)} GPU slots")
+        rng = random.Random(seed)
+        n_nodes = len({s.node for s in slots})
+        for node in range(n_nodes):
+            gids = self._node_gids(node)
+            slot_ids = [i for i, s in enumerate(self.slots) if s.node == node]
+            perm = list(range(len(slot_ids)))
+            rng.shuffle(perm)
+            for a, gid in enumerate(gids):
+                self.assign[gid] = slot_ids[perm[a]]
+        self._finish()
+
+
+def make_placement(strategy: PlacementStrategy, size, radius, slots) -> Placement:
+    if strategy == PlacementStrategy.Trivial:
+        return TrivialPlacement(size, radius, slots)
+    if strategy == PlacementStrategy.NodeAware:
+        return NodeAwarePlacement(size, radius, slots)
+    if strategy == PlacementStrategy.IntraNodeRandom:
+        return IntraNodeRandomPlacement(size, radius, slots)
+    raise ValueError(strategy)

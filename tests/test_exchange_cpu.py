@@ -1,0 +1,117 @@
+"""Full-region ripple exchange verification on the torch reference backend
+(CPU, single process). Covers self-wrap, multi-subdomain, symmetric radii
+1 and 2, asymmetric radius, multiple quantities, and swap semantics."""
+import numpy as np
+import pytest
+
+import stencil_amd as sa
+from stencil_amd import _C
+
+from util import check_full_regions, fill_interiors
+
+
+def make_dd(size, radius, n_domains=1):
+    dd = sa.DistributedDomain(*size, backend="torch")
+    dd.set_radius(radius)
+    dd.set_gpus([0] * n_domains)
+    return dd
+
+
+@pytest.mark.parametrize("r", [1, 2])
+@pytest.mark.parametrize("n_domains", [1, 2, 4, 8])
+def test_ripple_exchange(r, n_domains):
+    dd = make_dd((12, 10, 8), r, n_domains)
+    h = dd.add_data(np.float32, "q")
+    dd.realize()
+    fill_interiors(dd, h)
+    dd.exchange()
+    check_full_regions(dd, h)
+
+
+def test_ripple_asymmetric():
+    r = _C.Radius.constant(1)
+    r.set_dir(1, 0, 0, 2)
+    dd = make_dd((12, 10, 8), 0, 2)
+    dd.set_radius(r)
+    h = dd.add_data(np.float32, "q")
+    dd.realize()
+    fill_interiors(dd, h)
+    dd.exchange()
+    check_full_regions(dd, h)
+
+
+def test_multiple_quantities_exchange():
+    dd = make_dd((10, 10, 10), 1, 2)
+    h1 = dd.add_data(np.float32, "a")
+    h2 = dd.add_data(np.float32, "b")
+    dd.realize()
+    fill_interiors(dd, h1, scale=1.0)
+    fill_interiors(dd, h2, scale=2.0)
+    dd.exchange()
+    check_full_regions(dd, h1, scale=1.0)
+    check_full_regions(dd, h2, scale=2.0)
+
+
+def test_exchange_after_swap():
+    """halos must follow the curr buffer across swap()"""
+    dd = make_dd((8, 8, 8), 1, 2)
+    h = dd.add_data(np.float32, "q")
+    dd.realize()
+    fill_interiors(dd, h, scale=3.0)
+    dd.swap()  # ripple data now in next
+    fill_interiors(dd, h, scale=5.0)  # new data in curr
+    dd.exchange()
+    check_full_regions(dd, h, scale=5.0)
+    dd.swap()
+    # old interiors intact in (now) curr
+    for li in range(dd.num_local()):
+        lo, hi = dd.local_rect(li)
+        got = dd.read_global(li, lo, hi, h)
+        from util import ripple_block
+
+        assert np.array_equal(got, ripple_block(lo, hi, dd.size, 3.0))
+
+
+def test_interior_exterior_cover():
+    dd = make_dd((16, 12, 10), 2, 2)
+    h = dd.add_data(np.float32, "q")
+    dd.realize()
+    interiors = dd.get_interior()
+    exteriors = dd.get_exterior()
+    for li in range(dd.num_local()):
+        lo, hi = dd.local_rect(li)
+        vol = (hi[0] - lo[0]) * (hi[1] - lo[1]) * (hi[2] - lo[2])
+        (ilo, ihi) = interiors[li]
+        ivol = max(0, ihi[0] - ilo[0]) * max(0, ihi[1] - ilo[1]) * max(0, ihi[2] - ilo[2])
+        evol = sum(
+            (b[1][0] - b[0][0]) * (b[1][1] - b[0][1]) * (b[1][2] - b[0][2])
+            for b in exteriors[li]
+        )
+        assert ivol + evol == vol
+        # exterior boxes must not overlap the interior or each other
+        cells = np.zeros((hi[2] - lo[2], hi[1] - lo[1], hi[0] - lo[0]), dtype=np.int32)
+        cells[
+            ilo[2] - lo[2] : ihi[2] - lo[2],
+            ilo[1] - lo[1] : ihi[1] - lo[1],
+            ilo[0] - lo[0] : ihi[0] - lo[0],
+        ] += 1
+        for b in exteriors[li]:
+            cells[
+                b[0][2] - lo[2] : b[1][2] - lo[2],
+                b[0][1] - lo[1] : b[1][1] - lo[1],
+                b[0][0] - lo[0] : b[1][0] - lo[0],
+            ] += 1
+        assert (cells == 1).all()
+
+
+def test_paraview_dump(tmp_path):
+    dd = make_dd((6, 5, 4), 1, 1)
+    h = dd.add_data(np.float32, "temp")
+    dd.realize()
+    fill_interiors(dd, h)
+    dd.write_paraview(str(tmp_path / "pv"))
+    files = list(tmp_path.glob("pv*.txt"))
+    assert len(files) == 1
+    lines = files[0].read_text().strip().splitlines()
+    assert lines[0] == "Z,Y,X,temp"
+    assert len(lines) == 1 + 6 * 5 * 4
