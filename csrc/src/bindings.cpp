@@ -210,6 +210,22 @@ PYBIND11_MODULE(_C, m) {
       .def("num_domains", &ExchangeEngine::num_domains)
       .def("buffer_dlpack", [](ExchangeEngine &e, int64_t buf) {
         return make_dlpack_u8(e.buffer_ptr(buf), e.buffer_bytes(buf), e.buffer_device(buf));
+      })
+      .def("buffer_to_host",
+           [](ExchangeEngine &e, int64_t buf) {
+             std::string out(e.buffer_bytes(buf), '\0');
+             STENCIL_HIP(hipSetDevice(e.buffer_device(buf)));
+             STENCIL_HIP(hipMemcpy(out.data(), (void *)e.buffer_ptr(buf), out.size(),
+                                   hipMemcpyDeviceToHost));
+             return py::bytes(out);
+           })
+      .def("buffer_from_host", [](ExchangeEngine &e, int64_t buf, py::bytes data) {
+        std::string s = data;
+        if ((int64_t)s.size() != e.buffer_bytes(buf))
+          throw std::runtime_error("buffer_from_host: size mismatch");
+        STENCIL_HIP(hipSetDevice(e.buffer_device(buf)));
+        STENCIL_HIP(
+            hipMemcpy((void *)e.buffer_ptr(buf), s.data(), s.size(), hipMemcpyHostToDevice));
       });
 
   m.def("jacobi_step", &jacobi_step);

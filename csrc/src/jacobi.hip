@@ -85,7 +85,7 @@ __global__ void jacobi_kernel(JacobiParams p) {
       const float my = *(const float *)(rowC - p.pitch + ax * 4);
       const float pz = *(const float *)(rowC + p.plane + ax * 4);
       const float mz = *(const float *)(rowC - p.plane + ax * 4);
-      out = (px + mx + py + my + pz + mz) * (1.0f / 6.0f);
+      out = (px + mx + py + my + pz + mz) / 6.0f;
     }
     *(float *)(dstBase + az * p.plane + ay * p.pitch + ax * 4) = out;
   }

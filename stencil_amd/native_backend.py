@@ -42,6 +42,8 @@ class NativeBackend:
         self._send_ops: List[Tuple[object, int, int]] = []  # (tensor, peer, tag)
         self._recv_ops: List[Tuple[object, int, int]] = []
         self._has_wire = False
+        self._wire_via_cpu = False
+        self._cpu_mirror = {}
 
     # ---- plan registration ----
     def register_plan(self, plan: ExchangePlan):
@@ -77,7 +79,28 @@ class NativeBackend:
             else:
                 self._recv_ops.append((tensor, item.peer_rank, tag))
         self._has_wire = bool(self._send_ops or self._recv_ops)
+        if self._has_wire:
+            self._wire_via_cpu = self._detect_cpu_wire()
+            if self._wire_via_cpu:
+                self._cpu_mirror = {
+                    id(t): torch.empty(t.shape, dtype=torch.uint8, device="cpu")
+                    for t, _, _ in self._send_ops + self._recv_ops
+                }
         self.engine.finalize()
+
+    @staticmethod
+    def _detect_cpu_wire() -> bool:
+        """True when the process group cannot carry CUDA tensors (plain gloo):
+        pack buffers are then staged through pinned host mirrors. RCCL (nccl)
+        carries the device buffers directly over xGMI."""
+        import os
+
+        if os.environ.get("STENCIL_AMD_WIRE", "") == "cpu":
+            return True
+        import torch.distributed as dist
+
+        backend = str(dist.get_backend())
+        return "nccl" not in backend
 
     # ---- per-iteration ----
     def exchange(self):
@@ -87,10 +110,25 @@ class NativeBackend:
 
             self.engine.launch_packs()
             self.engine.sync_packs()
-            ops = [dist.P2POp(dist.isend, t, peer, tag=tag) for t, peer, tag in self._send_ops]
-            ops += [dist.P2POp(dist.irecv, t, peer, tag=tag) for t, peer, tag in self._recv_ops]
-            for w in dist.batch_isend_irecv(ops):
-                w.wait()
+            if self._wire_via_cpu:
+                ops = []
+                for t, peer, tag in self._send_ops:
+                    m = self._cpu_mirror[id(t)]
+                    m.copy_(t)
+                    ops.append(dist.P2POp(dist.isend, m, peer, tag=tag))
+                for t, peer, tag in self._recv_ops:
+                    ops.append(dist.P2POp(dist.irecv, self._cpu_mirror[id(t)], peer, tag=tag))
+                for w in dist.batch_isend_irecv(ops):
+                    w.wait()
+                for t, _, _ in self._recv_ops:
+                    t.copy_(self._cpu_mirror[id(t)])
+            else:
+                ops = [dist.P2POp(dist.isend, t, peer, tag=tag) for t, peer, tag in self._send_ops]
+                ops += [
+                    dist.P2POp(dist.irecv, t, peer, tag=tag) for t, peer, tag in self._recv_ops
+                ]
+                for w in dist.batch_isend_irecv(ops):
+                    w.wait()
             self.engine.launch_unpacks()
         self.engine.sync_all()
 
