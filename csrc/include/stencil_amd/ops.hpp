@@ -16,4 +16,23 @@ void jacobi_step(ExchangeEngine &eng, int dom, int64_t qi, const Rect3 &region,
 void fill_f32(ExchangeEngine &eng, int dom, int64_t qi, const Rect3 &region, float value,
               bool nextBuf);
 
+// physical coefficients of the MHD solver (see csrc/src/mhd.hip)
+struct MhdCoeffs {
+  double dsx = 1.0, dsy = 1.0, dsz = 1.0;
+  double cs2 = 1.0;
+  double cp_inv = 1.0;
+  double nu = 5e-3, eta = 5e-3, chi = 5e-4;
+};
+
+// one RK3 substep (step in 0..2) of the 8-field 6th-order MHD system over
+// `region` (global coords); reads curr, updates next in place (Williamson
+// two-buffer form) -- caller swaps after each substep
+void mhd_substep(ExchangeEngine &eng, int dom, const Rect3 &region, int step, double dt,
+                 const MhdCoeffs &cf);
+
+// fill an fp64 region with base + amp*sin(kx*x + ky*y + kz*z + phase)
+// (deterministic smooth initial conditions, reproducible in NumPy)
+void init_harmonic_f64(ExchangeEngine &eng, int dom, int64_t qi, const Rect3 &region, double base,
+                       double amp, double kx, double ky, double kz, double phase, bool nextBuf);
+
 } // namespace stencil_amd

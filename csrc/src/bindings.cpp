@@ -231,6 +231,19 @@ PYBIND11_MODULE(_C, m) {
   m.def("jacobi_step", &jacobi_step);
   m.def("fill_f32", &fill_f32);
 
+  py::class_<MhdCoeffs>(m, "MhdCoeffs")
+      .def(py::init<>())
+      .def_readwrite("dsx", &MhdCoeffs::dsx)
+      .def_readwrite("dsy", &MhdCoeffs::dsy)
+      .def_readwrite("dsz", &MhdCoeffs::dsz)
+      .def_readwrite("cs2", &MhdCoeffs::cs2)
+      .def_readwrite("cp_inv", &MhdCoeffs::cp_inv)
+      .def_readwrite("nu", &MhdCoeffs::nu)
+      .def_readwrite("eta", &MhdCoeffs::eta)
+      .def_readwrite("chi", &MhdCoeffs::chi);
+  m.def("mhd_substep", &mhd_substep);
+  m.def("init_harmonic_f64", &init_harmonic_f64);
+
   m.def("device_count", []() {
     int n = 0;
     if (hipGetDeviceCount(&n) != hipSuccess) return 0;
