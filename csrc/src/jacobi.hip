@@ -62,35 +62,53 @@ __device__ __forceinline__ bool sphere_override(int64_t x, int64_t y, int64_t z,
   return false;
 }
 
-// Vectorized variant for wide regions: each thread owns 4 consecutive
-// x-cells at a 16 B-aligned address. Per 4 cells: 5 aligned float4 loads
-// (center + 4 neighbor rows) + 2 scalar edge loads + 1 float4 store. The
-// x-shifted px/mx values are recomposed from the center vector in
-// registers, so no unaligned vector loads are needed. Thread mapping is
-// (x-unit, y, z) via the 3D grid -- no div/mod per element.
+// Vectorized z-marching kernel for wide regions: each thread owns a
+// 4-wide x strip (16 B aligned) and marches ZCHUNK cells in z, keeping the
+// z-1 / z / z+1 center vectors in registers so the +-z neighbor planes are
+// never re-loaded. Per 4 cells and z step: 3 aligned float4 loads (new
+// center plane + y neighbors) + 2 scalar edge loads + 1 float4 store
+// (~14 B/cell issued vs 28 for the scalar kernel). The hot/cold sphere
+// test short-circuits on a 1D bounding check so the common case costs two
+// int compares.
+#define JAC_ZCHUNK 16
+
 __global__ void __launch_bounds__(256) jacobi_kernel_v4(JacobiParams p) {
   const int32_t u = blockIdx.x * blockDim.x + threadIdx.x; // x unit
   const int32_t ly = blockIdx.y * blockDim.y + threadIdx.y;
-  const int32_t lz = blockIdx.z;
-  if (ly >= p.extY || lz >= p.extZ) return;
+  const int32_t lz0 = blockIdx.z * JAC_ZCHUNK;
+  if (ly >= p.extY) return;
   const char *srcBase = *p.srcSlot;
   char *dstBase = *p.dstSlot;
-  const int64_t gy = p.loY + ly, gz = p.loZ + lz;
-  const int64_t ay = gy - p.allocY, az = gz - p.allocZ;
-  const char *rowC = srcBase + az * p.plane + ay * p.pitch;
-  char *rowD = dstBase + az * p.plane + ay * p.pitch;
+  const int64_t gy = p.loY + ly;
+  const int64_t ay = gy - p.allocY;
+  const int32_t zEnd = min((int32_t)(lz0 + JAC_ZCHUNK), p.extZ);
 
-  // row split: [loX, xa) scalar head, body4 aligned float4 units, tail
-  const int64_t a0 = p.loX - p.allocX; // alloc x of first cell
-  const int32_t head = (int32_t)(((a0 + 3) & ~3LL) - a0) > p.extX
-                           ? p.extX
-                           : (int32_t)(((a0 + 3) & ~3LL) - a0);
-  const int32_t body4 = (p.extX - head) / 4;
-  const int32_t tail = p.extX - head - body4 * 4;
+  // sphere geometry (int32: grids are far below 2^31 per axis)
+  const int32_t cw = (int32_t)(p.cHiX - p.cLoX);
+  const int32_t hotX = (int32_t)p.cLoX + cw / 3, coldX = (int32_t)p.cLoX + cw * 2 / 3;
+  const int32_t cY = (int32_t)(p.cLoY + p.cHiY) / 2, cZ = (int32_t)(p.cLoZ + p.cHiZ) / 2;
+  const int32_t srad = cw / 10;
 
-  auto scalar_cell = [&](int32_t lx) {
+  auto sphere4 = [&](int32_t gx, int32_t gyy, int32_t gzz, float4 &out) {
+    const int32_t dy = gyy - cY, dz = gzz - cZ;
+    const int32_t yz2 = dy * dy + dz * dz;
+    // truncated-int sqrt: only yz2 >= (r+1)^2 guarantees exclusion
+    if (yz2 >= (srad + 1) * (srad + 1)) return;
+#pragma unroll
+    for (int i = 0; i < 4; ++i) {
+      const int32_t dxh = gx + i - hotX, dxc = gx + i - coldX;
+      if ((int32_t)__fsqrt_rn((float)(dxh * dxh + yz2)) <= srad)
+        (&out.x)[i] = 1.0f;
+      else if ((int32_t)__fsqrt_rn((float)(dxc * dxc + yz2)) <= srad)
+        (&out.x)[i] = 0.0f;
+    }
+  };
+
+  auto scalar_cell = [&](int32_t lx, int32_t lz) {
     const int64_t gx = p.loX + lx;
-    const int64_t ax = gx - p.allocX;
+    const int64_t gz = p.loZ + lz;
+    const int64_t ax = gx - p.allocX, az = gz - p.allocZ;
+    const char *rowC = srcBase + az * p.plane + ay * p.pitch;
     float out;
     if (!sphere_override(gx, gy, gz, p, out)) {
       const float px = *(const float *)(rowC + (ax + 1) * 4);
@@ -101,35 +119,49 @@ __global__ void __launch_bounds__(256) jacobi_kernel_v4(JacobiParams p) {
       const float mz = *(const float *)(rowC - p.plane + ax * 4);
       out = (px + mx + py + my + pz + mz) / 6.0f;
     }
-    *(float *)(rowD + ax * 4) = out;
+    *(float *)(dstBase + az * p.plane + ay * p.pitch + ax * 4) = out;
   };
+
+  const int64_t a0 = p.loX - p.allocX;
+  const int32_t head = (int32_t)(((a0 + 3) & ~3LL) - a0) > p.extX
+                           ? p.extX
+                           : (int32_t)(((a0 + 3) & ~3LL) - a0);
+  const int32_t body4 = (p.extX - head) / 4;
+  const int32_t tail = p.extX - head - body4 * 4;
 
   if (u < body4) {
     const int64_t ax = a0 + head + (int64_t)u * 4; // 4-aligned
-    const int64_t gx = ax + p.allocX;
-    const float4 c = *(const float4 *)(rowC + ax * 4);
-    const float left = *(const float *)(rowC + (ax - 1) * 4);
-    const float right = *(const float *)(rowC + (ax + 4) * 4);
-    const float4 py = *(const float4 *)(rowC + p.pitch + ax * 4);
-    const float4 my = *(const float4 *)(rowC - p.pitch + ax * 4);
-    const float4 pz = *(const float4 *)(rowC + p.plane + ax * 4);
-    const float4 mz = *(const float4 *)(rowC - p.plane + ax * 4);
-    float4 out;
-    out.x = (c.y + left + py.x + my.x + pz.x + mz.x) / 6.0f;
-    out.y = (c.z + c.x + py.y + my.y + pz.y + mz.y) / 6.0f;
-    out.z = (c.w + c.y + py.z + my.z + pz.z + mz.z) / 6.0f;
-    out.w = (right + c.z + py.w + my.w + pz.w + mz.w) / 6.0f;
-    // hot/cold sphere cells are rare: recompute those lanes scalar
-    float ov;
-    if (sphere_override(gx + 0, gy, gz, p, ov)) out.x = ov;
-    if (sphere_override(gx + 1, gy, gz, p, ov)) out.y = ov;
-    if (sphere_override(gx + 2, gy, gz, p, ov)) out.z = ov;
-    if (sphere_override(gx + 3, gy, gz, p, ov)) out.w = ov;
-    *(float4 *)(rowD + ax * 4) = out;
+    const int32_t gx = (int32_t)(ax + p.allocX);
+    const int64_t az0 = p.loZ + lz0 - p.allocZ;
+    const char *col = srcBase + az0 * p.plane + ay * p.pitch + ax * 4;
+    char *dcol = dstBase + az0 * p.plane + ay * p.pitch + ax * 4;
+    // rolling center vectors
+    float4 cm = *(const float4 *)(col - p.plane);
+    float4 cc = *(const float4 *)(col);
+    for (int32_t lz = lz0; lz < zEnd; ++lz) {
+      const float4 cp = *(const float4 *)(col + p.plane);
+      const float left = *(const float *)(col - 4);
+      const float right = *(const float *)(col + 16);
+      const float4 py = *(const float4 *)(col + p.pitch);
+      const float4 my = *(const float4 *)(col - p.pitch);
+      float4 out;
+      out.x = (cc.y + left + py.x + my.x + cp.x + cm.x) / 6.0f;
+      out.y = (cc.z + cc.x + py.y + my.y + cp.y + cm.y) / 6.0f;
+      out.z = (cc.w + cc.y + py.z + my.z + cp.z + cm.z) / 6.0f;
+      out.w = (right + cc.z + py.w + my.w + cp.w + cm.w) / 6.0f;
+      sphere4(gx, (int32_t)gy, (int32_t)(p.loZ + lz), out);
+      *(float4 *)dcol = out;
+      cm = cc;
+      cc = cp;
+      col += p.plane;
+      dcol += p.plane;
+    }
   } else if (u == body4) {
-    for (int32_t lx = 0; lx < head; ++lx) scalar_cell(lx);
+    for (int32_t lz = lz0; lz < zEnd; ++lz)
+      for (int32_t lx = 0; lx < head; ++lx) scalar_cell(lx, lz);
   } else if (u == body4 + 1) {
-    for (int32_t lx = p.extX - tail; lx < p.extX; ++lx) scalar_cell(lx);
+    for (int32_t lz = lz0; lz < zEnd; ++lz)
+      for (int32_t lx = p.extX - tail; lx < p.extX; ++lx) scalar_cell(lx, lz);
   }
 }
 
@@ -219,7 +251,8 @@ void jacobi_step(ExchangeEngine &eng, int dom, int64_t qi, const Rect3 &region,
     const int64_t head = std::min<int64_t>(((a0 + 3) & ~3LL) - a0, ext.x);
     const int64_t units = (ext.x - head) / 4 + 2;
     dim3 block(64, 4, 1);
-    dim3 grid((uint32_t)((units + 63) / 64), (uint32_t)((ext.y + 3) / 4), (uint32_t)ext.z);
+    dim3 grid((uint32_t)((units + 63) / 64), (uint32_t)((ext.y + 3) / 4),
+              (uint32_t)((ext.z + 15) / 16)); // 16 == JAC_ZCHUNK
     hipLaunchKernelGGL(jacobi_kernel_v4, grid, block, 0, eng.compute_stream(dom), p);
   } else {
     hipLaunchKernelGGL(jacobi_kernel, dim3(grid_for(ext.flatten(), 256)), dim3(256), 0,
