@@ -94,6 +94,10 @@ public:
   char **dev_curr_slots() const { return devCurrRaw_; }
   char **dev_next_slots() const { return devNextRaw_; }
 
+  // IPC export of a quantity buffer (hipIpcMemHandle_t blob) for the
+  // cross-process direct-write transport
+  std::string ipc_handle(int64_t qi, bool next) const;
+
   // blocking element-region copies (pos in allocation coords)
   void region_to_host(void *dst, const Vec3 &pos, const Vec3 &ext, int64_t qi, bool fromNext = false) const;
   void region_from_host(const void *src, const Vec3 &pos, const Vec3 &ext, int64_t qi, bool toNext = false) const;

@@ -75,6 +75,22 @@ public:
   // direct-write copy of one region for ALL quantities:
   // srcDom's curr -> dstDom's curr (possibly a peer GPU over xGMI)
   void add_translate(int srcDom, int dstDom, const Vec3 &srcPos, const Vec3 &dstPos, const Vec3 &ext);
+
+  //// cross-process direct writes (HIP IPC over xGMI; the MI355X analog of
+  //// the reference's tx_colocated direct-access senders)
+  // open a remote rank's domain buffers from their IPC handles; handles are
+  // hipIpcMemHandle_t blobs, one per (quantity, parity). `openDev` is the
+  // LOCAL device whose kernels will write the remote memory.
+  int64_t create_remote_view(int openDev, const std::vector<std::string> &currHandles,
+                             const std::vector<std::string> &nextHandles,
+                             const std::vector<int64_t> &pitches, const std::vector<int64_t> &ysizes,
+                             const std::vector<int64_t> &elemSizes);
+  // direct-write one region (all quantities) into a remote view's curr
+  void add_translate_view(int srcDom, int64_t view, const Vec3 &srcPos, const Vec3 &dstPos,
+                          const Vec3 &ext);
+  // swap every view's curr/next pointer sets (call in lockstep with
+  // LocalDomain::swap on every rank)
+  void flip_views();
   // staging buffer on `dom`'s GPU; returns buffer id
   int64_t create_buffer(int dom, int64_t bytes);
   // gather region of quantity qi of dom's curr into buffer at byte offset
@@ -115,8 +131,16 @@ private:
     int dev = -1;
   };
   struct TranslateSpec {
-    int srcDom, dstDom;
+    int srcDom, dstDom; // dstDom: local domain index, or remote view id
+    bool dstIsView = false;
     Vec3 srcPos, dstPos, ext;
+  };
+  struct RemoteView {
+    int openDev = -1;
+    int parity = 0;
+    std::vector<char *> base[2]; // [parity][qi] opened pointers
+    std::vector<int64_t> pitch, ysize, elemSize;
+    char **devSlots = nullptr; // local device array, refreshed on flip
   };
   struct PackSpec {
     int dom;
@@ -131,6 +155,7 @@ private:
   hipStream_t pack_stream_(int dev);
 
   std::vector<std::shared_ptr<LocalDomain>> domains_;
+  std::vector<RemoteView> views_;
   std::vector<Buffer> buffers_;
   std::vector<TranslateSpec> translateSpecs_;
   std::vector<PackSpec> packSpecs_;

@@ -169,6 +169,9 @@ PYBIND11_MODULE(_C, m) {
              return py::bytes(out);
            },
            py::arg("pos"), py::arg("ext"), py::arg("qi"), py::arg("from_next") = false)
+      .def("ipc_handle",
+           [](const LocalDomain &d, int64_t qi, bool next) { return py::bytes(d.ipc_handle(qi, next)); },
+           py::arg("qi"), py::arg("next") = false)
       .def("region_from_host",
            [](const LocalDomain &d, py::bytes data, const Vec3 &pos, const Vec3 &ext, int64_t qi,
               bool toNext) {
@@ -193,6 +196,17 @@ PYBIND11_MODULE(_C, m) {
       .def("enable_peer_all", &ExchangeEngine::enable_peer_all)
       .def_static("can_access_peer", &ExchangeEngine::can_access_peer)
       .def("add_translate", &ExchangeEngine::add_translate)
+      .def("create_remote_view",
+           [](ExchangeEngine &e, int openDev, const std::vector<py::bytes> &cur,
+              const std::vector<py::bytes> &nxt, const std::vector<int64_t> &pitches,
+              const std::vector<int64_t> &ysizes, const std::vector<int64_t> &es) {
+             std::vector<std::string> c, n;
+             for (auto &b : cur) c.push_back(b);
+             for (auto &b : nxt) n.push_back(b);
+             return e.create_remote_view(openDev, c, n, pitches, ysizes, es);
+           })
+      .def("add_translate_view", &ExchangeEngine::add_translate_view)
+      .def("flip_views", &ExchangeEngine::flip_views)
       .def("create_buffer", &ExchangeEngine::create_buffer)
       .def("add_pack", &ExchangeEngine::add_pack)
       .def("add_unpack", &ExchangeEngine::add_unpack)

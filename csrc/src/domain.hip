@@ -71,6 +71,14 @@ void LocalDomain::swapUpload_() {
   STENCIL_HIP(hipMemcpy(devNextRaw_, x.data(), n * sizeof(char *), hipMemcpyHostToDevice));
 }
 
+std::string LocalDomain::ipc_handle(int64_t qi, bool next) const {
+  STENCIL_HIP(hipSetDevice(dev_));
+  hipIpcMemHandle_t h;
+  const Pitched &p = next ? next_.at(qi) : curr_.at(qi);
+  STENCIL_HIP(hipIpcGetMemHandle(&h, p.ptr));
+  return std::string((const char *)&h, sizeof(h));
+}
+
 Vec3 LocalDomain::halo_pos(const Vec3 &dir, const Vec3 &sz, const Radius &radius, bool halo) {
   Vec3 ret;
   for (int i = 0; i < 3; ++i) {

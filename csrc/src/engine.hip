@@ -4,6 +4,7 @@
 #include "stencil_amd/hip_check.hpp"
 
 #include <algorithm>
+#include <cstring>
 #include <initializer_list>
 #include <numeric>
 #include <stdexcept>
@@ -126,6 +127,13 @@ ExchangeEngine::~ExchangeEngine() {
     if (s) (void)hipStreamDestroy(s);
   for (auto &b : buffers_)
     if (b.ptr) (void)hipFree(b.ptr);
+  for (auto &v : views_) {
+    (void)hipSetDevice(v.openDev);
+    for (int par = 0; par < 2; ++par)
+      for (char *p : v.base[par])
+        if (p) (void)hipIpcCloseMemHandle(p);
+    if (v.devSlots) (void)hipFree(v.devSlots);
+  }
 }
 
 bool ExchangeEngine::can_access_peer(int src, int dst) {
@@ -154,7 +162,56 @@ void ExchangeEngine::enable_peer_all() {
 
 void ExchangeEngine::add_translate(int srcDom, int dstDom, const Vec3 &srcPos, const Vec3 &dstPos,
                                    const Vec3 &ext) {
-  translateSpecs_.push_back({srcDom, dstDom, srcPos, dstPos, ext});
+  translateSpecs_.push_back({srcDom, dstDom, false, srcPos, dstPos, ext});
+}
+
+int64_t ExchangeEngine::create_remote_view(int openDev, const std::vector<std::string> &currHandles,
+                                           const std::vector<std::string> &nextHandles,
+                                           const std::vector<int64_t> &pitches,
+                                           const std::vector<int64_t> &ysizes,
+                                           const std::vector<int64_t> &elemSizes) {
+  const size_t nq = currHandles.size();
+  if (nextHandles.size() != nq || pitches.size() != nq || ysizes.size() != nq ||
+      elemSizes.size() != nq)
+    throw std::runtime_error("create_remote_view: size mismatch");
+  RemoteView v;
+  v.openDev = openDev;
+  v.pitch = pitches;
+  v.ysize = ysizes;
+  v.elemSize = elemSizes;
+  STENCIL_HIP(hipSetDevice(openDev));
+  auto open_one = [&](const std::string &blob) {
+    if (blob.size() != sizeof(hipIpcMemHandle_t))
+      throw std::runtime_error("create_remote_view: bad handle size");
+    hipIpcMemHandle_t h;
+    std::memcpy(&h, blob.data(), sizeof(h));
+    void *p = nullptr;
+    STENCIL_HIP(hipIpcOpenMemHandle(&p, h, hipIpcMemLazyEnablePeerAccess));
+    return (char *)p;
+  };
+  for (size_t qi = 0; qi < nq; ++qi) {
+    v.base[0].push_back(open_one(currHandles[qi]));
+    v.base[1].push_back(open_one(nextHandles[qi]));
+  }
+  STENCIL_HIP(hipMalloc((void **)&v.devSlots, nq * sizeof(char *)));
+  STENCIL_HIP(
+      hipMemcpy(v.devSlots, v.base[0].data(), nq * sizeof(char *), hipMemcpyHostToDevice));
+  views_.push_back(std::move(v));
+  return (int64_t)views_.size() - 1;
+}
+
+void ExchangeEngine::add_translate_view(int srcDom, int64_t view, const Vec3 &srcPos,
+                                        const Vec3 &dstPos, const Vec3 &ext) {
+  translateSpecs_.push_back({srcDom, (int)view, true, srcPos, dstPos, ext});
+}
+
+void ExchangeEngine::flip_views() {
+  for (auto &v : views_) {
+    v.parity ^= 1;
+    STENCIL_HIP(hipSetDevice(v.openDev));
+    STENCIL_HIP(hipMemcpy(v.devSlots, v.base[v.parity].data(),
+                          v.base[v.parity].size() * sizeof(char *), hipMemcpyHostToDevice));
+  }
 }
 
 int64_t ExchangeEngine::create_buffer(int dom, int64_t bytes) {
@@ -192,20 +249,33 @@ void ExchangeEngine::build_batches_(const std::vector<TranslateSpec> &ts,
 
   for (const auto &t : ts) {
     LocalDomain &s = *domains_[t.srcDom];
-    LocalDomain &d = *domains_[t.dstDom];
     for (int64_t qi = 0; qi < s.num_data(); ++qi) {
       const int64_t es = s.elem_size(qi);
       const Pitched &sp = s.curr(qi); // pitch/ysize only; base via slot
-      const Pitched &dp = d.curr(qi);
+      int64_t dPitch, dPlane;
+      char *const *dSlot;
+      if (t.dstIsView) {
+        RemoteView &v = views_[t.dstDom];
+        if (v.elemSize[qi] != es) throw std::runtime_error("view elem size mismatch");
+        dPitch = v.pitch[qi];
+        dPlane = v.pitch[qi] * v.ysize[qi];
+        dSlot = (char *const *)(v.devSlots + qi);
+      } else {
+        LocalDomain &d = *domains_[t.dstDom];
+        const Pitched &dp = d.curr(qi);
+        dPitch = dp.pitch;
+        dPlane = dp.plane();
+        dSlot = (char *const *)(d.dev_curr_slots() + qi);
+      }
       CopyJob j{};
       j.srcSlot = (const char *const *)(s.dev_curr_slots() + qi);
-      j.dstSlot = (char *const *)(d.dev_curr_slots() + qi);
+      j.dstSlot = dSlot;
       j.srcOff = t.srcPos.z * sp.plane() + t.srcPos.y * sp.pitch + t.srcPos.x * es;
-      j.dstOff = t.dstPos.z * dp.plane() + t.dstPos.y * dp.pitch + t.dstPos.x * es;
+      j.dstOff = t.dstPos.z * dPlane + t.dstPos.y * dPitch + t.dstPos.x * es;
       j.srcPitch = sp.pitch;
       j.srcPlane = sp.plane();
-      j.dstPitch = dp.pitch;
-      j.dstPlane = dp.plane();
+      j.dstPitch = dPitch;
+      j.dstPlane = dPlane;
       const int64_t rowBytes = t.ext.x * es;
       const int w = pick_word(rowBytes, {j.srcOff, j.dstOff, j.srcPitch, j.dstPitch});
       j.wordBytes = w;

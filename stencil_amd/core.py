@@ -34,7 +34,8 @@ class Method(enum.Flag):
     NONE = 0
     DIRECT_KERNEL = 1  # same-process direct-write translate kernels (xGMI peer stores)
     RCCL = 2  # cross-process packed RCCL point-to-point
-    DEFAULT = DIRECT_KERNEL | RCCL
+    IPC_KERNEL = 4  # colocated cross-process direct writes via HIP IPC over xGMI
+    DEFAULT = DIRECT_KERNEL | RCCL | IPC_KERNEL
 
 
 class DataHandle:
@@ -77,7 +78,7 @@ class DistributedDomain:
         self.backend = None
         self._realized = False
         # per-method exchanged bytes (one full exchange)
-        self.bytes_by_method: Dict[str, int] = {"direct_kernel": 0, "rccl": 0}
+        self.bytes_by_method: Dict[str, int] = {"direct_kernel": 0, "rccl": 0, "ipc_kernel": 0}
         self.time_exchange = 0.0
         self.time_swap = 0.0
 
@@ -160,7 +161,12 @@ class DistributedDomain:
             raise RuntimeError("same-rank halos require Method.DIRECT_KERNEL")
         if not (self.methods & Method.RCCL) and (plan.sends or plan.recvs):
             raise RuntimeError("cross-rank halos require Method.RCCL")
-        self.backend.register_plan(plan)
+        ctx = {
+            "comm": self.comm,
+            "placement": self.placement,
+            "ipc": bool(self.methods & Method.IPC_KERNEL),
+        }
+        self.backend.register_plan(plan, ctx)
         self.plan = plan
         self._count_bytes(plan)
         if self.output_prefix:
@@ -172,8 +178,19 @@ class DistributedDomain:
         self.bytes_by_method["direct_kernel"] = sum(
             t.ext[0] * t.ext[1] * t.ext[2] * es_total for t in plan.translates
         )
+        colo = (
+            set(self.comm.colocated_ranks()) - {self.comm.rank}
+            if getattr(self.backend, "_ipc_active", False)
+            else set()
+        )
+        self.bytes_by_method["ipc_kernel"] = sum(
+            m.volume() * es_total for s in plan.sends if s.peer_rank in colo for m in s.messages
+        )
         self.bytes_by_method["rccl"] = sum(
-            m.volume() * es_total for s in plan.sends for m in s.messages
+            m.volume() * es_total
+            for s in plan.sends
+            if s.peer_rank not in colo
+            for m in s.messages
         )
 
     def exchange_bytes_for_method(self, method: Method) -> int:
@@ -182,6 +199,8 @@ class DistributedDomain:
             total += self.bytes_by_method["direct_kernel"]
         if method & Method.RCCL:
             total += self.bytes_by_method["rccl"]
+        if method & Method.IPC_KERNEL:
+            total += self.bytes_by_method["ipc_kernel"]
         return total
 
     # ---- iteration ----

@@ -91,7 +91,7 @@ class TorchBackend:
         self._send_bufs = []  # (buffer, peer, tag, [(chunk, dom, pos, ext, qi)])
         self._recv_bufs = []
 
-    def register_plan(self, plan: ExchangePlan):
+    def register_plan(self, plan: ExchangePlan, ctx=None):
         elem_sizes = [es for es, _ in self.data_defs]
         for t in plan.translates:
             src = self.domains[t.src_local]

@@ -1,7 +1,11 @@
 """Cross-process exchange with the NATIVE backend on one GPU (2 ranks,
-gloo wire with CPU staging): exercises DLPack buffer export, the HIP
-pack/unpack kernels, and the cross-process wire format. On multi-GPU
-nodes the same path runs over RCCL with device buffers."""
+gloo control plane). Two transports:
+- wire: packed buffers staged over gloo (exercises DLPack export and the
+  HIP pack/unpack kernels; on multi-GPU nodes the same path runs over
+  RCCL with device buffers)
+- ipc: direct-write translate kernels into the peer process's buffers via
+  hipIpcMemHandle (the xGMI colocated path)
+"""
 import multiprocessing as mp
 import os
 
@@ -11,11 +15,12 @@ import pytest
 pytestmark = pytest.mark.gpu
 
 
-def _worker(rank, world, port, q):
+def _worker(rank, world, port, q, use_ipc):
     try:
         os.environ["MASTER_ADDR"] = "127.0.0.1"
         os.environ["MASTER_PORT"] = str(port)
         os.environ["STENCIL_AMD_WIRE"] = "cpu"  # single GPU: stage over gloo
+        os.environ["STENCIL_AMD_IPC"] = "1" if use_ipc else "0"
         import torch.distributed as dist
 
         dist.init_process_group("gloo", rank=rank, world_size=world)
@@ -30,6 +35,10 @@ def _worker(rank, world, port, q):
         dd.set_gpus([0])
         h = dd.add_data(np.float32, "q")
         dd.realize()
+        if use_ipc:
+            assert dd.backend._ipc_active, "IPC transport did not activate"
+            assert dd.bytes_by_method["ipc_kernel"] > 0
+            assert dd.bytes_by_method["rccl"] == 0
         fill_interiors(dd, h)
         dd.exchange()
         check_full_regions(dd, h)
@@ -45,10 +54,11 @@ def _worker(rank, world, port, q):
         q.put((rank, f"FAIL: {e}\n{traceback.format_exc()}"))
 
 
-def test_native_two_ranks_one_gpu():
+@pytest.mark.parametrize("use_ipc,port", [(False, 29717), (True, 29721)])
+def test_native_two_ranks_one_gpu(use_ipc, port):
     ctx = mp.get_context("spawn")
     q = ctx.Queue()
-    procs = [ctx.Process(target=_worker, args=(r, 2, 29717, q)) for r in range(2)]
+    procs = [ctx.Process(target=_worker, args=(r, 2, port, q, use_ipc)) for r in range(2)]
     for p in procs:
         p.start()
     results = [q.get(timeout=300) for _ in procs]
