@@ -15,6 +15,7 @@
 #include "stencil_amd/ops.hpp"
 #include "stencil_amd/partition.hpp"
 #include "stencil_amd/qap.hpp"
+#include "stencil_amd/topo.hpp"
 
 namespace py = pybind11;
 using namespace stencil_amd;
@@ -257,6 +258,17 @@ PYBIND11_MODULE(_C, m) {
       .def_readwrite("chi", &MhdCoeffs::chi);
   m.def("mhd_substep", &mhd_substep);
   m.def("init_harmonic_f64", &init_harmonic_f64);
+
+  // topology utilities (csrc/src/topo.hip)
+  m.def("gpu_distance", &gpu_distance);
+  m.def("peer_copy_bandwidth", &peer_copy_bandwidth, py::arg("src"), py::arg("dst"),
+        py::arg("bytes"), py::arg("iters") = 10);
+  py::class_<GpuInfo>(m, "GpuInfo")
+      .def_readonly("name", &GpuInfo::name)
+      .def_readonly("pci", &GpuInfo::pci)
+      .def_readonly("total_mem", &GpuInfo::totalMem)
+      .def_readonly("cu_count", &GpuInfo::cuCount);
+  m.def("gpu_info", &gpu_info);
 
   m.def("device_count", []() {
     int n = 0;

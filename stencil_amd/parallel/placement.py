@@ -167,17 +167,18 @@ def _comm_matrix(gids: List[int], placement: Placement, radius: "_C.Radius") -> 
 
 
 def _bandwidth_matrix(cudas: List[int]) -> "_C.SqMat":
-    """distance = 1/bandwidth between GPU slots. Same-GPU is near-zero
-    distance; distinct MI355X GPUs are one xGMI hop (uniform)."""
+    """distance between GPU slots from the native link discovery
+    (hipExtGetLinkTypeAndHopCount; same-GPU near-zero, xGMI peers one hop,
+    non-peer heavily penalized). Without GPUs a uniform matrix is used."""
     n = len(cudas)
     d = _C.SqMat(n, 0.0)
+    have_gpu = _C.device_count() > 0
     for i in range(n):
         for j in range(n):
-            if cudas[i] == cudas[j]:
-                d.set(i, j, 0.1)
+            if not have_gpu:
+                d.set(i, j, 0.1 if cudas[i] == cudas[j] else 1.0)
             else:
-                peer = _C.ExchangeEngine.can_access_peer(cudas[i], cudas[j])
-                d.set(i, j, 1.0 if peer else 5.0)
+                d.set(i, j, _C.gpu_distance(cudas[i], cudas[j]))
     return d
 
 
