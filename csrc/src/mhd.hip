@@ -148,78 +148,120 @@ struct Stencil {
   }
 };
 
-__global__ void __launch_bounds__(256) mhd_substep_kernel(MhdParams p) {
-  const int32_t lx = blockIdx.x * blockDim.x + threadIdx.x;
-  const int32_t ly = blockIdx.y * blockDim.y + threadIdx.y;
-  const int32_t lz = blockIdx.z;
-  if (lx >= p.extX || ly >= p.extY || lz >= p.extZ) return;
+// The solver is split into two kernels per substep: the monolithic form
+// allocated 356 VGPR+AGPR (1 wave/SIMD) and latency-bound on its dependent
+// cache loads. Split, each kernel fits >=2 waves/SIMD; the duplicated
+// center reads are served by L1/L2.
+
+struct MhdCommon {
+  const char *base[8];
+  int64_t pitch, plane;
+  char *out[8];
+};
+
+__device__ __forceinline__ MhdCommon mhd_setup(const MhdParams &p, int32_t lx, int32_t ly,
+                                               int32_t lz) {
   const int64_t ax = p.loX + lx - p.allocX;
   const int64_t ay = p.loY + ly - p.allocY;
   const int64_t az = p.loZ + lz - p.allocZ;
   const int64_t cellOff = az * p.plane + ay * p.pitch + ax * 8;
-
-  Stencil st;
-  st.pitch = p.pitch;
-  st.plane = p.plane;
+  MhdCommon c;
+  c.pitch = p.pitch;
+  c.plane = p.plane;
 #pragma unroll
-  for (int q = 0; q < 8; ++q) st.base[q] = p.currSlots[q] + cellOff;
+  for (int q = 0; q < 8; ++q) {
+    c.base[q] = p.currSlots[q] + cellOff;
+    c.out[q] = p.nextSlots[q] + cellOff;
+  }
+  return c;
+}
 
+__device__ __forceinline__ void write_rk3(const MhdParams &p, const Stencil &st, char *out, int q,
+                                          double r) {
+  const double cur = st.c(q);
+  const double prev = *(const double *)out;
+  *(double *)out = cur + p.beta * (p.alpha_over_beta_prev * (cur - prev) + p.dt * r);
+}
+
+// kernel 1: continuity + entropy + induction (lnrho, ss, aa). First and
+// second derivatives only, no cross terms.
+__global__ void __launch_bounds__(256) mhd_scalar_kernel(MhdParams p) {
+  const int32_t lx = blockIdx.x * blockDim.x + threadIdx.x;
+  const int32_t ly = blockIdx.y * blockDim.y + threadIdx.y;
+  const int32_t lz = blockIdx.z;
+  if (lx >= p.extX || ly >= p.extY || lz >= p.extZ) return;
+  const MhdCommon c = mhd_setup(p, lx, ly, lz);
+  Stencil st;
+  st.pitch = c.pitch;
+  st.plane = c.plane;
+#pragma unroll
+  for (int q = 0; q < 8; ++q) st.base[q] = c.base[q];
   const double ix = 1.0 / p.dsx, iy = 1.0 / p.dsy, iz = 1.0 / p.dsz;
-
   const Vec3d uu = {st.c(UUX), st.c(UUY), st.c(UUZ)};
-  const Vec3d glnrho = st.grad(LNRHO, ix, iy, iz);
-  const Vec3d gss = st.grad(SS, ix, iy, iz);
-
-  // velocity gradient / laplacian / grad(div u)
-  const double uxx = st.dx(UUX, ix), uxy = st.dy(UUX, iy), uxz = st.dz(UUX, iz);
-  const double uyx = st.dx(UUY, ix), uyy = st.dy(UUY, iy), uyz = st.dz(UUY, iz);
-  const double uzx = st.dx(UUZ, ix), uzy = st.dy(UUZ, iy), uzz = st.dz(UUZ, iz);
-  const double divu = uxx + uyy + uzz;
-  const Vec3d lap_u = {st.lap(UUX, ix, iy, iz), st.lap(UUY, ix, iy, iz), st.lap(UUZ, ix, iy, iz)};
-  const Vec3d graddiv_u = {
-      st.dxx(UUX, ix * ix) + st.dxy(UUY, ix, iy) + st.dxz(UUZ, ix, iz),
-      st.dxy(UUX, ix, iy) + st.dyy(UUY, iy * iy) + st.dyz(UUZ, iy, iz),
-      st.dxz(UUX, ix, iz) + st.dyz(UUY, iy, iz) + st.dzz(UUZ, iz * iz),
-  };
-
-  // magnetic field B = curl(A), current j = grad(div A) - lap(A)
+  {
+    const Vec3d glnrho = st.grad(LNRHO, ix, iy, iz);
+    const double divu = st.dx(UUX, ix) + st.dy(UUY, iy) + st.dz(UUZ, iz);
+    write_rk3(p, st, c.out[LNRHO], LNRHO, -dot(uu, glnrho) - divu);
+  }
+  {
+    const Vec3d gss = st.grad(SS, ix, iy, iz);
+    write_rk3(p, st, c.out[SS], SS, -dot(uu, gss) + p.chi * st.lap(SS, ix, iy, iz));
+  }
   const Vec3d B = {st.dy(AAZ, iy) - st.dz(AAY, iz), st.dz(AAX, iz) - st.dx(AAZ, ix),
                    st.dx(AAY, ix) - st.dy(AAX, iy)};
-  const Vec3d lap_a = {st.lap(AAX, ix, iy, iz), st.lap(AAY, ix, iy, iz), st.lap(AAZ, ix, iy, iz)};
+  const Vec3d uxB = cross(uu, B);
+  write_rk3(p, st, c.out[AAX], AAX, uxB.x + p.eta * st.lap(AAX, ix, iy, iz));
+  write_rk3(p, st, c.out[AAY], AAY, uxB.y + p.eta * st.lap(AAY, ix, iy, iz));
+  write_rk3(p, st, c.out[AAZ], AAZ, uxB.z + p.eta * st.lap(AAZ, ix, iy, iz));
+}
+
+// kernel 2: momentum (uu). Needs B, j (with cross derivatives of A),
+// pressure gradient, advection, viscosity (cross derivatives of u).
+__global__ void __launch_bounds__(256) mhd_momentum_kernel(MhdParams p) {
+  const int32_t lx = blockIdx.x * blockDim.x + threadIdx.x;
+  const int32_t ly = blockIdx.y * blockDim.y + threadIdx.y;
+  const int32_t lz = blockIdx.z;
+  if (lx >= p.extX || ly >= p.extY || lz >= p.extZ) return;
+  const MhdCommon c = mhd_setup(p, lx, ly, lz);
+  Stencil st;
+  st.pitch = c.pitch;
+  st.plane = c.plane;
+#pragma unroll
+  for (int q = 0; q < 8; ++q) st.base[q] = c.base[q];
+  const double ix = 1.0 / p.dsx, iy = 1.0 / p.dsy, iz = 1.0 / p.dsz;
+  const Vec3d uu = {st.c(UUX), st.c(UUY), st.c(UUZ)};
+  const Vec3d B = {st.dy(AAZ, iy) - st.dz(AAY, iz), st.dz(AAX, iz) - st.dx(AAZ, ix),
+                   st.dx(AAY, ix) - st.dy(AAX, iy)};
+  const Vec3d lap_a = {st.lap(AAX, ix, iy, iz), st.lap(AAY, ix, iy, iz),
+                       st.lap(AAZ, ix, iy, iz)};
   const Vec3d graddiv_a = {
       st.dxx(AAX, ix * ix) + st.dxy(AAY, ix, iy) + st.dxz(AAZ, ix, iz),
       st.dxy(AAX, ix, iy) + st.dyy(AAY, iy * iy) + st.dyz(AAZ, iy, iz),
       st.dxz(AAX, ix, iz) + st.dyz(AAY, iy, iz) + st.dzz(AAZ, iz * iz),
   };
-  const Vec3d j = graddiv_a - lap_a;
+  const Vec3d jxB = cross(graddiv_a - lap_a, B);
   const double rho_inv = exp(-st.c(LNRHO));
 
-  // right-hand sides
-  double rhs[8];
-  rhs[LNRHO] = -dot(uu, glnrho) - divu;
-  const Vec3d ugradu = {uu.x * uxx + uu.y * uxy + uu.z * uxz,
-                        uu.x * uyx + uu.y * uyy + uu.z * uyz,
-                        uu.x * uzx + uu.y * uzy + uu.z * uzz};
-  const Vec3d jxB = cross(j, B);
-  const Vec3d press = glnrho + p.cp_inv * gss;
-  rhs[UUX] = -ugradu.x - p.cs2 * press.x + rho_inv * jxB.x + p.nu * (lap_u.x + graddiv_u.x / 3.0);
-  rhs[UUY] = -ugradu.y - p.cs2 * press.y + rho_inv * jxB.y + p.nu * (lap_u.y + graddiv_u.y / 3.0);
-  rhs[UUZ] = -ugradu.z - p.cs2 * press.z + rho_inv * jxB.z + p.nu * (lap_u.z + graddiv_u.z / 3.0);
-  const Vec3d uxB = cross(uu, B);
-  rhs[AAX] = uxB.x + p.eta * lap_a.x;
-  rhs[AAY] = uxB.y + p.eta * lap_a.y;
-  rhs[AAZ] = uxB.z + p.eta * lap_a.z;
-  rhs[SS] = -dot(uu, gss) + p.chi * st.lap(SS, ix, iy, iz);
-
-  // Williamson RK3 two-buffer update: next holds the previous substep's
-  // state; write the new state over it
-#pragma unroll
-  for (int q = 0; q < 8; ++q) {
-    char *out = p.nextSlots[q] + cellOff;
-    const double cur = st.c(q);
-    const double prev = *(const double *)out;
-    const double w = p.alpha_over_beta_prev * (cur - prev) + p.dt * rhs[q];
-    *(double *)out = cur + p.beta * w;
+  {
+    const double ugradu = uu.x * st.dx(UUX, ix) + uu.y * st.dy(UUX, iy) + uu.z * st.dz(UUX, iz);
+    const double press = st.dx(LNRHO, ix) + p.cp_inv * st.dx(SS, ix);
+    const double graddiv = st.dxx(UUX, ix * ix) + st.dxy(UUY, ix, iy) + st.dxz(UUZ, ix, iz);
+    const double visc = p.nu * (st.lap(UUX, ix, iy, iz) + graddiv / 3.0);
+    write_rk3(p, st, c.out[UUX], UUX, -ugradu - p.cs2 * press + rho_inv * jxB.x + visc);
+  }
+  {
+    const double ugradu = uu.x * st.dx(UUY, ix) + uu.y * st.dy(UUY, iy) + uu.z * st.dz(UUY, iz);
+    const double press = st.dy(LNRHO, iy) + p.cp_inv * st.dy(SS, iy);
+    const double graddiv = st.dxy(UUX, ix, iy) + st.dyy(UUY, iy * iy) + st.dyz(UUZ, iy, iz);
+    const double visc = p.nu * (st.lap(UUY, ix, iy, iz) + graddiv / 3.0);
+    write_rk3(p, st, c.out[UUY], UUY, -ugradu - p.cs2 * press + rho_inv * jxB.y + visc);
+  }
+  {
+    const double ugradu = uu.x * st.dx(UUZ, ix) + uu.y * st.dy(UUZ, iy) + uu.z * st.dz(UUZ, iz);
+    const double press = st.dz(LNRHO, iz) + p.cp_inv * st.dz(SS, iz);
+    const double graddiv = st.dxz(UUX, ix, iz) + st.dyz(UUY, iy, iz) + st.dzz(UUZ, iz * iz);
+    const double visc = p.nu * (st.lap(UUZ, ix, iy, iz) + graddiv / 3.0);
+    write_rk3(p, st, c.out[UUZ], UUZ, -ugradu - p.cs2 * press + rho_inv * jxB.z + visc);
   }
 }
 
@@ -266,7 +308,9 @@ void mhd_substep(ExchangeEngine &eng, int dom, const Rect3 &region, int step, do
   STENCIL_HIP(hipSetDevice(d.gpu()));
   dim3 block(64, 4, 1);
   dim3 grid((uint32_t)((ext.x + 63) / 64), (uint32_t)((ext.y + 3) / 4), (uint32_t)ext.z);
-  hipLaunchKernelGGL(mhd_substep_kernel, grid, block, 0, eng.compute_stream(dom), p);
+  hipLaunchKernelGGL(mhd_scalar_kernel, grid, block, 0, eng.compute_stream(dom), p);
+  STENCIL_HIP(hipGetLastError());
+  hipLaunchKernelGGL(mhd_momentum_kernel, grid, block, 0, eng.compute_stream(dom), p);
   STENCIL_HIP(hipGetLastError());
 }
 
