@@ -55,8 +55,10 @@ struct CopyBatch {
   CopyJob *dJobs = nullptr;
   int64_t *dPrefix = nullptr;
   int64_t nBlocks = 0;
+  hipGraphExec_t graphExec = nullptr; // optional hipGraph replay
 
   void finalize_upload();
+  void capture_graph(); // capture the batch launch into a hipGraph
   void launch(hipStream_t stream);
   void destroy();
 };

@@ -4,6 +4,7 @@
 #include "stencil_amd/hip_check.hpp"
 
 #include <algorithm>
+#include <cstdlib>
 #include <cstring>
 #include <initializer_list>
 #include <numeric>
@@ -97,15 +98,41 @@ void CopyBatch::finalize_upload() {
       hipMemcpy(dPrefix, prefix.data(), prefix.size() * sizeof(int64_t), hipMemcpyHostToDevice));
 }
 
+// Optional: wrap the batch launch in a hipGraph and replay it. With the
+// job-table design an exchange is already 1-2 launches per GPU (the
+// reference needed graphs to amortize its 26 x nQuant launch storm,
+// src/packer.cu:96-106), so this mainly shaves host launch latency.
+// Enabled by STENCIL_AMD_GRAPHS=1.
+void CopyBatch::capture_graph() {
+  if (jobs.empty()) return;
+  STENCIL_HIP(hipSetDevice(dev));
+  hipStream_t cap;
+  STENCIL_HIP(hipStreamCreateWithFlags(&cap, hipStreamNonBlocking));
+  STENCIL_HIP(hipStreamBeginCapture(cap, hipStreamCaptureModeThreadLocal));
+  hipLaunchKernelGGL(copy_batch_kernel, dim3((uint32_t)nBlocks), dim3(kBlock), 0, cap, dJobs,
+                     dPrefix, (int)jobs.size());
+  hipGraph_t graph;
+  STENCIL_HIP(hipStreamEndCapture(cap, &graph));
+  STENCIL_HIP(hipGraphInstantiate(&graphExec, graph, nullptr, nullptr, 0));
+  STENCIL_HIP(hipGraphDestroy(graph));
+  STENCIL_HIP(hipStreamDestroy(cap));
+}
+
 void CopyBatch::launch(hipStream_t stream) {
   if (jobs.empty()) return;
   STENCIL_HIP(hipSetDevice(dev));
+  if (graphExec) {
+    STENCIL_HIP(hipGraphLaunch(graphExec, stream));
+    return;
+  }
   hipLaunchKernelGGL(copy_batch_kernel, dim3((uint32_t)nBlocks), dim3(kBlock), 0, stream, dJobs,
                      dPrefix, (int)jobs.size());
   STENCIL_HIP(hipGetLastError());
 }
 
 void CopyBatch::destroy() {
+  if (graphExec) (void)hipGraphExecDestroy(graphExec);
+  graphExec = nullptr;
   if (dJobs) (void)hipFree(dJobs);
   if (dPrefix) (void)hipFree(dPrefix);
   dJobs = nullptr;
@@ -237,9 +264,13 @@ void ExchangeEngine::add_unpack(int dom, int64_t buf, int64_t offset, const Vec3
 void ExchangeEngine::finalize() {
   if (finalized_) throw std::runtime_error("ExchangeEngine::finalize called twice");
   build_batches_(translateSpecs_, packSpecs_);
-  for (auto &b : translateBatches_) b.finalize_upload();
-  for (auto &b : packBatches_) b.finalize_upload();
-  for (auto &b : unpackBatches_) b.finalize_upload();
+  const char *g = getenv("STENCIL_AMD_GRAPHS");
+  const bool graphs = g && g[0] == '1';
+  for (auto *set : {&translateBatches_, &packBatches_, &unpackBatches_})
+    for (auto &b : *set) {
+      b.finalize_upload();
+      if (graphs) b.capture_graph();
+    }
   finalized_ = true;
 }
 

@@ -81,6 +81,9 @@ class DistributedDomain:
         self.bytes_by_method: Dict[str, int] = {"direct_kernel": 0, "rccl": 0, "ipc_kernel": 0}
         self.time_exchange = 0.0
         self.time_swap = 0.0
+        # setup-phase timers (reference STENCIL_SETUP_STATS,
+        # stencil.hpp:103-112)
+        self.setup_times: Dict[str, float] = {}
 
     # ---- configuration (before realize) ----
     def set_radius(self, r):
@@ -123,11 +126,15 @@ class DistributedDomain:
 
     def do_placement(self):
         """compute partition + placement only (no allocation)"""
+        t0 = time.perf_counter()
         self.comm = Comm()
         if self.gpus is None:
             self.gpus = self._default_gpus()
         slots = gather_slots(self.comm, self.gpus)
+        self.setup_times["topo"] = time.perf_counter() - t0
+        t0 = time.perf_counter()
         self.placement = make_placement(self.strategy, self.size, self.radius, slots)
+        self.setup_times["placement"] = time.perf_counter() - t0
         return self.placement
 
     def realize(self):
@@ -147,6 +154,7 @@ class DistributedDomain:
                     self.placement.get_cuda(idx),
                 )
             )
+        t0 = time.perf_counter()
         if self.backend_kind == "native":
             from .native_backend import NativeBackend
 
@@ -155,8 +163,11 @@ class DistributedDomain:
             from .torch_backend import TorchBackend
 
             self.backend = TorchBackend(specs, self._data, self.radius, self.torch_device)
+        self.setup_times["realize"] = time.perf_counter() - t0
 
+        t0 = time.perf_counter()
         plan = plan_exchange(self.placement, self.radius, rank)
+        self.setup_times["plan"] = time.perf_counter() - t0
         if not (self.methods & Method.DIRECT_KERNEL) and plan.translates:
             raise RuntimeError("same-rank halos require Method.DIRECT_KERNEL")
         if not (self.methods & Method.RCCL) and (plan.sends or plan.recvs):
@@ -166,7 +177,9 @@ class DistributedDomain:
             "placement": self.placement,
             "ipc": bool(self.methods & Method.IPC_KERNEL),
         }
+        t0 = time.perf_counter()
         self.backend.register_plan(plan, ctx)
+        self.setup_times["create"] = time.perf_counter() - t0
         self.plan = plan
         self._count_bytes(plan)
         if self.output_prefix:
