@@ -126,33 +126,60 @@ class NativeBackend:
         instead of IPC events)"""
         comm, placement = ctx["comm"], ctx["placement"]
         radius = self.radius
-        export = []
-        for d in self.domains:
-            nq = d.num_data()
-            export.append(
-                {
-                    "gpu": d.gpu(),
-                    "pitch": [d.curr(qi).pitch for qi in range(nq)],
-                    "ysize": [d.curr(qi).ysize for qi in range(nq)],
-                    "es": [d.elem_size(qi) for qi in range(nq)],
-                    "curr": [d.ipc_handle(qi, False) for qi in range(nq)],
-                    "next": [d.ipc_handle(qi, True) for qi in range(nq)],
-                }
-            )
+        # phase 1: export handles. A failing rank must still reach the
+        # allgather (its peers would otherwise hang), so errors are
+        # exported as data and re-raised on EVERY rank afterwards.
+        try:
+            export = []
+            for d in self.domains:
+                nq = d.num_data()
+                export.append(
+                    {
+                        "gpu": d.gpu(),
+                        "pitch": [d.curr(qi).pitch for qi in range(nq)],
+                        "ysize": [d.curr(qi).ysize for qi in range(nq)],
+                        "es": [d.elem_size(qi) for qi in range(nq)],
+                        "curr": [d.ipc_handle(qi, False) for qi in range(nq)],
+                        "next": [d.ipc_handle(qi, True) for qi in range(nq)],
+                    }
+                )
+        except Exception as e:
+            export = {"error": str(e)}
         infos = comm.allgather_object(export)
+        errs = [i["error"] for i in infos if isinstance(i, dict) and "error" in i]
+        if errs:
+            raise RuntimeError(f"IPC export failed on some rank: {errs[0]}")
 
+        # phase 2: open every view first (a failure here leaves no
+        # translate specs registered, keeping the RCCL fallback clean),
+        # then register the direct-write jobs.
         views = {}
+        open_err = None
+        try:
+            for s in ipc_sends:
+                dst_idx = placement.dimensionize(s.dst_gid)
+                dst_li = placement.get_subdomain_id(dst_idx)
+                src_gpu = self.domains[s.local_id].gpu()
+                key = (s.peer_rank, dst_li, src_gpu)
+                if key not in views:
+                    info = infos[s.peer_rank][dst_li]
+                    views[key] = self.engine.create_remote_view(
+                        src_gpu, info["curr"], info["next"], info["pitch"], info["ysize"], info["es"]
+                    )
+        except Exception as e:
+            open_err = str(e)
+        # consensus: either EVERY rank uses IPC or none does (a mixed state
+        # would deadlock: IPC ranks at the barrier, RCCL ranks at a recv)
+        votes = comm.allgather_object(open_err)
+        bad = [v for v in votes if v is not None]
+        if bad:
+            raise RuntimeError(f"IPC open failed on some rank: {bad[0]}")
         for s in ipc_sends:
             dst_idx = placement.dimensionize(s.dst_gid)
             dst_li = placement.get_subdomain_id(dst_idx)
             dst_size = placement.subdomain_size(dst_idx)
             src_gpu = self.domains[s.local_id].gpu()
             key = (s.peer_rank, dst_li, src_gpu)
-            if key not in views:
-                info = infos[s.peer_rank][dst_li]
-                views[key] = self.engine.create_remote_view(
-                    src_gpu, info["curr"], info["next"], info["pitch"], info["ysize"], info["es"]
-                )
             dom = self.domains[s.local_id]
             for m in s.messages:
                 nd = _vec3(tuple(-c for c in m.dir))
