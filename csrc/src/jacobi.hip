@@ -14,6 +14,8 @@
 #include <hip/hip_runtime.h>
 
 #include <algorithm>
+#include <cstdio>
+#include <cstdlib>
 
 #include "stencil_amd/domain.hpp"
 #include "stencil_amd/engine.hpp"
@@ -150,7 +152,10 @@ __global__ void __launch_bounds__(256) jacobi_kernel_v4(JacobiParams p) {
       out.z = (cc.w + cc.y + py.z + my.z + cp.z + cm.z) / 6.0f;
       out.w = (right + cc.z + py.w + my.w + cp.w + cm.w) / 6.0f;
       sphere4(gx, (int32_t)gy, (int32_t)(p.loZ + lz), out);
-      *(float4 *)dcol = out;
+      // next is write-only this iteration: bypass cache pollution
+      typedef float vfloat4 __attribute__((ext_vector_type(4)));
+      vfloat4 ov = {out.x, out.y, out.z, out.w};
+      __builtin_nontemporal_store(ov, (vfloat4 *)dcol);
       cm = cc;
       cc = cp;
       col += p.plane;
@@ -246,12 +251,23 @@ void jacobi_step(ExchangeEngine &eng, int dom, int64_t qi, const Rect3 &region,
   p.cHiZ = computeRegion.hi.z;
   STENCIL_HIP(hipSetDevice(d.gpu()));
   if (ext.x >= 8 && ext.y <= 0x7fffffff) {
-    // vectorized row-mapped kernel
+    // vectorized row-mapped kernel; block shape tunable via env
     const int64_t a0 = region.lo.x - full.lo.x;
     const int64_t head = std::min<int64_t>(((a0 + 3) & ~3LL) - a0, ext.x);
     const int64_t units = (ext.x - head) / 4 + 2;
-    dim3 block(64, 4, 1);
-    dim3 grid((uint32_t)((units + 63) / 64), (uint32_t)((ext.y + 3) / 4),
+    static int bx = 0, by = 0;
+    if (!bx) {
+      bx = 64;
+      by = 4;
+      if (const char *e = getenv("STENCIL_JAC_BLOCK")) {
+        if (sscanf(e, "%dx%d", &bx, &by) != 2 || bx * by != 256) {
+          bx = 64;
+          by = 4;
+        }
+      }
+    }
+    dim3 block((uint32_t)bx, (uint32_t)by, 1);
+    dim3 grid((uint32_t)((units + bx - 1) / bx), (uint32_t)((ext.y + by - 1) / by),
               (uint32_t)((ext.z + 15) / 16)); // 16 == JAC_ZCHUNK
     hipLaunchKernelGGL(jacobi_kernel_v4, grid, block, 0, eng.compute_stream(dom), p);
   } else {

@@ -56,6 +56,7 @@ class NativeBackend:
 
         self._ipc_active = False
         self._colo_group = None
+        self._ipc_error = None
         ipc_sends, ipc_recv_peers = [], set()
         if ctx and ctx.get("ipc", True) and os.environ.get("STENCIL_AMD_IPC", "1") != "0":
             comm = ctx["comm"]
@@ -72,8 +73,10 @@ class NativeBackend:
                     )
                     self._ipc_active = True
                 except Exception as e:  # pragma: no cover - fallback path
+                    import traceback
                     import warnings
 
+                    self._ipc_error = traceback.format_exc()
                     warnings.warn(f"HIP IPC transport unavailable ({e}); falling back to RCCL")
                     ipc_sends = []
         self._make_colo_groups(ctx)

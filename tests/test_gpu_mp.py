@@ -36,7 +36,9 @@ def _worker(rank, world, port, q, use_ipc):
         h = dd.add_data(np.float32, "q")
         dd.realize()
         if use_ipc:
-            assert dd.backend._ipc_active, "IPC transport did not activate"
+            assert dd.backend._ipc_active, (
+                f"IPC transport did not activate: {dd.backend._ipc_error}"
+            )
             assert dd.bytes_by_method["ipc_kernel"] > 0
             assert dd.bytes_by_method["rccl"] == 0
         fill_interiors(dd, h)
