@@ -1,0 +1,80 @@
+"""Placement strategy tests (no GPU): trivial round-robin, node-aware QAP
+over fake multi-node slot lists, random control, and the placement query
+interface consistency."""
+import pytest
+
+from stencil_amd import _C
+from stencil_amd.parallel.placement import (
+    IntraNodeRandomPlacement,
+    NodeAwarePlacement,
+    Slot,
+    TrivialPlacement,
+)
+
+
+def slots_multi_node(n_nodes, gpus_per_node):
+    out = []
+    rank = 0
+    for node in range(n_nodes):
+        for g in range(gpus_per_node):
+            out.append(Slot(rank, 0, g, node))
+            rank += 1
+    return out
+
+
+@pytest.mark.parametrize("cls", [TrivialPlacement, NodeAwarePlacement, IntraNodeRandomPlacement])
+@pytest.mark.parametrize("n_nodes,gpn", [(1, 8), (2, 4), (4, 2)])
+def test_placement_is_bijection(cls, n_nodes, gpn):
+    r = _C.Radius.constant(1)
+    slots = slots_multi_node(n_nodes, gpn)
+    p = cls((64, 64, 64), r, slots)
+    d = p.dim()
+    assert d[0] * d[1] * d[2] == n_nodes * gpn
+    seen = set()
+    for z in range(d[2]):
+        for y in range(d[1]):
+            for x in range(d[0]):
+                idx = (x, y, z)
+                key = (p.get_rank(idx), p.get_subdomain_id(idx))
+                assert key not in seen
+                seen.add(key)
+                # inverse mapping consistent
+                assert p.get_idx(*key) == idx
+
+
+def test_node_aware_keeps_subdomains_on_their_node():
+    """the node-level split must assign each node's block of subdomains to
+    ranks of that node"""
+    r = _C.Radius.constant(2)
+    slots = slots_multi_node(2, 4)
+    p = NodeAwarePlacement((64, 64, 64), r, slots)
+    d = p.dim()
+    node_of_rank = {s.rank: s.node for s in slots}
+    for node in range(2):
+        gids = p._node_gids(node)
+        for gid in gids:
+            idx = p.dimensionize(gid)
+            assert node_of_rank[p.get_rank(idx)] == node
+
+
+def test_random_placement_seed_deterministic():
+    r = _C.Radius.constant(1)
+    a = IntraNodeRandomPlacement((32, 32, 32), r, slots_multi_node(1, 8), seed=0)
+    b = IntraNodeRandomPlacement((32, 32, 32), r, slots_multi_node(1, 8), seed=0)
+    assert a.assign == b.assign
+
+
+def test_uneven_nodes_raise():
+    r = _C.Radius.constant(1)
+    slots = slots_multi_node(2, 4)
+    slots.append(Slot(8, 0, 4, 1))  # node 1 gets a 5th GPU
+    with pytest.raises(ValueError):
+        TrivialPlacement((30, 30, 30), r, slots)
+
+
+def test_prime_gpu_count_partitions():
+    """7 GPUs -> a 7-way split along one axis still covers the grid"""
+    r = _C.Radius.constant(1)
+    p = TrivialPlacement((28, 30, 30), r, slots_multi_node(1, 7))
+    d = p.dim()
+    assert d[0] * d[1] * d[2] == 7
