@@ -67,6 +67,24 @@ def build(verbose=True):
         if verbose:
             print("[link]", " ".join(cmd), flush=True)
         subprocess.check_call(cmd)
+
+    # native C++ example binaries (no Python dependency)
+    core_objs = [o for o in objs if o.stem != "bindings"]
+    for ex in sorted((REPO / "examples").glob("*.cpp")):
+        exe = BUILD / ex.stem
+        if exe.exists() and exe.stat().st_mtime > max(
+            [ex.stat().st_mtime, hmtime] + [o.stat().st_mtime for o in core_objs]
+        ):
+            continue
+        exo = BUILD / (ex.stem + "_main.o")
+        cmd = [HIPCC, "-x", "hip", "-c", str(ex), "-o", str(exo)] + cflags
+        if verbose:
+            print("[exe-compile]", " ".join(cmd), flush=True)
+        subprocess.check_call(cmd)
+        cmd = [HIPCC, str(exo)] + [str(o) for o in core_objs] + ["-o", str(exe)]
+        if verbose:
+            print("[exe-link]", " ".join(cmd), flush=True)
+        subprocess.check_call(cmd)
     return out
 
 
