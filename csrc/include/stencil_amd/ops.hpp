@@ -27,8 +27,10 @@ struct MhdCoeffs {
 // one RK3 substep (step in 0..2) of the 8-field 6th-order MHD system over
 // `region` (global coords); reads curr, updates next in place (Williamson
 // two-buffer form) -- caller swaps after each substep
+// scratchBuf: engine buffer (>= 3 * region volume * 8 bytes) holding the
+// intermediate Lorentz force between the kernel passes
 void mhd_substep(ExchangeEngine &eng, int dom, const Rect3 &region, int step, double dt,
-                 const MhdCoeffs &cf);
+                 const MhdCoeffs &cf, int64_t scratchBuf);
 
 // fill an fp64 region with base + amp*sin(kx*x + ky*y + kz*z + phase)
 // (deterministic smooth initial conditions, reproducible in NumPy)

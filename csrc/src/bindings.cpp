@@ -170,6 +170,8 @@ PYBIND11_MODULE(_C, m) {
              return py::bytes(out);
            },
            py::arg("pos"), py::arg("ext"), py::arg("qi"), py::arg("from_next") = false)
+      .def("curr_pitch", [](const LocalDomain &d, int64_t qi) { return d.curr(qi).pitch; })
+      .def("curr_ysize", [](const LocalDomain &d, int64_t qi) { return d.curr(qi).ysize; })
       .def("ipc_handle",
            [](const LocalDomain &d, int64_t qi, bool next) { return py::bytes(d.ipc_handle(qi, next)); },
            py::arg("qi"), py::arg("next") = false)

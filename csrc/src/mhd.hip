@@ -215,9 +215,53 @@ __global__ void __launch_bounds__(256) mhd_scalar_kernel(MhdParams p) {
   write_rk3(p, st, c.out[AAZ], AAZ, uxB.z + p.eta * st.lap(AAZ, ix, iy, iz));
 }
 
-// kernel 2: momentum (uu). Needs B, j (with cross derivatives of A),
-// pressure gradient, advection, viscosity (cross derivatives of u).
-__global__ void __launch_bounds__(256) mhd_momentum_kernel(MhdParams p) {
+// kernel 2a: Lorentz force j x B into a scratch array (j needs the cross
+// derivatives of A -- the most load-heavy part of the solver; isolating it
+// keeps each kernel's register set small enough for >=2 waves/SIMD)
+struct MhdScratch {
+  char *ptr;           // 3 consecutive (z,y,x) fp64 arrays over the region
+  int64_t rowStride;   // extX * 8
+  int64_t planeStride; // rowStride * extY
+  int64_t compStride;  // planeStride * extZ
+};
+
+__global__ void __launch_bounds__(256) mhd_lorentz_kernel(MhdParams p, MhdScratch sc) {
+  const int32_t lx = blockIdx.x * blockDim.x + threadIdx.x;
+  const int32_t ly = blockIdx.y * blockDim.y + threadIdx.y;
+  const int32_t lz = blockIdx.z;
+  if (lx >= p.extX || ly >= p.extY || lz >= p.extZ) return;
+  const MhdCommon c = mhd_setup(p, lx, ly, lz);
+  Stencil st;
+  st.pitch = c.pitch;
+  st.plane = c.plane;
+#pragma unroll
+  for (int q = 0; q < 8; ++q) st.base[q] = c.base[q];
+  const double ix = 1.0 / p.dsx, iy = 1.0 / p.dsy, iz = 1.0 / p.dsz;
+  char *out = sc.ptr + (int64_t)lz * sc.planeStride + (int64_t)ly * sc.rowStride + (int64_t)lx * 8;
+  // j = grad(div A) - lap(A), expanded per component so each is a small
+  // independent expression (2 cross + 2 second derivatives) and the
+  // register live-set stays flat:
+  //   j.x = dxy Ay + dxz Az - dyy Ax - dzz Ax   (etc. cyclically)
+  {
+    const double jx = st.dxy(AAY, ix, iy) + st.dxz(AAZ, ix, iz) - st.dyy(AAX, iy * iy) -
+                      st.dzz(AAX, iz * iz);
+    __builtin_nontemporal_store(jx, (double *)out);
+  }
+  {
+    const double jy = st.dxy(AAX, ix, iy) + st.dyz(AAZ, iy, iz) - st.dxx(AAY, ix * ix) -
+                      st.dzz(AAY, iz * iz);
+    __builtin_nontemporal_store(jy, (double *)(out + sc.compStride));
+  }
+  {
+    const double jz = st.dxz(AAX, ix, iz) + st.dyz(AAY, iy, iz) - st.dxx(AAZ, ix * ix) -
+                      st.dyy(AAZ, iy * iy);
+    __builtin_nontemporal_store(jz, (double *)(out + 2 * sc.compStride));
+  }
+}
+
+// kernel 2b: momentum update (advection + pressure + viscosity + the
+// precomputed Lorentz force)
+__global__ void __launch_bounds__(256, 3) mhd_momentum_kernel(MhdParams p, MhdScratch sc) {
   const int32_t lx = blockIdx.x * blockDim.x + threadIdx.x;
   const int32_t ly = blockIdx.y * blockDim.y + threadIdx.y;
   const int32_t lz = blockIdx.z;
@@ -230,17 +274,15 @@ __global__ void __launch_bounds__(256) mhd_momentum_kernel(MhdParams p) {
   for (int q = 0; q < 8; ++q) st.base[q] = c.base[q];
   const double ix = 1.0 / p.dsx, iy = 1.0 / p.dsy, iz = 1.0 / p.dsz;
   const Vec3d uu = {st.c(UUX), st.c(UUY), st.c(UUZ)};
-  const Vec3d B = {st.dy(AAZ, iy) - st.dz(AAY, iz), st.dz(AAX, iz) - st.dx(AAZ, ix),
-                   st.dx(AAY, ix) - st.dy(AAX, iy)};
-  const Vec3d lap_a = {st.lap(AAX, ix, iy, iz), st.lap(AAY, ix, iy, iz),
-                       st.lap(AAZ, ix, iy, iz)};
-  const Vec3d graddiv_a = {
-      st.dxx(AAX, ix * ix) + st.dxy(AAY, ix, iy) + st.dxz(AAZ, ix, iz),
-      st.dxy(AAX, ix, iy) + st.dyy(AAY, iy * iy) + st.dyz(AAZ, iy, iz),
-      st.dxz(AAX, ix, iz) + st.dyz(AAY, iy, iz) + st.dzz(AAZ, iz * iz),
-  };
-  const Vec3d jxB = cross(graddiv_a - lap_a, B);
   const double rho_inv = exp(-st.c(LNRHO));
+  const char *jb =
+      sc.ptr + (int64_t)lz * sc.planeStride + (int64_t)ly * sc.rowStride + (int64_t)lx * 8;
+  const Vec3d B = {st.dy(AAZ, 1.0 / p.dsy) - st.dz(AAY, 1.0 / p.dsz),
+                   st.dz(AAX, 1.0 / p.dsz) - st.dx(AAZ, 1.0 / p.dsx),
+                   st.dx(AAY, 1.0 / p.dsx) - st.dy(AAX, 1.0 / p.dsy)};
+  const Vec3d j = {*(const double *)jb, *(const double *)(jb + sc.compStride),
+                   *(const double *)(jb + 2 * sc.compStride)};
+  const Vec3d jxB = cross(j, B);
 
   {
     const double ugradu = uu.x * st.dx(UUX, ix) + uu.y * st.dy(UUX, iy) + uu.z * st.dz(UUX, iz);
@@ -268,7 +310,7 @@ __global__ void __launch_bounds__(256) mhd_momentum_kernel(MhdParams p) {
 } // namespace
 
 void mhd_substep(ExchangeEngine &eng, int dom, const Rect3 &region, int step, double dt,
-                 const MhdCoeffs &cf) {
+                 const MhdCoeffs &cf, int64_t scratchBuf) {
   LocalDomain &d = eng.domain(dom);
   if (d.num_data() != 8 || d.elem_size(0) != 8)
     throw std::runtime_error("mhd_substep: domain must have 8 fp64 quantities");
@@ -308,9 +350,18 @@ void mhd_substep(ExchangeEngine &eng, int dom, const Rect3 &region, int step, do
   STENCIL_HIP(hipSetDevice(d.gpu()));
   dim3 block(64, 4, 1);
   dim3 grid((uint32_t)((ext.x + 63) / 64), (uint32_t)((ext.y + 3) / 4), (uint32_t)ext.z);
+  MhdScratch sc{};
+  sc.rowStride = ext.x * 8;
+  sc.planeStride = sc.rowStride * ext.y;
+  sc.compStride = sc.planeStride * ext.z;
+  if (eng.buffer_bytes(scratchBuf) < 3 * sc.compStride)
+    throw std::runtime_error("mhd_substep: scratch buffer too small for region");
+  sc.ptr = (char *)eng.buffer_ptr(scratchBuf);
   hipLaunchKernelGGL(mhd_scalar_kernel, grid, block, 0, eng.compute_stream(dom), p);
   STENCIL_HIP(hipGetLastError());
-  hipLaunchKernelGGL(mhd_momentum_kernel, grid, block, 0, eng.compute_stream(dom), p);
+  hipLaunchKernelGGL(mhd_lorentz_kernel, grid, block, 0, eng.compute_stream(dom), p, sc);
+  STENCIL_HIP(hipGetLastError());
+  hipLaunchKernelGGL(mhd_momentum_kernel, grid, block, 0, eng.compute_stream(dom), p, sc);
   STENCIL_HIP(hipGetLastError());
 }
 

@@ -99,6 +99,15 @@ class Astaroth:
         self.dd.realize()
         self.interiors = self.dd.get_interior()
         self.exteriors = self.dd.get_exterior()
+        # per-domain scratch for the Lorentz-force pass (3 fp64 fields over
+        # the largest region a substep computes = the full compute region)
+        self.scratch = []
+        eng = self.dd.backend.engine if hasattr(self.dd.backend, "engine") else None
+        if eng is not None:
+            for li in range(self.dd.num_local()):
+                lo, hi = self.dd.local_rect(li)
+                vol = (hi[0] - lo[0]) * (hi[1] - lo[1]) * (hi[2] - lo[2])
+                self.scratch.append(eng.create_buffer(li, 3 * vol * 8))
 
     def init_fields(self):
         """harmonic initial conditions on every interior (device-side)"""
@@ -130,7 +139,10 @@ class Astaroth:
         if compute and overlap:
             for li in range(dd.num_local()):
                 ilo, ihi = self.interiors[li]
-                _C.mhd_substep(eng, li, _C.Rect3(_C.Vec3(*ilo), _C.Vec3(*ihi)), s, dt, self.cf)
+                _C.mhd_substep(
+                    eng, li, _C.Rect3(_C.Vec3(*ilo), _C.Vec3(*ihi)), s, dt, self.cf,
+                    self.scratch[li],
+                )
         dd.exchange()
         if compute:
             for li in range(dd.num_local()):
@@ -140,7 +152,10 @@ class Astaroth:
                     else [dd.local_rect(li)]
                 )
                 for blo, bhi in boxes:
-                    _C.mhd_substep(eng, li, _C.Rect3(_C.Vec3(*blo), _C.Vec3(*bhi)), s, dt, self.cf)
+                    _C.mhd_substep(
+                        eng, li, _C.Rect3(_C.Vec3(*blo), _C.Vec3(*bhi)), s, dt, self.cf,
+                        self.scratch[li],
+                    )
         dd.backend.sync_compute()
         dd.swap()
 

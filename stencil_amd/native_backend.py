@@ -139,8 +139,8 @@ class NativeBackend:
                 export.append(
                     {
                         "gpu": d.gpu(),
-                        "pitch": [d.curr(qi).pitch for qi in range(nq)],
-                        "ysize": [d.curr(qi).ysize for qi in range(nq)],
+                        "pitch": [d.curr_pitch(qi) for qi in range(nq)],
+                        "ysize": [d.curr_ysize(qi) for qi in range(nq)],
                         "es": [d.elem_size(qi) for qi in range(nq)],
                         "curr": [d.ipc_handle(qi, False) for qi in range(nq)],
                         "next": [d.ipc_handle(qi, True) for qi in range(nq)],
