@@ -310,6 +310,33 @@ class DistributedDomain:
         ext = (arr.shape[2], arr.shape[1], arr.shape[0])
         self.backend.write_region(li, arr.tobytes(), pos, ext, handle.index, to_next)
 
+    # ---- checkpoint/restore ----
+    def save_checkpoint(self, path: str):
+        """save every quantity's interior of every local domain (npz; one
+        file per rank). Beyond the reference's capabilities (it only had
+        ParaView text dumps)."""
+        rank = self.comm.rank
+        arrays = {}
+        for li in range(self.num_local()):
+            lo, hi = self.local_rect(li)
+            gid = self.placement.linearize(self.placement.get_idx(rank, li))
+            for i, (es, name) in enumerate(self._data):
+                h = DataHandle(i, es, name)
+                arrays[f"d{gid}_q{i}"] = self.read_global(li, lo, hi, h)
+        np.savez_compressed(f"{path}.rank{rank}.npz", **arrays)
+
+    def load_checkpoint(self, path: str):
+        """restore interiors saved by save_checkpoint (same partition and
+        placement required)"""
+        rank = self.comm.rank
+        with np.load(f"{path}.rank{rank}.npz") as data:
+            for li in range(self.num_local()):
+                lo, hi = self.local_rect(li)
+                gid = self.placement.linearize(self.placement.get_idx(rank, li))
+                for i, (es, name) in enumerate(self._data):
+                    h = DataHandle(i, es, name)
+                    self.write_global(li, lo, data[f"d{gid}_q{i}"], h)
+
     # ---- observability ----
     def _write_plan_files(self, plan):
         """plan_<rank>.txt + rank x rank byte matrix (reference

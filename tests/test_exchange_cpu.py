@@ -115,3 +115,21 @@ def test_paraview_dump(tmp_path):
     lines = files[0].read_text().strip().splitlines()
     assert lines[0] == "Z,Y,X,temp"
     assert len(lines) == 1 + 6 * 5 * 4
+
+
+def test_checkpoint_roundtrip(tmp_path):
+    from util import ripple_block
+
+    dd = make_dd((8, 8, 8), 1, 2)
+    h = dd.add_data(np.float32, "q")
+    dd.realize()
+    fill_interiors(dd, h, scale=7.0)
+    dd.save_checkpoint(str(tmp_path / "ck"))
+
+    dd2 = make_dd((8, 8, 8), 1, 2)
+    h2 = dd2.add_data(np.float32, "q")
+    dd2.realize()
+    dd2.load_checkpoint(str(tmp_path / "ck"))
+    for li in range(dd2.num_local()):
+        lo, hi = dd2.local_rect(li)
+        assert np.array_equal(dd2.read_global(li, lo, hi, h2), ripple_block(lo, hi, dd2.size, 7.0))
