@@ -23,6 +23,8 @@
 #include <hip/hip_runtime.h>
 
 #include <algorithm>
+#include <cstdio>
+#include <cstdlib>
 #include <stdexcept>
 
 #include "stencil_amd/domain.hpp"
@@ -188,7 +190,7 @@ __device__ __forceinline__ void write_rk3(const MhdParams &p, const Stencil &st,
 __global__ void __launch_bounds__(256) mhd_scalar_kernel(MhdParams p) {
   const int32_t lx = blockIdx.x * blockDim.x + threadIdx.x;
   const int32_t ly = blockIdx.y * blockDim.y + threadIdx.y;
-  const int32_t lz = blockIdx.z;
+  const int32_t lz = blockIdx.z * blockDim.z + threadIdx.z;
   if (lx >= p.extX || ly >= p.extY || lz >= p.extZ) return;
   const MhdCommon c = mhd_setup(p, lx, ly, lz);
   Stencil st;
@@ -228,7 +230,7 @@ struct MhdScratch {
 __global__ void __launch_bounds__(256) mhd_lorentz_kernel(MhdParams p, MhdScratch sc) {
   const int32_t lx = blockIdx.x * blockDim.x + threadIdx.x;
   const int32_t ly = blockIdx.y * blockDim.y + threadIdx.y;
-  const int32_t lz = blockIdx.z;
+  const int32_t lz = blockIdx.z * blockDim.z + threadIdx.z;
   if (lx >= p.extX || ly >= p.extY || lz >= p.extZ) return;
   const MhdCommon c = mhd_setup(p, lx, ly, lz);
   Stencil st;
@@ -264,7 +266,7 @@ __global__ void __launch_bounds__(256) mhd_lorentz_kernel(MhdParams p, MhdScratc
 __global__ void __launch_bounds__(256, 3) mhd_momentum_kernel(MhdParams p, MhdScratch sc) {
   const int32_t lx = blockIdx.x * blockDim.x + threadIdx.x;
   const int32_t ly = blockIdx.y * blockDim.y + threadIdx.y;
-  const int32_t lz = blockIdx.z;
+  const int32_t lz = blockIdx.z * blockDim.z + threadIdx.z;
   if (lx >= p.extX || ly >= p.extY || lz >= p.extZ) return;
   const MhdCommon c = mhd_setup(p, lx, ly, lz);
   Stencil st;
@@ -348,8 +350,21 @@ void mhd_substep(ExchangeEngine &eng, int dom, const Rect3 &region, int step, do
   p.alpha_over_beta_prev = (step == 0) ? 0.0 : ALPHA[step] / BETA[step - 1];
   p.beta = BETA[step];
   STENCIL_HIP(hipSetDevice(d.gpu()));
-  dim3 block(64, 4, 1);
-  dim3 grid((uint32_t)((ext.x + 63) / 64), (uint32_t)((ext.y + 3) / 4), (uint32_t)ext.z);
+  static int bx = 0, by = 0, bz = 0;
+  if (!bx) {
+    bx = 64;
+    by = 4;
+    bz = 1;
+    if (const char *e = getenv("STENCIL_MHD_BLOCK"))
+      if (sscanf(e, "%dx%dx%d", &bx, &by, &bz) != 3 || bx * by * bz != 256) {
+        bx = 64;
+        by = 4;
+        bz = 1;
+      }
+  }
+  dim3 block((uint32_t)bx, (uint32_t)by, (uint32_t)bz);
+  dim3 grid((uint32_t)((ext.x + bx - 1) / bx), (uint32_t)((ext.y + by - 1) / by),
+            (uint32_t)((ext.z + bz - 1) / bz));
   MhdScratch sc{};
   sc.rowStride = ext.x * 8;
   sc.planeStride = sc.rowStride * ext.y;
