@@ -111,8 +111,9 @@ public:
   void sync_packs(); // also used after unpack
   void sync_all();
 
-  //// compute-stream helpers for apps
-  hipStream_t compute_stream(int dom);
+  //// compute-stream helpers for apps (two streams per domain so
+  //// exterior-shell kernels can overlap the interior kernel)
+  hipStream_t compute_stream(int dom, int which = 0);
   uintptr_t compute_stream_handle(int dom) { return (uintptr_t)compute_stream(dom); }
   void sync_compute();
   // make dom's compute stream wait until all exchange streams are idle
@@ -164,7 +165,8 @@ private:
 
   std::map<int, hipStream_t> commStreams_; // per device: translate launches
   std::map<int, hipStream_t> packStreams_; // per device: pack/unpack launches
-  std::vector<hipStream_t> computeStreams_; // per domain
+  std::vector<hipStream_t> computeStreams_;  // per domain, stream 0
+  std::vector<hipStream_t> computeStreams2_; // per domain, stream 1
 
   std::vector<CopyBatch> translateBatches_; // one per device with jobs
   std::vector<CopyBatch> packBatches_;

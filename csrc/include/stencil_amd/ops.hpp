@@ -10,7 +10,7 @@ namespace stencil_amd {
 // next = avg of 6 face neighbors of curr, with the reference's hot/cold
 // sphere sources fixed inside `computeRegion` (bin/jacobi3d.cu:40-85).
 void jacobi_step(ExchangeEngine &eng, int dom, int64_t qi, const Rect3 &region,
-                 const Rect3 &computeRegion);
+                 const Rect3 &computeRegion, int streamId = 0);
 
 // fill an fp32 region with `value` (curr or next buffer)
 void fill_f32(ExchangeEngine &eng, int dom, int64_t qi, const Rect3 &region, float value,
@@ -30,7 +30,7 @@ struct MhdCoeffs {
 // scratchBuf: engine buffer (>= 3 * region volume * 8 bytes) holding the
 // intermediate Lorentz force between the kernel passes
 void mhd_substep(ExchangeEngine &eng, int dom, const Rect3 &region, int step, double dt,
-                 const MhdCoeffs &cf, int64_t scratchBuf);
+                 const MhdCoeffs &cf, int64_t scratchBuf, int streamId = 0);
 
 // fill an fp64 region with base + amp*sin(kx*x + ky*y + kz*z + phase)
 // (deterministic smooth initial conditions, reproducible in NumPy)

@@ -245,7 +245,8 @@ PYBIND11_MODULE(_C, m) {
             hipMemcpy((void *)e.buffer_ptr(buf), s.data(), s.size(), hipMemcpyHostToDevice));
       });
 
-  m.def("jacobi_step", &jacobi_step);
+  m.def("jacobi_step", &jacobi_step, py::arg("eng"), py::arg("dom"), py::arg("qi"),
+        py::arg("region"), py::arg("compute_region"), py::arg("stream_id") = 0);
   m.def("fill_f32", &fill_f32);
 
   py::class_<MhdCoeffs>(m, "MhdCoeffs")
@@ -258,7 +259,8 @@ PYBIND11_MODULE(_C, m) {
       .def_readwrite("nu", &MhdCoeffs::nu)
       .def_readwrite("eta", &MhdCoeffs::eta)
       .def_readwrite("chi", &MhdCoeffs::chi);
-  m.def("mhd_substep", &mhd_substep);
+  m.def("mhd_substep", &mhd_substep, py::arg("eng"), py::arg("dom"), py::arg("region"),
+        py::arg("step"), py::arg("dt"), py::arg("cf"), py::arg("scratch"), py::arg("stream_id") = 0);
   m.def("init_harmonic_f64", &init_harmonic_f64);
 
   // topology utilities (csrc/src/topo.hip)

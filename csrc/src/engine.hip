@@ -142,6 +142,7 @@ void CopyBatch::destroy() {
 ExchangeEngine::ExchangeEngine(std::vector<std::shared_ptr<LocalDomain>> domains)
     : domains_(std::move(domains)) {
   computeStreams_.resize(domains_.size(), nullptr);
+  computeStreams2_.resize(domains_.size(), nullptr);
 }
 
 ExchangeEngine::~ExchangeEngine() {
@@ -151,6 +152,8 @@ ExchangeEngine::~ExchangeEngine() {
   for (auto &kv : commStreams_) (void)hipStreamDestroy(kv.second);
   for (auto &kv : packStreams_) (void)hipStreamDestroy(kv.second);
   for (auto s : computeStreams_)
+    if (s) (void)hipStreamDestroy(s);
+  for (auto s : computeStreams2_)
     if (s) (void)hipStreamDestroy(s);
   for (auto &b : buffers_)
     if (b.ptr) (void)hipFree(b.ptr);
@@ -381,14 +384,15 @@ hipStream_t ExchangeEngine::pack_stream_(int dev) {
   return s;
 }
 
-hipStream_t ExchangeEngine::compute_stream(int dom) {
-  if (!computeStreams_[dom]) {
+hipStream_t ExchangeEngine::compute_stream(int dom, int which) {
+  auto &vec = which ? computeStreams2_ : computeStreams_;
+  if (!vec[dom]) {
     STENCIL_HIP(hipSetDevice(domains_[dom]->gpu()));
     hipStream_t s;
     STENCIL_HIP(hipStreamCreateWithFlags(&s, hipStreamNonBlocking));
-    computeStreams_[dom] = s;
+    vec[dom] = s;
   }
-  return computeStreams_[dom];
+  return vec[dom];
 }
 
 void ExchangeEngine::launch_translates() {
@@ -419,11 +423,12 @@ void ExchangeEngine::sync_all() {
 }
 
 void ExchangeEngine::sync_compute() {
-  for (size_t i = 0; i < computeStreams_.size(); ++i)
-    if (computeStreams_[i]) {
-      STENCIL_HIP(hipSetDevice(domains_[i]->gpu()));
-      STENCIL_HIP(hipStreamSynchronize(computeStreams_[i]));
-    }
+  for (auto *vec : {&computeStreams_, &computeStreams2_})
+    for (size_t i = 0; i < vec->size(); ++i)
+      if ((*vec)[i]) {
+        STENCIL_HIP(hipSetDevice(domains_[i]->gpu()));
+        STENCIL_HIP(hipStreamSynchronize((*vec)[i]));
+      }
 }
 
 } // namespace stencil_amd

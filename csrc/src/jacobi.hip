@@ -225,7 +225,7 @@ uint32_t grid_for(int64_t total, int block) {
 } // namespace
 
 void jacobi_step(ExchangeEngine &eng, int dom, int64_t qi, const Rect3 &region,
-                 const Rect3 &computeRegion) {
+                 const Rect3 &computeRegion, int streamId) {
   LocalDomain &d = eng.domain(dom);
   if (d.elem_size(qi) != 4) throw std::runtime_error("jacobi_step: quantity must be fp32");
   const Vec3 ext = region.extent();
@@ -271,10 +271,10 @@ void jacobi_step(ExchangeEngine &eng, int dom, int64_t qi, const Rect3 &region,
     dim3 block((uint32_t)bx, (uint32_t)by, 1);
     dim3 grid((uint32_t)((units + bx - 1) / bx), (uint32_t)((ext.y + by - 1) / by),
               (uint32_t)((ext.z + 15) / 16)); // 16 == JAC_ZCHUNK
-    hipLaunchKernelGGL(jacobi_kernel_v4, grid, block, 0, eng.compute_stream(dom), p);
+    hipLaunchKernelGGL(jacobi_kernel_v4, grid, block, 0, eng.compute_stream(dom, streamId), p);
   } else {
     hipLaunchKernelGGL(jacobi_kernel, dim3(grid_for(ext.flatten(), 256)), dim3(256), 0,
-                       eng.compute_stream(dom), p);
+                       eng.compute_stream(dom, streamId), p);
   }
   STENCIL_HIP(hipGetLastError());
 }

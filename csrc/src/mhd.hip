@@ -312,7 +312,7 @@ __global__ void __launch_bounds__(256, 3) mhd_momentum_kernel(MhdParams p, MhdSc
 } // namespace
 
 void mhd_substep(ExchangeEngine &eng, int dom, const Rect3 &region, int step, double dt,
-                 const MhdCoeffs &cf, int64_t scratchBuf) {
+                 const MhdCoeffs &cf, int64_t scratchBuf, int streamId) {
   LocalDomain &d = eng.domain(dom);
   if (d.num_data() != 8 || d.elem_size(0) != 8)
     throw std::runtime_error("mhd_substep: domain must have 8 fp64 quantities");
@@ -350,16 +350,18 @@ void mhd_substep(ExchangeEngine &eng, int dom, const Rect3 &region, int step, do
   p.alpha_over_beta_prev = (step == 0) ? 0.0 : ALPHA[step] / BETA[step - 1];
   p.beta = BETA[step];
   STENCIL_HIP(hipSetDevice(d.gpu()));
+  // 32x4x2 measured best on gfx950 (block sweep in gpurun_out/gpu6.log:
+  // 3D-ish tiles reuse the y/z derivative lines within a block)
   static int bx = 0, by = 0, bz = 0;
   if (!bx) {
-    bx = 64;
+    bx = 32;
     by = 4;
-    bz = 1;
+    bz = 2;
     if (const char *e = getenv("STENCIL_MHD_BLOCK"))
       if (sscanf(e, "%dx%dx%d", &bx, &by, &bz) != 3 || bx * by * bz != 256) {
-        bx = 64;
+        bx = 32;
         by = 4;
-        bz = 1;
+        bz = 2;
       }
   }
   dim3 block((uint32_t)bx, (uint32_t)by, (uint32_t)bz);
@@ -372,11 +374,11 @@ void mhd_substep(ExchangeEngine &eng, int dom, const Rect3 &region, int step, do
   if (eng.buffer_bytes(scratchBuf) < 3 * sc.compStride)
     throw std::runtime_error("mhd_substep: scratch buffer too small for region");
   sc.ptr = (char *)eng.buffer_ptr(scratchBuf);
-  hipLaunchKernelGGL(mhd_scalar_kernel, grid, block, 0, eng.compute_stream(dom), p);
+  hipLaunchKernelGGL(mhd_scalar_kernel, grid, block, 0, eng.compute_stream(dom, streamId), p);
   STENCIL_HIP(hipGetLastError());
-  hipLaunchKernelGGL(mhd_lorentz_kernel, grid, block, 0, eng.compute_stream(dom), p, sc);
+  hipLaunchKernelGGL(mhd_lorentz_kernel, grid, block, 0, eng.compute_stream(dom, streamId), p, sc);
   STENCIL_HIP(hipGetLastError());
-  hipLaunchKernelGGL(mhd_momentum_kernel, grid, block, 0, eng.compute_stream(dom), p, sc);
+  hipLaunchKernelGGL(mhd_momentum_kernel, grid, block, 0, eng.compute_stream(dom, streamId), p, sc);
   STENCIL_HIP(hipGetLastError());
 }
 

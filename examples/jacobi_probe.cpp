@@ -112,6 +112,69 @@ __global__ void __launch_bounds__(256) probe8(P p) {
   }
 }
 
+// variant mirroring the PRODUCTION kernel structure: slot indirection for
+// the base pointers, per-z sphere bounding test, head/tail scalar units
+__global__ void __launch_bounds__(256) probe_prod(P p, const char *const *srcSlot,
+                                                  char *const *dstSlot, int32_t withSphere) {
+  const int32_t u = blockIdx.x * blockDim.x + threadIdx.x;
+  const int32_t y = blockIdx.y * blockDim.y + threadIdx.y;
+  const int32_t z0 = blockIdx.z * ZCH;
+  if (y >= p.ny) return;
+  const char *srcBase = *srcSlot;
+  char *dstBase = *dstSlot;
+  const int32_t zEnd = min(z0 + ZCH, p.nz);
+  const int32_t body4 = p.nx4;
+  if (u < body4) {
+    const char *col = srcBase + (int64_t)(z0 + 3) * p.plane + (int64_t)(y + 3) * p.pitch + 16 + u * 16;
+    char *dcol = dstBase + (int64_t)(z0 + 3) * p.plane + (int64_t)(y + 3) * p.pitch + 16 + u * 16;
+    float4 cm = *(const float4 *)(col - p.plane);
+    float4 cc = *(const float4 *)(col);
+    for (int32_t z = z0; z < zEnd; ++z) {
+      const float4 cp = *(const float4 *)(col + p.plane);
+      const float left = *(const float *)(col - 4);
+      const float right = *(const float *)(col + 16);
+      const float4 py = *(const float4 *)(col + p.pitch);
+      const float4 my = *(const float4 *)(col - p.pitch);
+      float4 out;
+      out.x = (cc.y + left + py.x + my.x + cp.x + cm.x) / 6.0f;
+      out.y = (cc.z + cc.x + py.y + my.y + cp.y + cm.y) / 6.0f;
+      out.z = (cc.w + cc.y + py.z + my.z + cp.z + cm.z) / 6.0f;
+      out.w = (right + cc.z + py.w + my.w + cp.w + cm.w) / 6.0f;
+      if (withSphere) {
+        const int32_t dy = y - p.ny / 2, dz = z - p.nz / 2;
+        const int32_t yz2 = dy * dy + dz * dz;
+        const int32_t srad = p.nx4 * 4 / 10;
+        if (yz2 < (srad + 1) * (srad + 1)) {
+#pragma unroll
+          for (int i = 0; i < 4; ++i) {
+            const int32_t dxh = u * 4 + i - p.nx4;
+            if ((int32_t)__fsqrt_rn((float)(dxh * dxh + yz2)) <= srad) (&out.x)[i] = 1.0f;
+          }
+        }
+      }
+      vfloat4 ov = {out.x, out.y, out.z, out.w};
+      __builtin_nontemporal_store(ov, (vfloat4 *)dcol);
+      cm = cc;
+      cc = cp;
+      col += p.plane;
+      dcol += p.plane;
+    }
+  } else if (u == body4) { // head/tail scalar cells (mirrors production)
+    for (int32_t z = z0; z < zEnd; ++z) {
+      const char *rowC = srcBase + (int64_t)(z + 3) * p.plane + (int64_t)(y + 3) * p.pitch;
+      char *rowD = dstBase + (int64_t)(z + 3) * p.plane + (int64_t)(y + 3) * p.pitch;
+      for (int32_t lx = 0; lx < 2; ++lx) {
+        const int64_t ax = 3 + lx;
+        const float v = (*(const float *)(rowC + (ax + 1) * 4) + *(const float *)(rowC + (ax - 1) * 4) +
+                         *(const float *)(rowC + p.pitch + ax * 4) + *(const float *)(rowC - p.pitch + ax * 4) +
+                         *(const float *)(rowC + p.plane + ax * 4) + *(const float *)(rowC - p.plane + ax * 4)) /
+                        6.0f;
+        *(float *)(rowD + ax * 4) = v;
+      }
+    }
+  }
+}
+
 int main(int argc, char **argv) {
   const int n = argc > 1 ? atoi(argv[1]) : 752;
   const int rounds = argc > 2 ? atoi(argv[2]) : 10;
@@ -136,16 +199,32 @@ int main(int argc, char **argv) {
   CHECK(hipEventCreate(&e0));
   CHECK(hipEventCreate(&e1));
   const double cells = (double)p.nx4 * 4 * p.ny * p.nz;
-  const char *names[4] = {"copy", "xz", "full", "full8"};
-  double best[4] = {1e30, 1e30, 1e30, 1e30};
+  // device slots for the production-mirror variants
+  char **slots;
+  CHECK(hipMalloc(&slots, 2 * sizeof(char *)));
+  CHECK(hipMemcpy(slots, &a, sizeof(char *), hipMemcpyHostToDevice));
+  CHECK(hipMemcpy(slots + 1, &b, sizeof(char *), hipMemcpyHostToDevice));
+  dim3 grdp((p.nx4 + 2 + 63) / 64, (p.ny + 3) / 4, (p.nz + ZCH - 1) / ZCH);
+
+  const int NV = 6;
+  const char *names[NV] = {"copy", "xz", "full", "full8", "prod", "prod+sph"};
+  double best[NV] = {1e30, 1e30, 1e30, 1e30, 1e30, 1e30};
   for (int r = 0; r < rounds; ++r) {
-    for (int v = 0; v < 4; ++v) {
+    for (int v = 0; v < NV; ++v) {
       CHECK(hipEventRecord(e0));
       switch (v) {
       case 0: hipLaunchKernelGGL(probe<0>, grd, blk, 0, 0, p); break;
       case 1: hipLaunchKernelGGL(probe<1>, grd, blk, 0, 0, p); break;
       case 2: hipLaunchKernelGGL(probe<2>, grd, blk, 0, 0, p); break;
       case 3: hipLaunchKernelGGL(probe8, grd8, blk, 0, 0, p); break;
+      case 4:
+        hipLaunchKernelGGL(probe_prod, grdp, blk, 0, 0, p, (const char *const *)slots,
+                           (char *const *)(slots + 1), 0);
+        break;
+      case 5:
+        hipLaunchKernelGGL(probe_prod, grdp, blk, 0, 0, p, (const char *const *)slots,
+                           (char *const *)(slots + 1), 1);
+        break;
       }
       CHECK(hipEventRecord(e1));
       CHECK(hipEventSynchronize(e1));
@@ -154,7 +233,7 @@ int main(int argc, char **argv) {
       best[v] = std::min(best[v], (double)ms);
     }
   }
-  for (int v = 0; v < 4; ++v)
+  for (int v = 0; v < NV; ++v)
     printf("%-6s %8.3f ms  %7.1f Gcell/s  %6.2f TB/s(8B/cell)\n", names[v], best[v],
            cells / best[v] / 1e6, cells * 8 / best[v] / 1e9);
   return 0;

@@ -222,6 +222,21 @@ class DistributedDomain:
         self.backend.exchange()
         self.time_exchange += time.perf_counter() - t0
 
+    def exchange_begin(self):
+        """asynchronous exchange start (native backend): all device work is
+        enqueued; call exchange_end() before reading halos"""
+        if hasattr(self.backend, "exchange_begin"):
+            self.backend.exchange_begin()
+        # torch backend has no async path; everything happens in end()
+
+    def exchange_end(self):
+        t0 = time.perf_counter()
+        if hasattr(self.backend, "exchange_end"):
+            self.backend.exchange_end()
+        else:
+            self.backend.exchange()
+        self.time_exchange += time.perf_counter() - t0
+
     def swap(self):
         t0 = time.perf_counter()
         self.backend.swap()

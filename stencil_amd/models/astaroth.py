@@ -99,15 +99,18 @@ class Astaroth:
         self.dd.realize()
         self.interiors = self.dd.get_interior()
         self.exteriors = self.dd.get_exterior()
-        # per-domain scratch for the Lorentz-force pass (3 fp64 fields over
-        # the largest region a substep computes = the full compute region)
+        # per-domain Lorentz-force scratch, one per compute stream
+        # (interior runs on stream 0, exterior shells on stream 1 so they
+        # overlap; each stream needs its own scratch)
         self.scratch = []
+        self.scratch_ext = []
         eng = self.dd.backend.engine if hasattr(self.dd.backend, "engine") else None
         if eng is not None:
             for li in range(self.dd.num_local()):
                 lo, hi = self.dd.local_rect(li)
                 vol = (hi[0] - lo[0]) * (hi[1] - lo[1]) * (hi[2] - lo[2])
                 self.scratch.append(eng.create_buffer(li, 3 * vol * 8))
+                self.scratch_ext.append(eng.create_buffer(li, 3 * vol * 8))
 
     def init_fields(self):
         """harmonic initial conditions on every interior (device-side)"""
@@ -152,9 +155,10 @@ class Astaroth:
                     else [dd.local_rect(li)]
                 )
                 for blo, bhi in boxes:
+                    # stream 1: exterior shells overlap the interior kernel
                     _C.mhd_substep(
                         eng, li, _C.Rect3(_C.Vec3(*blo), _C.Vec3(*bhi)), s, dt, self.cf,
-                        self.scratch[li],
+                        self.scratch_ext[li], 1 if overlap else 0,
                     )
         dd.backend.sync_compute()
         dd.swap()

@@ -51,10 +51,13 @@ class Jacobi3D:
                 ilo, ihi = self.interiors[li]
                 dd.backend.jacobi_step(li, self.h.index, ilo, ihi, self.compute_lo, self.compute_hi)
             dd.exchange()
+            # exterior shells go on the second compute stream: they only
+            # depend on the exchange (synced above), so they overlap the
+            # still-running interior kernel
             for li in range(dd.num_local()):
                 for blo, bhi in self.exteriors[li]:
                     dd.backend.jacobi_step(
-                        li, self.h.index, blo, bhi, self.compute_lo, self.compute_hi
+                        li, self.h.index, blo, bhi, self.compute_lo, self.compute_hi, stream_id=1
                     )
         else:
             dd.exchange()

@@ -222,12 +222,20 @@ class NativeBackend:
         return "nccl" not in backend
 
     # ---- per-iteration ----
-    def exchange(self):
+    def exchange_begin(self):
+        """stream-ordered first half: launch all local/IPC translates and
+        (if cross-rank) the pack kernels. Returns immediately; GPU work
+        overlaps whatever the app runs on its compute streams."""
         self.engine.launch_translates()
+        if self._has_wire:
+            self.engine.launch_packs()
+
+    def exchange_end(self):
+        """second half: move wire buffers, unpack, and block until every
+        halo is in place."""
         if self._has_wire:
             import torch.distributed as dist
 
-            self.engine.launch_packs()
             self.engine.sync_packs()
             if self._wire_via_cpu:
                 ops = []
@@ -251,11 +259,13 @@ class NativeBackend:
             self.engine.launch_unpacks()
         self.engine.sync_all()
         if self._ipc_active:
-            # all colocated ranks' direct writes are complete after the
-            # barrier (each rank synced its own translate kernels above)
             import torch.distributed as dist
 
             dist.barrier(group=self._colo_group)
+
+    def exchange(self):
+        self.exchange_begin()
+        self.exchange_end()
 
     def swap(self):
         for d in self.domains:
@@ -278,8 +288,9 @@ class NativeBackend:
         _C.fill_f32(self.engine, li, qi, _rect3(region_lo, region_hi), value, next_buf)
 
     def jacobi_step(self, li: int, qi: int, region_lo: Vec, region_hi: Vec,
-                    c_lo: Vec, c_hi: Vec):
-        _C.jacobi_step(self.engine, li, qi, _rect3(region_lo, region_hi), _rect3(c_lo, c_hi))
+                    c_lo: Vec, c_hi: Vec, stream_id: int = 0):
+        _C.jacobi_step(self.engine, li, qi, _rect3(region_lo, region_hi), _rect3(c_lo, c_hi),
+                       stream_id)
 
     def sync_compute(self):
         self.engine.sync_compute()

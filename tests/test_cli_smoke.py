@@ -1,0 +1,66 @@
+"""Benchmark CLI smoke tests (torch backend, tiny sizes, CPU): the
+benchmark suite must stay runnable."""
+import os
+import subprocess
+import sys
+
+import pytest
+
+REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+
+
+def run_cli(script, *args, timeout=300):
+    return subprocess.run(
+        [sys.executable, os.path.join(REPO, "benchmarks", script), *args],
+        capture_output=True,
+        text=True,
+        timeout=timeout,
+        cwd=REPO,
+    )
+
+
+def test_jacobi3d_cli():
+    out = run_cli(
+        "jacobi3d.py", "--backend", "torch", "--size", "24", "--iters", "2", "--gpus", "2"
+    )
+    assert out.returncode == 0, out.stderr
+    assert out.stdout.startswith("jacobi3d,weak")
+
+
+def test_jacobi3d_strong_no_overlap():
+    out = run_cli(
+        "jacobi3d.py",
+        "--backend",
+        "torch",
+        "--size",
+        "24",
+        "--iters",
+        "2",
+        "--strong",
+        "--no-overlap",
+        "--trivial",
+    )
+    assert out.returncode == 0, out.stderr
+    assert "strong" in out.stdout
+
+
+def test_bench_exchange_cli():
+    out = run_cli("bench_exchange.py", "--backend", "torch", "--size", "16", "--iters", "3")
+    assert out.returncode == 0, out.stderr
+    lines = out.stdout.strip().splitlines()
+    assert lines[0].startswith("pattern,")
+    assert len(lines) == 6  # header + 5 patterns
+
+
+def test_exchange_scaling_cli():
+    out = run_cli(
+        "exchange_scaling.py", "--backend", "torch", "--size", "16", "--iters", "3"
+    )
+    assert out.returncode == 0, out.stderr
+    assert out.stdout.startswith("exchange,weak")
+
+
+def test_bench_qap_cli():
+    out = run_cli("bench_qap.py", "--n", "6")
+    assert out.returncode == 0, out.stderr
+    assert out.stdout.startswith("case,")
