@@ -3,6 +3,8 @@
 #include "stencil_amd/engine.hpp"
 #include "stencil_amd/hip_check.hpp"
 
+#include <roctracer/roctx.h>
+
 #include <algorithm>
 #include <cstdlib>
 #include <cstring>
@@ -395,21 +397,31 @@ hipStream_t ExchangeEngine::compute_stream(int dom, int which) {
   return vec[dom];
 }
 
+// roctx ranges give rocprof-sys/rocprofv3 timelines the same phase
+// annotations the reference had via NVTX (tracing)
 void ExchangeEngine::launch_translates() {
+  roctxRangePush("stencil::translate");
   for (auto &b : translateBatches_) b.launch(comm_stream_(b.dev));
+  roctxRangePop();
 }
 void ExchangeEngine::launch_packs() {
+  roctxRangePush("stencil::pack");
   for (auto &b : packBatches_) b.launch(pack_stream_(b.dev));
+  roctxRangePop();
 }
 void ExchangeEngine::launch_unpacks() {
+  roctxRangePush("stencil::unpack");
   for (auto &b : unpackBatches_) b.launch(pack_stream_(b.dev));
+  roctxRangePop();
 }
 
 void ExchangeEngine::sync_translates() {
+  roctxRangePush("stencil::sync_translates");
   for (auto &b : translateBatches_) {
     STENCIL_HIP(hipSetDevice(b.dev));
     STENCIL_HIP(hipStreamSynchronize(comm_stream_(b.dev)));
   }
+  roctxRangePop();
 }
 void ExchangeEngine::sync_packs() {
   for (auto &kv : packStreams_) {

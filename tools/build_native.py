@@ -63,7 +63,10 @@ def build(verbose=True):
     suffix = sysconfig.get_config_var("EXT_SUFFIX")
     out = REPO / "stencil_amd" / f"_C{suffix}"
     if not out.exists() or any(o.stat().st_mtime > out.stat().st_mtime for o in objs):
-        cmd = [HIPCC, "-shared", "-fPIC", "-o", str(out)] + [str(o) for o in objs]
+        cmd = [HIPCC, "-shared", "-fPIC", "-o", str(out)] + [str(o) for o in objs] + [
+            "-L/opt/rocm/lib",
+            "-lroctx64",
+        ]
         if verbose:
             print("[link]", " ".join(cmd), flush=True)
         subprocess.check_call(cmd)
@@ -81,7 +84,12 @@ def build(verbose=True):
         if verbose:
             print("[exe-compile]", " ".join(cmd), flush=True)
         subprocess.check_call(cmd)
-        cmd = [HIPCC, str(exo)] + [str(o) for o in core_objs] + ["-o", str(exe)]
+        cmd = [HIPCC, str(exo)] + [str(o) for o in core_objs] + [
+            "-L/opt/rocm/lib",
+            "-lroctx64",
+            "-o",
+            str(exe),
+        ]
         if verbose:
             print("[exe-link]", " ".join(cmd), flush=True)
         subprocess.check_call(cmd)
