@@ -1,5 +1,6 @@
 // ExchangeEngine implementation: job-table construction and the batched
 // copy kernel. See engine.hpp for the design rationale.
+#include "stencil_amd/device_util.hpp"
 #include "stencil_amd/engine.hpp"
 #include "stencil_amd/hip_check.hpp"
 
@@ -36,8 +37,8 @@ __global__ void copy_batch_kernel(const CopyJob *__restrict__ jobs, const int64_
     }
   }
   const CopyJob j = jobs[lo];
-  const char *src = (j.srcSlot ? *j.srcSlot : j.srcDirect) + j.srcOff;
-  char *dst = (j.dstSlot ? *j.dstSlot : j.dstDirect) + j.dstOff;
+  const char *src = uniform_ptr((j.srcSlot ? *j.srcSlot : j.srcDirect) + j.srcOff);
+  char *dst = uniform_ptr((j.dstSlot ? *j.dstSlot : j.dstDirect) + j.dstOff);
 
   const int64_t jobBlocks = prefix[lo + 1] - prefix[lo];
   const int64_t stride = jobBlocks * blockDim.x;

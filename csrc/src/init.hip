@@ -3,6 +3,7 @@
 // give deterministic smooth initial conditions reproducible in NumPy).
 #include <hip/hip_runtime.h>
 
+#include "stencil_amd/device_util.hpp"
 #include "stencil_amd/domain.hpp"
 #include "stencil_amd/engine.hpp"
 #include "stencil_amd/hip_check.hpp"
@@ -30,7 +31,7 @@ __global__ void __launch_bounds__(256) init_harmonic_f64_kernel(InitParams p) {
   if (lx >= p.extX || ly >= p.extY || lz >= p.extZ) return;
   const int64_t gx = p.loX + lx, gy = p.loY + ly, gz = p.loZ + lz;
   const double v = p.base + p.amp * sin(p.kx * gx + p.ky * gy + p.kz * gz + p.phase);
-  char *base = *p.slot;
+  char *base = uniform_ptr(*p.slot);
   *(double *)(base + (gz - p.allocZ) * p.plane + (gy - p.allocY) * p.pitch + (gx - p.allocX) * 8) =
       v;
 }

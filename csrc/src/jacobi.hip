@@ -17,6 +17,7 @@
 #include <cstdio>
 #include <cstdlib>
 
+#include "stencil_amd/device_util.hpp"
 #include "stencil_amd/domain.hpp"
 #include "stencil_amd/engine.hpp"
 #include "stencil_amd/hip_check.hpp"
@@ -81,8 +82,8 @@ __global__ void __launch_bounds__(256) jacobi_kernel_v4(JacobiParams p) {
   const int32_t ly = blockIdx.y * blockDim.y + threadIdx.y;
   const int32_t lz0 = blockIdx.z * JAC_ZCHUNK;
   if (ly >= p.extY) return;
-  const char *srcBase = *p.srcSlot;
-  char *dstBase = *p.dstSlot;
+  const char *srcBase = uniform_ptr(*p.srcSlot);
+  char *dstBase = uniform_ptr(*p.dstSlot);
   const int64_t gy = p.loY + ly;
   const int64_t ay = gy - p.allocY;
   const int32_t zEnd = min((int32_t)(lz0 + JAC_ZCHUNK), p.extZ);
@@ -173,8 +174,8 @@ __global__ void __launch_bounds__(256) jacobi_kernel_v4(JacobiParams p) {
 }
 
 __global__ void jacobi_kernel(JacobiParams p) {
-  const char *srcBase = *p.srcSlot;
-  char *dstBase = *p.dstSlot;
+  const char *srcBase = uniform_ptr(*p.srcSlot);
+  char *dstBase = uniform_ptr(*p.dstSlot);
   const int64_t total = (int64_t)p.extX * p.extY * p.extZ;
   const int64_t stride = (int64_t)gridDim.x * blockDim.x;
   for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < total; i += stride) {
@@ -203,7 +204,7 @@ __global__ void jacobi_kernel(JacobiParams p) {
 
 __global__ void fill_kernel(char *const *dstSlot, int64_t pitch, int64_t plane, int64_t offBytes,
                             int32_t extX, int32_t extY, int32_t extZ, float value) {
-  char *base = *dstSlot + offBytes;
+  char *base = uniform_ptr(*dstSlot) + offBytes;
   const int64_t total = (int64_t)extX * extY * extZ;
   const int64_t stride = (int64_t)gridDim.x * blockDim.x;
   for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < total; i += stride) {

@@ -27,6 +27,7 @@
 #include <cstdlib>
 #include <stdexcept>
 
+#include "stencil_amd/device_util.hpp"
 #include "stencil_amd/domain.hpp"
 #include "stencil_amd/engine.hpp"
 #include "stencil_amd/hip_check.hpp"
@@ -172,8 +173,8 @@ __device__ __forceinline__ MhdCommon mhd_setup(const MhdParams &p, int32_t lx, i
   c.plane = p.plane;
 #pragma unroll
   for (int q = 0; q < 8; ++q) {
-    c.base[q] = p.currSlots[q] + cellOff;
-    c.out[q] = p.nextSlots[q] + cellOff;
+    c.base[q] = uniform_ptr(p.currSlots[q]) + cellOff;
+    c.out[q] = uniform_ptr(p.nextSlots[q]) + cellOff;
   }
   return c;
 }
