@@ -133,3 +133,25 @@ def test_checkpoint_roundtrip(tmp_path):
     for li in range(dd2.num_local()):
         lo, hi = dd2.local_rect(li)
         assert np.array_equal(dd2.read_global(li, lo, hi, h2), ripple_block(lo, hi, dd2.size, 7.0))
+
+
+def test_methods_restriction():
+    import stencil_amd as sa
+
+    dd = make_dd((8, 8, 8), 1, 2)
+    dd.set_methods(sa.Method.DIRECT_KERNEL)  # same-rank only: fine
+    h = dd.add_data(np.float32, "q")
+    dd.realize()
+    fill_interiors(dd, h)
+    dd.exchange()
+    check_full_regions(dd, h)
+
+
+def test_exchange_begin_end():
+    dd = make_dd((8, 8, 8), 1, 2)
+    h = dd.add_data(np.float32, "q")
+    dd.realize()
+    fill_interiors(dd, h)
+    dd.exchange_begin()
+    dd.exchange_end()
+    check_full_regions(dd, h)
