@@ -164,11 +164,15 @@ __global__ void __launch_bounds__(256) jacobi_kernel_v4(JacobiParams p) {
       col += p.plane;
       dcol += p.plane;
     }
-  } else if (u == body4) {
-    for (int32_t lz = lz0; lz < zEnd; ++lz)
+  } else if (u < body4 + JAC_ZCHUNK) {
+    // head cells: one z-plane per lane (a single lane looping all 16 z
+    // planes was the kernel's straggler: ~0.17 ms on a 750^3 interior)
+    const int32_t lz = lz0 + (u - body4);
+    if (lz < zEnd)
       for (int32_t lx = 0; lx < head; ++lx) scalar_cell(lx, lz);
-  } else if (u == body4 + 1) {
-    for (int32_t lz = lz0; lz < zEnd; ++lz)
+  } else if (u < body4 + 2 * JAC_ZCHUNK) {
+    const int32_t lz = lz0 + (u - body4 - JAC_ZCHUNK);
+    if (lz < zEnd)
       for (int32_t lx = p.extX - tail; lx < p.extX; ++lx) scalar_cell(lx, lz);
   }
 }
@@ -257,7 +261,7 @@ void jacobi_step(ExchangeEngine &eng, int dom, int64_t qi, const Rect3 &region,
     // vectorized row-mapped kernel; block shape tunable via env
     const int64_t a0 = region.lo.x - full.lo.x;
     const int64_t head = std::min<int64_t>(((a0 + 3) & ~3LL) - a0, ext.x);
-    const int64_t units = (ext.x - head) / 4 + 2;
+    const int64_t units = (ext.x - head) / 4 + 2 * 16; // + head/tail z lanes
     static int bx = 0, by = 0;
     if (!bx) {
       bx = 64;
