@@ -371,8 +371,13 @@ hipStream_t ExchangeEngine::comm_stream_(int dev) {
   auto it = commStreams_.find(dev);
   if (it != commStreams_.end()) return it->second;
   STENCIL_HIP(hipSetDevice(dev));
+  // highest-priority stream: halo kernels preempt compute-stream work so
+  // the exchange window (and its contention with the interior kernel)
+  // stays short (reference: RcStream Priority::HIGH, tx_cuda.cuh:68)
+  int lo = 0, hi = 0;
+  STENCIL_HIP(hipDeviceGetStreamPriorityRange(&lo, &hi));
   hipStream_t s;
-  STENCIL_HIP(hipStreamCreateWithFlags(&s, hipStreamNonBlocking));
+  STENCIL_HIP(hipStreamCreateWithPriority(&s, hipStreamNonBlocking, hi));
   commStreams_[dev] = s;
   return s;
 }
