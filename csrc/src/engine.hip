@@ -371,13 +371,11 @@ hipStream_t ExchangeEngine::comm_stream_(int dev) {
   auto it = commStreams_.find(dev);
   if (it != commStreams_.end()) return it->second;
   STENCIL_HIP(hipSetDevice(dev));
-  // highest-priority stream: halo kernels preempt compute-stream work so
-  // the exchange window (and its contention with the interior kernel)
-  // stays short (reference: RcStream Priority::HIGH, tx_cuda.cuh:68)
-  int lo = 0, hi = 0;
-  STENCIL_HIP(hipDeviceGetStreamPriorityRange(&lo, &hi));
+  // NOTE: a high-priority stream here was measured 10% SLOWER on the
+  // jacobi step (the preempting halo kernels serialized ahead of the
+  // interior kernel instead of sharing the chip) -- keep default priority
   hipStream_t s;
-  STENCIL_HIP(hipStreamCreateWithPriority(&s, hipStreamNonBlocking, hi));
+  STENCIL_HIP(hipStreamCreateWithFlags(&s, hipStreamNonBlocking));
   commStreams_[dev] = s;
   return s;
 }
