@@ -37,4 +37,11 @@ void mhd_substep(ExchangeEngine &eng, int dom, const Rect3 &region, int step, do
 void init_harmonic_f64(ExchangeEngine &eng, int dom, int64_t qi, const Rect3 &region, double base,
                        double amp, double kx, double ky, double kz, double phase, bool nextBuf);
 
+// min/max/RMS of a quantity over a region (reference: reductions.cuh)
+struct FieldStats {
+  double min, max, rms;
+};
+FieldStats field_stats(ExchangeEngine &eng, int dom, int64_t qi, const Rect3 &region,
+                       bool nextBuf = false);
+
 } // namespace stencil_amd

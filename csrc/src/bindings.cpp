@@ -262,6 +262,12 @@ PYBIND11_MODULE(_C, m) {
   m.def("mhd_substep", &mhd_substep, py::arg("eng"), py::arg("dom"), py::arg("region"),
         py::arg("step"), py::arg("dt"), py::arg("cf"), py::arg("scratch"), py::arg("stream_id") = 0);
   m.def("init_harmonic_f64", &init_harmonic_f64);
+  py::class_<FieldStats>(m, "FieldStats")
+      .def_readonly("min", &FieldStats::min)
+      .def_readonly("max", &FieldStats::max)
+      .def_readonly("rms", &FieldStats::rms);
+  m.def("field_stats", &field_stats, py::arg("eng"), py::arg("dom"), py::arg("qi"),
+        py::arg("region"), py::arg("next_buf") = false);
 
   // topology utilities (csrc/src/topo.hip)
   m.def("gpu_distance", &gpu_distance);

@@ -156,3 +156,18 @@ def test_jacobi_overlap_equals_no_overlap():
         )
     for a, b in zip(outs[0], outs[1]):
         np.testing.assert_array_equal(a, b)
+
+
+def test_field_stats_matches_numpy():
+    dd = make_dd((14, 12, 10), 1, 1)
+    h = dd.add_data(np.float32, "q")
+    dd.realize()
+    fill_interiors(dd, h, scale=2.5)
+    lo, hi = dd.local_rect(0)
+    st = _C.field_stats(
+        dd.backend.engine, 0, 0, _C.Rect3(_C.Vec3(*lo), _C.Vec3(*hi))
+    )
+    arr = dd.read_global(0, lo, hi, h).astype(np.float64)
+    assert st.min == pytest.approx(arr.min())
+    assert st.max == pytest.approx(arr.max())
+    assert st.rms == pytest.approx(np.sqrt((arr ** 2).mean()), rel=1e-12)
