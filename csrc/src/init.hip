@@ -14,12 +14,12 @@ namespace stencil_amd {
 namespace {
 
 struct InitParams {
-  char *const *slot;
+  char *base;
   int64_t pitch, plane;
   int64_t allocX, allocY, allocZ;
   int64_t loX, loY, loZ;
   int32_t extX, extY, extZ;
-  double base, amp;
+  double base0, amp;
   double kx, ky, kz; // radians per cell
   double phase;
 };
@@ -30,10 +30,9 @@ __global__ void __launch_bounds__(256) init_harmonic_f64_kernel(InitParams p) {
   const int32_t lz = blockIdx.z;
   if (lx >= p.extX || ly >= p.extY || lz >= p.extZ) return;
   const int64_t gx = p.loX + lx, gy = p.loY + ly, gz = p.loZ + lz;
-  const double v = p.base + p.amp * sin(p.kx * gx + p.ky * gy + p.kz * gz + p.phase);
-  char *base = uniform_ptr(*p.slot);
-  *(double *)(base + (gz - p.allocZ) * p.plane + (gy - p.allocY) * p.pitch + (gx - p.allocX) * 8) =
-      v;
+  const double v = p.base0 + p.amp * sin(p.kx * gx + p.ky * gy + p.kz * gz + p.phase);
+  *(double *)(p.base + (gz - p.allocZ) * p.plane + (gy - p.allocY) * p.pitch +
+              (gx - p.allocX) * 8) = v;
 }
 
 } // namespace
@@ -45,7 +44,7 @@ void init_harmonic_f64(ExchangeEngine &eng, int dom, int64_t qi, const Rect3 &re
   const Vec3 ext = region.extent();
   if (ext.flatten() <= 0) return;
   InitParams p{};
-  p.slot = (char *const *)((nextBuf ? d.dev_next_slots() : d.dev_curr_slots()) + qi);
+  p.base = nextBuf ? d.next(qi).ptr : d.curr(qi).ptr;
   p.pitch = d.curr(qi).pitch;
   p.plane = d.curr(qi).plane();
   const Rect3 full = d.full_region();
@@ -58,7 +57,7 @@ void init_harmonic_f64(ExchangeEngine &eng, int dom, int64_t qi, const Rect3 &re
   p.extX = (int32_t)ext.x;
   p.extY = (int32_t)ext.y;
   p.extZ = (int32_t)ext.z;
-  p.base = base;
+  p.base0 = base;
   p.amp = amp;
   p.kx = kx;
   p.ky = ky;

@@ -28,8 +28,11 @@ namespace stencil_amd {
 namespace {
 
 struct JacobiParams {
-  const char *const *srcSlot; // domain curr base ptr slot
-  char *const *dstSlot;       // domain next base ptr slot
+  // raw base pointers passed per launch (kernarg pointers get global
+  // address-space inference + SGPR preload; slot-indirected bases cost
+  // ~18% on this kernel -- measured with the jacobi_probe prod variant)
+  const char *src;
+  char *dst;
   int64_t pitch, plane;       // byte strides (same for curr/next)
   // global coordinate of allocation element (0,0,0)
   int64_t allocX, allocY, allocZ;
@@ -82,8 +85,8 @@ __global__ void __launch_bounds__(256) jacobi_kernel_v4(JacobiParams p) {
   const int32_t ly = blockIdx.y * blockDim.y + threadIdx.y;
   const int32_t lz0 = blockIdx.z * JAC_ZCHUNK;
   if (ly >= p.extY) return;
-  const char *srcBase = uniform_ptr(*p.srcSlot);
-  char *dstBase = uniform_ptr(*p.dstSlot);
+  const char *srcBase = p.src;
+  char *dstBase = p.dst;
   const int64_t gy = p.loY + ly;
   const int64_t ay = gy - p.allocY;
   const int32_t zEnd = min((int32_t)(lz0 + JAC_ZCHUNK), p.extZ);
@@ -164,22 +167,21 @@ __global__ void __launch_bounds__(256) jacobi_kernel_v4(JacobiParams p) {
       col += p.plane;
       dcol += p.plane;
     }
-  } else if (u < body4 + JAC_ZCHUNK) {
-    // head cells: one z-plane per lane (a single lane looping all 16 z
-    // planes was the kernel's straggler: ~0.17 ms on a 750^3 interior)
-    const int32_t lz = lz0 + (u - body4);
-    if (lz < zEnd)
+  } else if (u == body4) {
+    // (spreading these cells over one-lane-per-z-plane measured 10%
+    // SLOWER overall -- the extra block column cost more than the
+    // straggler lane; keep the compact form)
+    for (int32_t lz = lz0; lz < zEnd; ++lz)
       for (int32_t lx = 0; lx < head; ++lx) scalar_cell(lx, lz);
-  } else if (u < body4 + 2 * JAC_ZCHUNK) {
-    const int32_t lz = lz0 + (u - body4 - JAC_ZCHUNK);
-    if (lz < zEnd)
+  } else if (u == body4 + 1) {
+    for (int32_t lz = lz0; lz < zEnd; ++lz)
       for (int32_t lx = p.extX - tail; lx < p.extX; ++lx) scalar_cell(lx, lz);
   }
 }
 
 __global__ void jacobi_kernel(JacobiParams p) {
-  const char *srcBase = uniform_ptr(*p.srcSlot);
-  char *dstBase = uniform_ptr(*p.dstSlot);
+  const char *srcBase = p.src;
+  char *dstBase = p.dst;
   const int64_t total = (int64_t)p.extX * p.extY * p.extZ;
   const int64_t stride = (int64_t)gridDim.x * blockDim.x;
   for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < total; i += stride) {
@@ -206,9 +208,8 @@ __global__ void jacobi_kernel(JacobiParams p) {
   }
 }
 
-__global__ void fill_kernel(char *const *dstSlot, int64_t pitch, int64_t plane, int64_t offBytes,
+__global__ void fill_kernel(char *base, int64_t pitch, int64_t plane,
                             int32_t extX, int32_t extY, int32_t extZ, float value) {
-  char *base = uniform_ptr(*dstSlot) + offBytes;
   const int64_t total = (int64_t)extX * extY * extZ;
   const int64_t stride = (int64_t)gridDim.x * blockDim.x;
   for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < total; i += stride) {
@@ -236,8 +237,8 @@ void jacobi_step(ExchangeEngine &eng, int dom, int64_t qi, const Rect3 &region,
   const Vec3 ext = region.extent();
   if (ext.flatten() <= 0) return;
   JacobiParams p{};
-  p.srcSlot = (const char *const *)(d.dev_curr_slots() + qi);
-  p.dstSlot = (char *const *)(d.dev_next_slots() + qi);
+  p.src = d.curr(qi).ptr;
+  p.dst = d.next(qi).ptr;
   p.pitch = d.curr(qi).pitch;
   p.plane = d.curr(qi).plane();
   const Rect3 full = d.full_region();
@@ -261,7 +262,7 @@ void jacobi_step(ExchangeEngine &eng, int dom, int64_t qi, const Rect3 &region,
     // vectorized row-mapped kernel; block shape tunable via env
     const int64_t a0 = region.lo.x - full.lo.x;
     const int64_t head = std::min<int64_t>(((a0 + 3) & ~3LL) - a0, ext.x);
-    const int64_t units = (ext.x - head) / 4 + 2 * 16; // + head/tail z lanes
+    const int64_t units = (ext.x - head) / 4 + 2;
     static int bx = 0, by = 0;
     if (!bx) {
       bx = 64;
@@ -294,10 +295,10 @@ void fill_f32(ExchangeEngine &eng, int dom, int64_t qi, const Rect3 &region, flo
   const Vec3 pos = region.lo - full.lo; // allocation coords
   const Pitched &pp = d.curr(qi);
   const int64_t off = pos.z * pp.plane() + pos.y * pp.pitch + pos.x * 4;
-  char *const *slot = (char *const *)((nextBuf ? d.dev_next_slots() : d.dev_curr_slots()) + qi);
+  char *base = (nextBuf ? d.next(qi).ptr : d.curr(qi).ptr) + off;
   STENCIL_HIP(hipSetDevice(d.gpu()));
   hipLaunchKernelGGL(fill_kernel, dim3(grid_for(ext.flatten(), 256)), dim3(256), 0,
-                     eng.compute_stream(dom), slot, pp.pitch, pp.plane(), off, (int32_t)ext.x,
+                     eng.compute_stream(dom), base, pp.pitch, pp.plane(), (int32_t)ext.x,
                      (int32_t)ext.y, (int32_t)ext.z, value);
   STENCIL_HIP(hipGetLastError());
 }

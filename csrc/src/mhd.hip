@@ -38,8 +38,9 @@ namespace stencil_amd {
 namespace {
 
 struct MhdParams {
-  const char *const *currSlots; // 8 device slots
-  char *const *nextSlots;
+  // raw per-launch base pointers (kernarg > slot indirection; see jacobi)
+  const char *curr[8];
+  char *next[8];
   int64_t pitch, plane; // byte strides (identical for all 8 fp64 fields)
   int64_t allocX, allocY, allocZ;
   int64_t loX, loY, loZ;
@@ -173,8 +174,8 @@ __device__ __forceinline__ MhdCommon mhd_setup(const MhdParams &p, int32_t lx, i
   c.plane = p.plane;
 #pragma unroll
   for (int q = 0; q < 8; ++q) {
-    c.base[q] = uniform_ptr(p.currSlots[q]) + cellOff;
-    c.out[q] = uniform_ptr(p.nextSlots[q]) + cellOff;
+    c.base[q] = p.curr[q] + cellOff;
+    c.out[q] = p.next[q] + cellOff;
   }
   return c;
 }
@@ -323,8 +324,10 @@ void mhd_substep(ExchangeEngine &eng, int dom, const Rect3 &region, int step, do
   static const double ALPHA[3] = {0.0, -5.0 / 9.0, -153.0 / 128.0};
   static const double BETA[3] = {1.0 / 3.0, 15.0 / 16.0, 8.0 / 15.0};
   MhdParams p{};
-  p.currSlots = (const char *const *)d.dev_curr_slots();
-  p.nextSlots = (char *const *)d.dev_next_slots();
+  for (int q = 0; q < 8; ++q) {
+    p.curr[q] = d.curr(q).ptr;
+    p.next[q] = d.next(q).ptr;
+  }
   p.pitch = d.curr(0).pitch;
   p.plane = d.curr(0).plane();
   for (int q = 1; q < 8; ++q)

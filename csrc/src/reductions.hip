@@ -20,7 +20,7 @@ namespace stencil_amd {
 namespace {
 
 struct RedParams {
-  const char *const *slot;
+  const char *base;
   int64_t pitch, plane;
   int64_t off; // byte offset of region start
   int32_t extX, extY, extZ;
@@ -30,7 +30,7 @@ struct RedParams {
 
 __global__ void __launch_bounds__(256) reduce_kernel(RedParams p) {
   __shared__ double sMin[256], sMax[256], sSq[256];
-  const char *base = uniform_ptr(*p.slot) + p.off;
+  const char *base = p.base + p.off;
   const int64_t total = (int64_t)p.extX * p.extY * p.extZ;
   const int64_t stride = (int64_t)gridDim.x * blockDim.x;
   double mn = DBL_MAX, mx = -DBL_MAX, sq = 0.0;
@@ -79,7 +79,7 @@ FieldStats field_stats(ExchangeEngine &eng, int dom, int64_t qi, const Rect3 &re
   const Vec3 pos = region.lo - full.lo;
   const Pitched &pp = d.curr(qi);
   RedParams p{};
-  p.slot = (const char *const *)((nextBuf ? d.dev_next_slots() : d.dev_curr_slots()) + qi);
+  p.base = nextBuf ? d.next(qi).ptr : d.curr(qi).ptr;
   p.pitch = pp.pitch;
   p.plane = pp.plane();
   p.off = pos.z * p.plane + pos.y * p.pitch + pos.x * es;
