@@ -95,18 +95,27 @@ public:
   void flip_views();
   // staging buffer on `dom`'s GPU; returns buffer id
   int64_t create_buffer(int dom, int64_t bytes);
-  // gather region of quantity qi of dom's curr into buffer at byte offset
-  void add_pack(int dom, int64_t buf, int64_t offset, const Vec3 &pos, const Vec3 &ext, int64_t qi);
+  // IPC export/import of staging buffers (for the staged colocated path:
+  // thin x-faces pack into the RECEIVER's staging buffer as one contiguous
+  // xGMI stream instead of scattered remote stores)
+  std::string buffer_ipc_handle(int64_t buf);
+  int64_t open_remote_buffer(int openDev, const std::string &handle, int64_t bytes);
+  // gather region of quantity qi of dom's curr into buffer at byte offset.
+  // `group` selects the launch group: 0 = wire (default), 1/2 = colocated
+  // staging parity 0/1 (double-buffered across exchanges).
+  void add_pack(int dom, int64_t buf, int64_t offset, const Vec3 &pos, const Vec3 &ext, int64_t qi,
+                int group = 0);
   // scatter buffer bytes into region of quantity qi of dom's curr
-  void add_unpack(int dom, int64_t buf, int64_t offset, const Vec3 &pos, const Vec3 &ext, int64_t qi);
+  void add_unpack(int dom, int64_t buf, int64_t offset, const Vec3 &pos, const Vec3 &ext,
+                  int64_t qi, int group = 0);
 
   // build + upload the per-GPU job tables
   void finalize();
 
   //// per-exchange execution (stream-ordered; host-sync via the sync_* calls)
   void launch_translates();
-  void launch_packs();
-  void launch_unpacks();
+  void launch_packs(int group = 0);
+  void launch_unpacks(int group = 0);
   void sync_translates();
   void sync_packs(); // also used after unpack
   void sync_all();
@@ -132,6 +141,7 @@ private:
     char *ptr = nullptr;
     int64_t bytes = 0;
     int dev = -1;
+    bool external = false; // IPC-opened (close, don't free)
   };
   struct TranslateSpec {
     int srcDom, dstDom; // dstDom: local domain index, or remote view id
@@ -151,6 +161,7 @@ private:
     Vec3 pos, ext;
     int64_t qi;
     bool unpack;
+    int group;
   };
 
   void build_batches_(const std::vector<TranslateSpec> &ts, const std::vector<PackSpec> &ps);
@@ -169,8 +180,9 @@ private:
   std::vector<hipStream_t> computeStreams2_; // per domain, stream 1
 
   std::vector<CopyBatch> translateBatches_; // one per device with jobs
-  std::vector<CopyBatch> packBatches_;
-  std::vector<CopyBatch> unpackBatches_;
+  static constexpr int kGroups = 3; // 0 = wire, 1/2 = colo staging parity
+  std::vector<CopyBatch> packBatches_[kGroups];
+  std::vector<CopyBatch> unpackBatches_[kGroups];
   bool finalized_ = false;
 };
 

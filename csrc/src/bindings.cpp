@@ -211,12 +211,20 @@ PYBIND11_MODULE(_C, m) {
       .def("add_translate_view", &ExchangeEngine::add_translate_view)
       .def("flip_views", &ExchangeEngine::flip_views)
       .def("create_buffer", &ExchangeEngine::create_buffer)
-      .def("add_pack", &ExchangeEngine::add_pack)
-      .def("add_unpack", &ExchangeEngine::add_unpack)
+      .def("add_pack", &ExchangeEngine::add_pack, py::arg("dom"), py::arg("buf"),
+           py::arg("offset"), py::arg("pos"), py::arg("ext"), py::arg("qi"), py::arg("group") = 0)
+      .def("add_unpack", &ExchangeEngine::add_unpack, py::arg("dom"), py::arg("buf"),
+           py::arg("offset"), py::arg("pos"), py::arg("ext"), py::arg("qi"), py::arg("group") = 0)
+      .def("buffer_ipc_handle",
+           [](ExchangeEngine &e, int64_t buf) { return py::bytes(e.buffer_ipc_handle(buf)); })
+      .def("open_remote_buffer",
+           [](ExchangeEngine &e, int openDev, py::bytes handle, int64_t bytes) {
+             return e.open_remote_buffer(openDev, std::string(handle), bytes);
+           })
       .def("finalize", &ExchangeEngine::finalize)
       .def("launch_translates", &ExchangeEngine::launch_translates)
-      .def("launch_packs", &ExchangeEngine::launch_packs)
-      .def("launch_unpacks", &ExchangeEngine::launch_unpacks)
+      .def("launch_packs", &ExchangeEngine::launch_packs, py::arg("group") = 0)
+      .def("launch_unpacks", &ExchangeEngine::launch_unpacks, py::arg("group") = 0)
       .def("sync_translates", &ExchangeEngine::sync_translates)
       .def("sync_packs", &ExchangeEngine::sync_packs)
       .def("sync_all", &ExchangeEngine::sync_all)

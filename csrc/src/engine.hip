@@ -150,16 +150,25 @@ ExchangeEngine::ExchangeEngine(std::vector<std::shared_ptr<LocalDomain>> domains
 
 ExchangeEngine::~ExchangeEngine() {
   for (auto &b : translateBatches_) b.destroy();
-  for (auto &b : packBatches_) b.destroy();
-  for (auto &b : unpackBatches_) b.destroy();
+  for (int g = 0; g < kGroups; ++g) {
+    for (auto &b : packBatches_[g]) b.destroy();
+    for (auto &b : unpackBatches_[g]) b.destroy();
+  }
   for (auto &kv : commStreams_) (void)hipStreamDestroy(kv.second);
   for (auto &kv : packStreams_) (void)hipStreamDestroy(kv.second);
   for (auto s : computeStreams_)
     if (s) (void)hipStreamDestroy(s);
   for (auto s : computeStreams2_)
     if (s) (void)hipStreamDestroy(s);
-  for (auto &b : buffers_)
-    if (b.ptr) (void)hipFree(b.ptr);
+  for (auto &b : buffers_) {
+    if (!b.ptr) continue;
+    if (b.external) {
+      (void)hipSetDevice(b.dev);
+      (void)hipIpcCloseMemHandle(b.ptr);
+    } else {
+      (void)hipFree(b.ptr);
+    }
+  }
   for (auto &v : views_) {
     (void)hipSetDevice(v.openDev);
     for (int par = 0; par < 2; ++par)
@@ -258,13 +267,37 @@ int64_t ExchangeEngine::create_buffer(int dom, int64_t bytes) {
 }
 
 void ExchangeEngine::add_pack(int dom, int64_t buf, int64_t offset, const Vec3 &pos, const Vec3 &ext,
-                              int64_t qi) {
-  packSpecs_.push_back({dom, buf, offset, pos, ext, qi, false});
+                              int64_t qi, int group) {
+  packSpecs_.push_back({dom, buf, offset, pos, ext, qi, false, group});
 }
 
 void ExchangeEngine::add_unpack(int dom, int64_t buf, int64_t offset, const Vec3 &pos,
-                                const Vec3 &ext, int64_t qi) {
-  packSpecs_.push_back({dom, buf, offset, pos, ext, qi, true});
+                                const Vec3 &ext, int64_t qi, int group) {
+  packSpecs_.push_back({dom, buf, offset, pos, ext, qi, true, group});
+}
+
+std::string ExchangeEngine::buffer_ipc_handle(int64_t buf) {
+  STENCIL_HIP(hipSetDevice(buffers_[buf].dev));
+  hipIpcMemHandle_t h;
+  STENCIL_HIP(hipIpcGetMemHandle(&h, buffers_[buf].ptr));
+  return std::string((const char *)&h, sizeof(h));
+}
+
+int64_t ExchangeEngine::open_remote_buffer(int openDev, const std::string &handle, int64_t bytes) {
+  if (handle.size() != sizeof(hipIpcMemHandle_t))
+    throw std::runtime_error("open_remote_buffer: bad handle size");
+  STENCIL_HIP(hipSetDevice(openDev));
+  hipIpcMemHandle_t h;
+  std::memcpy(&h, handle.data(), sizeof(h));
+  void *p = nullptr;
+  STENCIL_HIP(hipIpcOpenMemHandle(&p, h, hipIpcMemLazyEnablePeerAccess));
+  Buffer b;
+  b.ptr = (char *)p;
+  b.bytes = bytes;
+  b.dev = openDev;
+  b.external = true; // close, don't free
+  buffers_.push_back(b);
+  return (int64_t)buffers_.size() - 1;
 }
 
 void ExchangeEngine::finalize() {
@@ -272,7 +305,12 @@ void ExchangeEngine::finalize() {
   build_batches_(translateSpecs_, packSpecs_);
   const char *g = getenv("STENCIL_AMD_GRAPHS");
   const bool graphs = g && g[0] == '1';
-  for (auto *set : {&translateBatches_, &packBatches_, &unpackBatches_})
+  std::vector<std::vector<CopyBatch> *> sets = {&translateBatches_};
+  for (int gr = 0; gr < kGroups; ++gr) {
+    sets.push_back(&packBatches_[gr]);
+    sets.push_back(&unpackBatches_[gr]);
+  }
+  for (auto *set : sets)
     for (auto &b : *set) {
       b.finalize_upload();
       if (graphs) b.capture_graph();
@@ -282,7 +320,8 @@ void ExchangeEngine::finalize() {
 
 void ExchangeEngine::build_batches_(const std::vector<TranslateSpec> &ts,
                                     const std::vector<PackSpec> &ps) {
-  std::map<int, CopyBatch> tb, pb, ub;
+  std::map<int, CopyBatch> tb;
+  std::map<std::pair<int, int>, CopyBatch> pb, ub; // (group, gpu)
 
   for (const auto &t : ts) {
     LocalDomain &s = *domains_[t.srcDom];
@@ -357,14 +396,14 @@ void ExchangeEngine::build_batches_(const std::vector<TranslateSpec> &ts,
     j.extXw = (int32_t)(rowBytes / w);
     j.extY = (int32_t)p.ext.y;
     j.nWords = (int64_t)j.extXw * p.ext.y * p.ext.z;
-    auto &batch = (p.unpack ? ub : pb)[dom.gpu()];
+    auto &batch = (p.unpack ? ub : pb)[{p.group, dom.gpu()}];
     batch.dev = dom.gpu();
     batch.jobs.push_back(j);
   }
 
   for (auto &kv : tb) translateBatches_.push_back(std::move(kv.second));
-  for (auto &kv : pb) packBatches_.push_back(std::move(kv.second));
-  for (auto &kv : ub) unpackBatches_.push_back(std::move(kv.second));
+  for (auto &kv : pb) packBatches_[kv.first.first].push_back(std::move(kv.second));
+  for (auto &kv : ub) unpackBatches_[kv.first.first].push_back(std::move(kv.second));
 }
 
 hipStream_t ExchangeEngine::comm_stream_(int dev) {
@@ -408,14 +447,14 @@ void ExchangeEngine::launch_translates() {
   for (auto &b : translateBatches_) b.launch(comm_stream_(b.dev));
   roctxRangePop();
 }
-void ExchangeEngine::launch_packs() {
+void ExchangeEngine::launch_packs(int group) {
   roctxRangePush("stencil::pack");
-  for (auto &b : packBatches_) b.launch(pack_stream_(b.dev));
+  for (auto &b : packBatches_[group]) b.launch(pack_stream_(b.dev));
   roctxRangePop();
 }
-void ExchangeEngine::launch_unpacks() {
+void ExchangeEngine::launch_unpacks(int group) {
   roctxRangePush("stencil::unpack");
-  for (auto &b : unpackBatches_) b.launch(pack_stream_(b.dev));
+  for (auto &b : unpackBatches_[group]) b.launch(pack_stream_(b.dev));
   roctxRangePop();
 }
 

@@ -57,15 +57,17 @@ class NativeBackend:
         self._ipc_active = False
         self._colo_group = None
         self._ipc_error = None
+        self._colo_parity = 0
         ipc_sends, ipc_recv_peers = [], set()
         if ctx and ctx.get("ipc", True) and os.environ.get("STENCIL_AMD_IPC", "1") != "0":
             comm = ctx["comm"]
             colo = set(comm.colocated_ranks()) - {comm.rank}
             if colo:
                 ipc_sends = [s for s in plan.sends if s.peer_rank in colo]
-                ipc_recv_peers = {r.peer_rank for r in plan.recvs if r.peer_rank in colo}
+                ipc_recvs = [r for r in plan.recvs if r.peer_rank in colo]
+                ipc_recv_peers = {r.peer_rank for r in ipc_recvs}
                 try:
-                    self._setup_ipc(plan, ctx, ipc_sends)
+                    self._setup_ipc(plan, ctx, ipc_sends, ipc_recvs)
                     plan = ExchangePlan(
                         translates=plan.translates,
                         sends=[s for s in plan.sends if s.peer_rank not in colo],
@@ -121,14 +123,25 @@ class NativeBackend:
                 }
         self.engine.finalize()
 
-    def _setup_ipc(self, plan: ExchangePlan, ctx: dict, ipc_sends):
+    @staticmethod
+    def _is_thin(m) -> bool:
+        """thin-row messages (x-faces/edges) written directly into remote
+        memory scatter 4-24 B stores over xGMI; they go through the staged
+        path instead (coalesced pack into the receiver's staging buffer +
+        local unpack)"""
+        return m.ext[0] * 8 < 64  # conservative: row < 64 B at fp64
+
+    def _setup_ipc(self, plan: ExchangePlan, ctx: dict, ipc_sends, ipc_recvs):
         """exchange hipIpc handles among colocated ranks and register
         direct-write translate jobs into the peer processes' buffers
         (the reference's ColoHaloSender direct-access idea,
         src/tx_colocated.cu, re-done with xGMI stores + a gloo barrier
-        instead of IPC events)"""
+        instead of IPC events). Thin messages are double-buffered staged
+        (see _is_thin)."""
         comm, placement = ctx["comm"], ctx["placement"]
         radius = self.radius
+        elem_sizes = [es for es, _ in self.data_defs]
+        self._colo_parity = 0
         # phase 1: export handles. A failing rank must still reach the
         # allgather (its peers would otherwise hang), so errors are
         # exported as data and re-raised on EVERY rank afterwards.
@@ -146,10 +159,35 @@ class NativeBackend:
                         "next": [d.ipc_handle(qi, True) for qi in range(nq)],
                     }
                 )
+            # receiver side of the staged path: allocate a double-buffered
+            # staging buffer per incoming (src,dst) pair's thin messages,
+            # register the parity-0/1 unpack jobs, export the IPC handle
+            staging = {}
+            for r in ipc_recvs:
+                thin = [m for m in r.messages if self._is_thin(m)]
+                if not thin:
+                    continue
+                total, chunks = wire_layout(thin, elem_sizes)
+                buf = self.engine.create_buffer(r.local_id, 2 * total)
+                dom = self.domains[r.local_id]
+                for mi, qi, off, nbytes in chunks:
+                    m = thin[mi]
+                    nd = _vec3(tuple(-c for c in m.dir))
+                    pos = dom.halo_pos(nd, True)
+                    for parity in (0, 1):
+                        self.engine.add_unpack(
+                            r.local_id, buf, off + parity * total, pos, _vec3(m.ext), qi,
+                            group=1 + parity,
+                        )
+                staging[(r.src_gid, r.dst_gid)] = (
+                    self.engine.buffer_ipc_handle(buf),
+                    total,
+                )
+            export = {"domains": export, "staging": staging}
         except Exception as e:
             export = {"error": str(e)}
         infos = comm.allgather_object(export)
-        errs = [i["error"] for i in infos if isinstance(i, dict) and "error" in i]
+        errs = [i["error"] for i in infos if "error" in i]
         if errs:
             raise RuntimeError(f"IPC export failed on some rank: {errs[0]}")
 
@@ -157,6 +195,7 @@ class NativeBackend:
         # translate specs registered, keeping the RCCL fallback clean),
         # then register the direct-write jobs.
         views = {}
+        remote_staging = {}
         open_err = None
         try:
             for s in ipc_sends:
@@ -164,11 +203,17 @@ class NativeBackend:
                 dst_li = placement.get_subdomain_id(dst_idx)
                 src_gpu = self.domains[s.local_id].gpu()
                 key = (s.peer_rank, dst_li, src_gpu)
-                if key not in views:
-                    info = infos[s.peer_rank][dst_li]
+                if key not in views and any(not self._is_thin(m) for m in s.messages):
+                    info = infos[s.peer_rank]["domains"][dst_li]
                     views[key] = self.engine.create_remote_view(
                         src_gpu, info["curr"], info["next"], info["pitch"], info["ysize"], info["es"]
                     )
+                skey = (s.src_gid, s.dst_gid)
+                stg = infos[s.peer_rank]["staging"].get(skey)
+                if stg is not None and skey not in remote_staging:
+                    handle, total = stg
+                    rb = self.engine.open_remote_buffer(src_gpu, handle, 2 * total)
+                    remote_staging[skey] = (rb, total)
         except Exception as e:
             open_err = str(e)
         # consensus: either EVERY rank uses IPC or none does (a mixed state
@@ -184,11 +229,25 @@ class NativeBackend:
             src_gpu = self.domains[s.local_id].gpu()
             key = (s.peer_rank, dst_li, src_gpu)
             dom = self.domains[s.local_id]
-            for m in s.messages:
+            fat = [m for m in s.messages if not self._is_thin(m)]
+            thin = [m for m in s.messages if self._is_thin(m)]
+            for m in fat:
                 nd = _vec3(tuple(-c for c in m.dir))
                 src_pos = dom.halo_pos(_vec3(m.dir), False)
                 dst_pos = _C.halo_pos(nd, _vec3(dst_size), radius, True)
                 self.engine.add_translate_view(s.local_id, views[key], src_pos, dst_pos, _vec3(m.ext))
+            if thin:
+                rb, total = remote_staging[(s.src_gid, s.dst_gid)]
+                _t2, chunks = wire_layout(thin, elem_sizes)
+                assert _t2 == total, "staged wire layout mismatch"
+                for mi, qi, off, nbytes in chunks:
+                    m = thin[mi]
+                    pos = dom.halo_pos(_vec3(m.dir), False)
+                    for parity in (0, 1):
+                        self.engine.add_pack(
+                            s.local_id, rb, off + parity * total, pos, _vec3(m.ext), qi,
+                            group=1 + parity,
+                        )
 
     def _make_colo_groups(self, ctx):
         """per-node gloo subgroup for the post-translate IPC barrier.
@@ -227,6 +286,10 @@ class NativeBackend:
         (if cross-rank) the pack kernels. Returns immediately; GPU work
         overlaps whatever the app runs on its compute streams."""
         self.engine.launch_translates()
+        if self._ipc_active:
+            # staged thin messages: coalesced pack straight into the
+            # receiver's staging buffer (parity-selected half)
+            self.engine.launch_packs(1 + self._colo_parity)
         if self._has_wire:
             self.engine.launch_packs()
 
