@@ -58,6 +58,7 @@ class NativeBackend:
         self._colo_group = None
         self._ipc_error = None
         self._colo_parity = 0
+        self._staging_recv = {}
         ipc_sends, ipc_recv_peers = [], set()
         if ctx and ctx.get("ipc", True) and os.environ.get("STENCIL_AMD_IPC", "1") != "0":
             comm = ctx["comm"]
@@ -183,6 +184,7 @@ class NativeBackend:
                     self.engine.buffer_ipc_handle(buf),
                     total,
                 )
+                self._staging_recv[(r.src_gid, r.dst_gid)] = (buf, total)
             export = {"domains": export, "staging": staging}
         except Exception as e:
             export = {"error": str(e)}
