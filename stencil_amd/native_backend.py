@@ -324,9 +324,17 @@ class NativeBackend:
             self.engine.launch_unpacks()
         self.engine.sync_all()
         if self._ipc_active:
+            # all colocated ranks' direct writes and staged packs are
+            # complete after the barrier (each rank synced its own streams
+            # above); then unpack this exchange's staging parity locally.
+            # The barrier also keeps senders at most one exchange ahead,
+            # which is what makes the two staging parities sufficient.
             import torch.distributed as dist
 
             dist.barrier(group=self._colo_group)
+            self.engine.launch_unpacks(1 + self._colo_parity)
+            self.engine.sync_packs()
+            self._colo_parity ^= 1
 
     def exchange(self):
         self.exchange_begin()
