@@ -140,7 +140,7 @@ __global__ void __launch_bounds__(256) probe_prod(P p, const char *const *srcSlo
       out.y = (cc.z + cc.x + py.y + my.y + cp.y + cm.y) / 6.0f;
       out.z = (cc.w + cc.y + py.z + my.z + cp.z + cm.z) / 6.0f;
       out.w = (right + cc.z + py.w + my.w + cp.w + cm.w) / 6.0f;
-      if (withSphere) {
+      if (withSphere & 1) {
         const int32_t dy = y - p.ny / 2, dz = z - p.nz / 2;
         const int32_t yz2 = dy * dy + dz * dz;
         const int32_t srad = p.nx4 * 4 / 10;
@@ -159,7 +159,7 @@ __global__ void __launch_bounds__(256) probe_prod(P p, const char *const *srcSlo
       col += p.plane;
       dcol += p.plane;
     }
-  } else if (u == body4) { // head/tail scalar cells (mirrors production)
+  } else if (u == body4 && (withSphere & 2)) { // head/tail scalar cells (mirrors production)
     for (int32_t z = z0; z < zEnd; ++z) {
       const char *rowC = srcBase + (int64_t)(z + 3) * p.plane + (int64_t)(y + 3) * p.pitch;
       char *rowD = dstBase + (int64_t)(z + 3) * p.plane + (int64_t)(y + 3) * p.pitch;
@@ -206,9 +206,9 @@ int main(int argc, char **argv) {
   CHECK(hipMemcpy(slots + 1, &b, sizeof(char *), hipMemcpyHostToDevice));
   dim3 grdp((p.nx4 + 2 + 63) / 64, (p.ny + 3) / 4, (p.nz + ZCH - 1) / ZCH);
 
-  const int NV = 6;
-  const char *names[NV] = {"copy", "xz", "full", "full8", "prod", "prod+sph"};
-  double best[NV] = {1e30, 1e30, 1e30, 1e30, 1e30, 1e30};
+  const int NV = 7;
+  const char *names[NV] = {"copy", "xz", "full", "full8", "prod-notail", "prod+sph+tail", "prod+tail"};
+  double best[NV] = {1e30, 1e30, 1e30, 1e30, 1e30, 1e30, 1e30};
   for (int r = 0; r < rounds; ++r) {
     for (int v = 0; v < NV; ++v) {
       CHECK(hipEventRecord(e0));
@@ -223,7 +223,11 @@ int main(int argc, char **argv) {
         break;
       case 5:
         hipLaunchKernelGGL(probe_prod, grdp, blk, 0, 0, p, (const char *const *)slots,
-                           (char *const *)(slots + 1), 1);
+                           (char *const *)(slots + 1), 3);
+        break;
+      case 6:
+        hipLaunchKernelGGL(probe_prod, grdp, blk, 0, 0, p, (const char *const *)slots,
+                           (char *const *)(slots + 1), 2);
         break;
       }
       CHECK(hipEventRecord(e1));
