@@ -95,8 +95,13 @@ public:
   char **dev_next_slots() const { return devNextRaw_; }
 
   // IPC export of a quantity buffer (hipIpcMemHandle_t blob) for the
-  // cross-process direct-write transport
+  // cross-process direct-write transport. The handle refers to the
+  // allocation BASE; the importer must add pad_bytes(qi) to reach
+  // element (0,0,0).
   std::string ipc_handle(int64_t qi, bool next) const;
+  // leading alignment pad (so the interior x-start is 16 B aligned and
+  // vectorized stencil kernels have no head/tail cells)
+  int64_t pad_bytes(int64_t qi) const { return padBytes_.at(qi); }
 
   // blocking element-region copies (pos in allocation coords)
   void region_to_host(void *dst, const Vec3 &pos, const Vec3 &ext, int64_t qi, bool fromNext = false) const;
@@ -114,6 +119,8 @@ private:
   std::vector<int64_t> elemSize_;
   std::vector<std::string> name_;
   std::vector<Pitched> curr_, next_;
+  std::vector<int64_t> padBytes_;      // leading pad per quantity
+  std::vector<char *> allocBases_;     // hipMalloc bases (curr then next)
   char **devCurrRaw_ = nullptr; // device array of curr base ptrs
   char **devNextRaw_ = nullptr;
 };

@@ -86,7 +86,8 @@ public:
   int64_t create_remote_view(int openDev, const std::vector<std::string> &currHandles,
                              const std::vector<std::string> &nextHandles,
                              const std::vector<int64_t> &pitches, const std::vector<int64_t> &ysizes,
-                             const std::vector<int64_t> &elemSizes);
+                             const std::vector<int64_t> &elemSizes,
+                             const std::vector<int64_t> &pads);
   // direct-write one region (all quantities) into a remote view's curr
   void add_translate_view(int srcDom, int64_t view, const Vec3 &srcPos, const Vec3 &dstPos,
                           const Vec3 &ext);
@@ -152,7 +153,7 @@ private:
     int openDev = -1;
     int parity = 0;
     std::vector<char *> base[2]; // [parity][qi] opened pointers
-    std::vector<int64_t> pitch, ysize, elemSize;
+    std::vector<int64_t> pitch, ysize, elemSize, pads;
     char **devSlots = nullptr; // local device array, refreshed on flip
   };
   struct PackSpec {

@@ -171,6 +171,7 @@ PYBIND11_MODULE(_C, m) {
            },
            py::arg("pos"), py::arg("ext"), py::arg("qi"), py::arg("from_next") = false)
       .def("curr_pitch", [](const LocalDomain &d, int64_t qi) { return d.curr(qi).pitch; })
+      .def("pad_bytes", &LocalDomain::pad_bytes)
       .def("curr_ysize", [](const LocalDomain &d, int64_t qi) { return d.curr(qi).ysize; })
       .def("ipc_handle",
            [](const LocalDomain &d, int64_t qi, bool next) { return py::bytes(d.ipc_handle(qi, next)); },
@@ -202,11 +203,12 @@ PYBIND11_MODULE(_C, m) {
       .def("create_remote_view",
            [](ExchangeEngine &e, int openDev, const std::vector<py::bytes> &cur,
               const std::vector<py::bytes> &nxt, const std::vector<int64_t> &pitches,
-              const std::vector<int64_t> &ysizes, const std::vector<int64_t> &es) {
+              const std::vector<int64_t> &ysizes, const std::vector<int64_t> &es,
+              const std::vector<int64_t> &pads) {
              std::vector<std::string> c, n;
              for (auto &b : cur) c.push_back(b);
              for (auto &b : nxt) n.push_back(b);
-             return e.create_remote_view(openDev, c, n, pitches, ysizes, es);
+             return e.create_remote_view(openDev, c, n, pitches, ysizes, es, pads);
            })
       .def("add_translate_view", &ExchangeEngine::add_translate_view)
       .def("flip_views", &ExchangeEngine::flip_views)

@@ -15,10 +15,8 @@ LocalDomain::LocalDomain(const Vec3 &sz, const Vec3 &origin, int dev)
 LocalDomain::~LocalDomain() {
   if (!realized_) return;
   (void)hipSetDevice(dev_);
-  for (auto &p : curr_)
-    if (p.ptr) (void)hipFree(p.ptr);
-  for (auto &p : next_)
-    if (p.ptr) (void)hipFree(p.ptr);
+  for (char *b : allocBases_)
+    if (b) (void)hipFree(b);
   if (devCurrRaw_) (void)hipFree(devCurrRaw_);
   if (devNextRaw_) (void)hipFree(devNextRaw_);
 }
@@ -35,15 +33,30 @@ void LocalDomain::realize() {
   const Vec3 raw = raw_size();
   curr_.resize(elemSize_.size());
   next_.resize(elemSize_.size());
+  padBytes_.resize(elemSize_.size(), 0);
+  // the interior (overlap-mode) region starts at allocation x =
+  // radius.x(-1) + max(-x-direction radii); pad the allocation so that
+  // address is 16 B aligned and the vectorized stencil kernels have no
+  // scalar head/tail cells
+  int64_t shrink = 0;
+  for (int dy = -1; dy <= 1; ++dy)
+    for (int dz = -1; dz <= 1; ++dz) shrink = std::max(shrink, radius_.dir(-1, dy, dz));
   for (size_t qi = 0; qi < elemSize_.size(); ++qi) {
+    const int64_t es = elemSize_[qi];
     // pitch rows to 256 B so every y-row starts on an HBM-friendly boundary
-    const int64_t pitch = align_up(raw.x * elemSize_[qi], 256);
-    const int64_t bytes = pitch * raw.y * raw.z;
+    const int64_t pitch = align_up(raw.x * es, 256);
+    const int64_t interiorOff = (radius_.x(-1) + shrink) * es;
+    const int64_t pad = (16 - interiorOff % 16) % 16;
+    padBytes_[qi] = pad;
+    const int64_t bytes = pitch * raw.y * raw.z + pad + 256;
     for (Pitched *buf : {&curr_[qi], &next_[qi]}) {
       buf->pitch = pitch;
       buf->ysize = raw.y;
-      STENCIL_HIP(hipMalloc((void **)&buf->ptr, bytes));
-      STENCIL_HIP(hipMemset(buf->ptr, 0, bytes));
+      char *base = nullptr;
+      STENCIL_HIP(hipMalloc((void **)&base, bytes));
+      STENCIL_HIP(hipMemset(base, 0, bytes));
+      allocBases_.push_back(base);
+      buf->ptr = base + pad;
     }
   }
   // device pointer tables (fixed addresses; contents refreshed on swap)
@@ -75,7 +88,9 @@ std::string LocalDomain::ipc_handle(int64_t qi, bool next) const {
   STENCIL_HIP(hipSetDevice(dev_));
   hipIpcMemHandle_t h;
   const Pitched &p = next ? next_.at(qi) : curr_.at(qi);
-  STENCIL_HIP(hipIpcGetMemHandle(&h, p.ptr));
+  // the handle must be taken on the allocation base (ptr is pad-offset);
+  // importers add pad_bytes(qi)
+  STENCIL_HIP(hipIpcGetMemHandle(&h, p.ptr - padBytes_.at(qi)));
   return std::string((const char *)&h, sizeof(h));
 }
 

@@ -172,8 +172,8 @@ ExchangeEngine::~ExchangeEngine() {
   for (auto &v : views_) {
     (void)hipSetDevice(v.openDev);
     for (int par = 0; par < 2; ++par)
-      for (char *p : v.base[par])
-        if (p) (void)hipIpcCloseMemHandle(p);
+      for (size_t qi = 0; qi < v.base[par].size(); ++qi)
+        if (v.base[par][qi]) (void)hipIpcCloseMemHandle(v.base[par][qi] - v.pads[qi]);
     if (v.devSlots) (void)hipFree(v.devSlots);
   }
 }
@@ -211,16 +211,18 @@ int64_t ExchangeEngine::create_remote_view(int openDev, const std::vector<std::s
                                            const std::vector<std::string> &nextHandles,
                                            const std::vector<int64_t> &pitches,
                                            const std::vector<int64_t> &ysizes,
-                                           const std::vector<int64_t> &elemSizes) {
+                                           const std::vector<int64_t> &elemSizes,
+                                           const std::vector<int64_t> &pads) {
   const size_t nq = currHandles.size();
   if (nextHandles.size() != nq || pitches.size() != nq || ysizes.size() != nq ||
-      elemSizes.size() != nq)
+      elemSizes.size() != nq || pads.size() != nq)
     throw std::runtime_error("create_remote_view: size mismatch");
   RemoteView v;
   v.openDev = openDev;
   v.pitch = pitches;
   v.ysize = ysizes;
   v.elemSize = elemSizes;
+  v.pads = pads;
   STENCIL_HIP(hipSetDevice(openDev));
   auto open_one = [&](const std::string &blob) {
     if (blob.size() != sizeof(hipIpcMemHandle_t))
@@ -232,8 +234,10 @@ int64_t ExchangeEngine::create_remote_view(int openDev, const std::vector<std::s
     return (char *)p;
   };
   for (size_t qi = 0; qi < nq; ++qi) {
-    v.base[0].push_back(open_one(currHandles[qi]));
-    v.base[1].push_back(open_one(nextHandles[qi]));
+    // the IPC handle maps the allocation base; element (0,0,0) sits
+    // pads[qi] bytes in (LocalDomain alignment pad)
+    v.base[0].push_back(open_one(currHandles[qi]) + pads[qi]);
+    v.base[1].push_back(open_one(nextHandles[qi]) + pads[qi]);
   }
   STENCIL_HIP(hipMalloc((void **)&v.devSlots, nq * sizeof(char *)));
   STENCIL_HIP(

@@ -156,6 +156,7 @@ class NativeBackend:
                         "pitch": [d.curr_pitch(qi) for qi in range(nq)],
                         "ysize": [d.curr_ysize(qi) for qi in range(nq)],
                         "es": [d.elem_size(qi) for qi in range(nq)],
+                        "pad": [d.pad_bytes(qi) for qi in range(nq)],
                         "curr": [d.ipc_handle(qi, False) for qi in range(nq)],
                         "next": [d.ipc_handle(qi, True) for qi in range(nq)],
                     }
@@ -208,7 +209,8 @@ class NativeBackend:
                 if key not in views and any(not self._is_thin(m) for m in s.messages):
                     info = infos[s.peer_rank]["domains"][dst_li]
                     views[key] = self.engine.create_remote_view(
-                        src_gpu, info["curr"], info["next"], info["pitch"], info["ysize"], info["es"]
+                        src_gpu, info["curr"], info["next"], info["pitch"], info["ysize"],
+                        info["es"], info["pad"]
                     )
                 skey = (s.src_gid, s.dst_gid)
                 stg = infos[s.peer_rank]["staging"].get(skey)
