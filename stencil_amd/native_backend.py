@@ -46,6 +46,9 @@ class NativeBackend:
         self._cpu_mirror = {}
         self._ipc_active = False
         self._colo_group = None
+        self._ipc_error = None
+        self._colo_parity = 0
+        self._staging_recv = {}
 
     # ---- plan registration ----
     def register_plan(self, plan: ExchangePlan, ctx: Optional[dict] = None):
@@ -59,14 +62,12 @@ class NativeBackend:
         self._ipc_error = None
         self._colo_parity = 0
         self._staging_recv = {}
-        ipc_sends, ipc_recv_peers = [], set()
         if ctx and ctx.get("ipc", True) and os.environ.get("STENCIL_AMD_IPC", "1") != "0":
             comm = ctx["comm"]
             colo = set(comm.colocated_ranks()) - {comm.rank}
             if colo:
                 ipc_sends = [s for s in plan.sends if s.peer_rank in colo]
                 ipc_recvs = [r for r in plan.recvs if r.peer_rank in colo]
-                ipc_recv_peers = {r.peer_rank for r in ipc_recvs}
                 try:
                     self._setup_ipc(plan, ctx, ipc_sends, ipc_recvs)
                     plan = ExchangePlan(
