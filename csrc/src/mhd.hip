@@ -243,21 +243,24 @@ __global__ void __launch_bounds__(256) mhd_lorentz_kernel(MhdParams p, MhdScratc
   const double ix = 1.0 / p.dsx, iy = 1.0 / p.dsy, iz = 1.0 / p.dsz;
   char *out = sc.ptr + (int64_t)lz * sc.planeStride + (int64_t)ly * sc.rowStride + (int64_t)lx * 8;
   // j = grad(div A) - lap(A), expanded per component so each is a small
-  // independent expression (2 cross + 2 second derivatives):
+  // independent expression (2 cross + 2 second derivatives) and the
+  // register live-set stays flat:
   //   j.x = dxy Ay + dxz Az - dyy Ax - dzz Ax   (etc. cyclically)
-  // B shares the A line loads already on chip, so the full Lorentz force
-  // j x B is produced here (the momentum kernel then needs no A at all)
-  const double jx = st.dxy(AAY, ix, iy) + st.dxz(AAZ, ix, iz) - st.dyy(AAX, iy * iy) -
-                    st.dzz(AAX, iz * iz);
-  const double jy = st.dxy(AAX, ix, iy) + st.dyz(AAZ, iy, iz) - st.dxx(AAY, ix * ix) -
-                    st.dzz(AAY, iz * iz);
-  const double jz = st.dxz(AAX, ix, iz) + st.dyz(AAY, iy, iz) - st.dxx(AAZ, ix * ix) -
-                    st.dyy(AAZ, iy * iy);
-  const Vec3d B = {st.dy(AAZ, iy) - st.dz(AAY, iz), st.dz(AAX, iz) - st.dx(AAZ, ix),
-                   st.dx(AAY, ix) - st.dy(AAX, iy)};
-  __builtin_nontemporal_store(jy * B.z - jz * B.y, (double *)out);
-  __builtin_nontemporal_store(jz * B.x - jx * B.z, (double *)(out + sc.compStride));
-  __builtin_nontemporal_store(jx * B.y - jy * B.x, (double *)(out + 2 * sc.compStride));
+  {
+    const double jx = st.dxy(AAY, ix, iy) + st.dxz(AAZ, ix, iz) - st.dyy(AAX, iy * iy) -
+                      st.dzz(AAX, iz * iz);
+    __builtin_nontemporal_store(jx, (double *)out);
+  }
+  {
+    const double jy = st.dxy(AAX, ix, iy) + st.dyz(AAZ, iy, iz) - st.dxx(AAY, ix * ix) -
+                      st.dzz(AAY, iz * iz);
+    __builtin_nontemporal_store(jy, (double *)(out + sc.compStride));
+  }
+  {
+    const double jz = st.dxz(AAX, ix, iz) + st.dyz(AAY, iy, iz) - st.dxx(AAZ, ix * ix) -
+                      st.dyy(AAZ, iy * iy);
+    __builtin_nontemporal_store(jz, (double *)(out + 2 * sc.compStride));
+  }
 }
 
 // kernel 2b: momentum update (advection + pressure + viscosity + the
@@ -278,8 +281,12 @@ __global__ void __launch_bounds__(256, 3) mhd_momentum_kernel(MhdParams p, MhdSc
   const double rho_inv = exp(-st.c(LNRHO));
   const char *jb =
       sc.ptr + (int64_t)lz * sc.planeStride + (int64_t)ly * sc.rowStride + (int64_t)lx * 8;
-  const Vec3d jxB = {*(const double *)jb, *(const double *)(jb + sc.compStride),
-                     *(const double *)(jb + 2 * sc.compStride)};
+  const Vec3d B = {st.dy(AAZ, 1.0 / p.dsy) - st.dz(AAY, 1.0 / p.dsz),
+                   st.dz(AAX, 1.0 / p.dsz) - st.dx(AAZ, 1.0 / p.dsx),
+                   st.dx(AAY, 1.0 / p.dsx) - st.dy(AAX, 1.0 / p.dsy)};
+  const Vec3d j = {*(const double *)jb, *(const double *)(jb + sc.compStride),
+                   *(const double *)(jb + 2 * sc.compStride)};
+  const Vec3d jxB = cross(j, B);
 
   {
     const double ugradu = uu.x * st.dx(UUX, ix) + uu.y * st.dy(UUX, iy) + uu.z * st.dz(UUX, iz);
