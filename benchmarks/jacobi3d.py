@@ -35,6 +35,7 @@ def main():
     ap.add_argument("--gpus", type=int, default=1)
     ap.add_argument("--size", type=int, default=512, help="per-GPU (weak) or total (strong) edge")
     ap.add_argument("--strong", action="store_true")
+    ap.add_argument("--radius", type=int, default=1)
     ap.add_argument("--iters", type=int, default=30)
     ap.add_argument("--trivial", action="store_true", help="trivial placement")
     ap.add_argument("--no-overlap", action="store_true")
@@ -64,7 +65,7 @@ def main():
         size = tuple(args.size * d for d in dims)
 
     placement = PlacementStrategy.Trivial if args.trivial else PlacementStrategy.NodeAware
-    app = Jacobi3D(size, backend=args.backend, gpus=gpus, placement=placement)
+    app = Jacobi3D(size, backend=args.backend, gpus=gpus, placement=placement, radius=args.radius)
     app.realize()
 
     stats = Statistics()

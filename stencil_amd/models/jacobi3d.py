@@ -23,9 +23,13 @@ class Jacobi3D:
         gpus: Optional[List[int]] = None,
         placement: PlacementStrategy = PlacementStrategy.NodeAware,
         device: str = "cpu",
+        radius: int = 1,
     ):
+        """radius >= 1: halo depth (the 7-point kernel reads 1 cell; deeper
+        radii exercise the deeper-halo exchange, reference jacobi3d-strong
+        r=2 configuration)"""
         self.dd = DistributedDomain(*size, backend=backend, device=device)
-        self.dd.set_radius(1)
+        self.dd.set_radius(radius)
         self.dd.set_placement(placement)
         if gpus is not None:
             self.dd.set_gpus(gpus)
