@@ -37,14 +37,17 @@ def main():
     world = int(os.environ.get("WORLD_SIZE", "1"))
     rank = int(os.environ.get("RANK", "0"))
     dist = None
+    ctrl = None  # gloo control group (barriers/reductions off the GPU)
     if world > 1:
         import torch
         import torch.distributed as dist_mod
 
         dist = dist_mod
         local_rank = int(os.environ.get("LOCAL_RANK", rank))
-        torch.cuda.set_device(local_rank)
+        if torch.cuda.is_available():
+            torch.cuda.set_device(local_rank % torch.cuda.device_count())
         dist.init_process_group(backend="cpu:gloo,cuda:nccl")
+        ctrl = dist.new_group(backend="gloo")
         n_gpus = world
         gpus = None  # DistributedDomain defaults to [LOCAL_RANK]
     else:
@@ -67,7 +70,7 @@ def main():
 
     def barrier_sync():
         if dist is not None:
-            dist.barrier()
+            dist.barrier(group=ctrl)
         from stencil_amd import _C
 
         if args.backend == "native":
@@ -88,7 +91,7 @@ def main():
         import torch
 
         t = torch.tensor([elapsed], dtype=torch.float64)
-        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        dist.all_reduce(t, op=dist.ReduceOp.MAX, group=ctrl)
         elapsed = float(t.item())
 
     total_cells = size[0] * size[1] * size[2]
@@ -98,7 +101,7 @@ def main():
         import torch
 
         t = torch.tensor([xbytes], dtype=torch.float64)
-        dist.all_reduce(t, op=dist.ReduceOp.SUM)
+        dist.all_reduce(t, op=dist.ReduceOp.SUM, group=ctrl)
         xbytes = float(t.item())
 
     if rank == 0:
