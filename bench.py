@@ -125,8 +125,16 @@ def main():
                 "radius": 1,
                 "overlap": not args.no_overlap,
                 "exchange_bytes_per_iter": xbytes,
-                "parallelism": f"domain-decomposition {dims[0]}x{dims[1]}x{dims[2]}"
-                + (" multi-process-rccl" if world > 1 else " single-process-xgmi"),
+                "parallelism": f"domain-decomposition {dims[0]}x{dims[1]}x{dims[2]} "
+                + (
+                    (
+                        "multi-process-ipc-xgmi"
+                        if getattr(app.dd.backend, "_ipc_active", False)
+                        else "multi-process-rccl"
+                    )
+                    if world > 1
+                    else "single-process-xgmi"
+                ),
             },
         }
         print(json.dumps(out), flush=True)

@@ -56,11 +56,13 @@ def _worker(rank, world, port, q, use_ipc):
         q.put((rank, f"FAIL: {e}\n{traceback.format_exc()}"))
 
 
-@pytest.mark.parametrize("use_ipc,port", [(False, 29717), (True, 29721)])
-def test_native_two_ranks_one_gpu(use_ipc, port):
+@pytest.mark.parametrize(
+    "world,use_ipc,port", [(2, False, 29717), (2, True, 29721), (4, True, 29725)]
+)
+def test_native_multi_rank_one_gpu(world, use_ipc, port):
     ctx = mp.get_context("spawn")
     q = ctx.Queue()
-    procs = [ctx.Process(target=_worker, args=(r, 2, port, q, use_ipc)) for r in range(2)]
+    procs = [ctx.Process(target=_worker, args=(r, world, port, q, use_ipc)) for r in range(world)]
     for p in procs:
         p.start()
     results = [q.get(timeout=300) for _ in procs]
