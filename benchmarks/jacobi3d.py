@@ -36,6 +36,8 @@ def main():
     ap.add_argument("--size", type=int, default=512, help="per-GPU (weak) or total (strong) edge")
     ap.add_argument("--strong", action="store_true")
     ap.add_argument("--radius", type=int, default=1)
+    ap.add_argument("--halo-multiplier", type=int, default=1,
+                    help="exchange every m-th step with m*radius halos (temporal blocking)")
     ap.add_argument("--iters", type=int, default=30)
     ap.add_argument("--trivial", action="store_true", help="trivial placement")
     ap.add_argument("--no-overlap", action="store_true")
@@ -65,7 +67,8 @@ def main():
         size = tuple(args.size * d for d in dims)
 
     placement = PlacementStrategy.Trivial if args.trivial else PlacementStrategy.NodeAware
-    app = Jacobi3D(size, backend=args.backend, gpus=gpus, placement=placement, radius=args.radius)
+    app = Jacobi3D(size, backend=args.backend, gpus=gpus, placement=placement,
+                   radius=args.radius, halo_multiplier=args.halo_multiplier)
     app.realize()
 
     stats = Statistics()

@@ -24,12 +24,24 @@ class Jacobi3D:
         placement: PlacementStrategy = PlacementStrategy.NodeAware,
         device: str = "cpu",
         radius: int = 1,
+        halo_multiplier: int = 1,
     ):
         """radius >= 1: halo depth (the 7-point kernel reads 1 cell; deeper
         radii exercise the deeper-halo exchange, reference jacobi3d-strong
-        r=2 configuration)"""
+        r=2 configuration).
+
+        halo_multiplier m > 1 enables communication-avoiding temporal
+        blocking (the reference's v3 wish-list item, README.md:313-315):
+        halos are allocated and exchanged m*radius deep every m-th step;
+        between exchanges each step computes a region EXPANDED into the
+        still-valid halo by (m-1-phase)*radius cells. Results are
+        identical to exchanging every step."""
+        assert halo_multiplier >= 1
+        self.m = int(halo_multiplier)
+        self.kernel_radius = int(radius)
+        self._phase = 0
         self.dd = DistributedDomain(*size, backend=backend, device=device)
-        self.dd.set_radius(radius)
+        self.dd.set_radius(radius * self.m)
         self.dd.set_placement(placement)
         if gpus is not None:
             self.dd.set_gpus(gpus)
@@ -49,6 +61,9 @@ class Jacobi3D:
 
     def step(self, overlap: bool = True):
         dd = self.dd
+        if self.m > 1:
+            self._step_multiplied()
+            return
         if overlap:
             # interior compute (on compute streams) overlaps the exchange
             for li in range(dd.num_local()):
@@ -70,6 +85,24 @@ class Jacobi3D:
                 dd.backend.jacobi_step(li, self.h.index, lo, hi, self.compute_lo, self.compute_hi)
         dd.backend.sync_compute()
         dd.swap()
+
+    def _step_multiplied(self):
+        """temporal-blocking step: exchange every m-th call, compute an
+        expanded region in between (expansion shrinks by kernel_radius per
+        phase, reaching the exact compute region just before the next
+        exchange)"""
+        dd = self.dd
+        if self._phase == 0:
+            dd.exchange()
+        e = (self.m - 1 - self._phase) * self.kernel_radius
+        for li in range(dd.num_local()):
+            lo, hi = dd.local_rect(li)
+            glo = tuple(c - e for c in lo)
+            ghi = tuple(c + e for c in hi)
+            dd.backend.jacobi_step(li, self.h.index, glo, ghi, self.compute_lo, self.compute_hi)
+        dd.backend.sync_compute()
+        dd.swap()
+        self._phase = (self._phase + 1) % self.m
 
     def run(self, iters: int) -> float:
         t0 = time.perf_counter()

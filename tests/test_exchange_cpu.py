@@ -155,3 +155,27 @@ def test_exchange_begin_end():
     dd.exchange_begin()
     dd.exchange_end()
     check_full_regions(dd, h)
+
+
+@pytest.mark.parametrize("m", [2, 3])
+def test_halo_multiplier_matches_plain(m):
+    """temporal blocking (exchange every m-th step with m-deep halos) must
+    reproduce the every-step-exchange results exactly"""
+    from stencil_amd.models.jacobi3d import Jacobi3D
+
+    size = (18, 15, 12)
+    outs = []
+    for mult in (1, m):
+        app = Jacobi3D(size, backend="torch", gpus=[0, 0], halo_multiplier=mult)
+        app.realize()
+        fill_interiors(app.dd, app.h)
+        for _ in range(2 * m):
+            app.step()
+        outs.append(
+            [
+                app.dd.read_global(li, *app.dd.local_rect(li), app.h)
+                for li in range(app.dd.num_local())
+            ]
+        )
+    for a, b in zip(outs[0], outs[1]):
+        np.testing.assert_array_equal(a, b)

@@ -171,3 +171,24 @@ def test_field_stats_matches_numpy():
     assert st.min == pytest.approx(arr.min())
     assert st.max == pytest.approx(arr.max())
     assert st.rms == pytest.approx(np.sqrt((arr ** 2).mean()), rel=1e-12)
+
+
+def test_halo_multiplier_native_matches_plain():
+    from stencil_amd.models.jacobi3d import Jacobi3D
+
+    size = (18, 15, 12)
+    outs = []
+    for mult in (1, 2):
+        app = Jacobi3D(size, backend="native", gpus=[0, 0], halo_multiplier=mult)
+        app.realize()
+        fill_interiors(app.dd, app.h)
+        for _ in range(4):
+            app.step()
+        outs.append(
+            [
+                app.dd.read_global(li, *app.dd.local_rect(li), app.h)
+                for li in range(app.dd.num_local())
+            ]
+        )
+    for a, b in zip(outs[0], outs[1]):
+        np.testing.assert_array_equal(a, b)
