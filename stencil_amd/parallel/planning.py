@@ -96,6 +96,12 @@ def plan_exchange(placement, radius: "_C.Radius", rank: int) -> ExchangePlan:
                     dst_rank = placement.get_rank(dst_idx)
                     dst_size = placement.subdomain_size(dst_idx)
                     s_ext = _C.halo_extent(_vec3(neg), _vec3(dst_size), radius).tuple()
+                    if s_ext[0] * s_ext[1] * s_ext[2] == 0:
+                        # degenerate: a diagonal radius is set but a face
+                        # radius of one of its components is 0, so the halo
+                        # region has no volume (the reference fatals here,
+                        # src/packer.cu:83; we skip the empty message)
+                        continue
                     msg = Message(d, my_gid, dst_gid, s_ext)
                     if dst_rank == rank:
                         plan.translates.append(
@@ -113,7 +119,9 @@ def plan_exchange(placement, radius: "_C.Radius", rank: int) -> ExchangePlan:
                     src_rank = placement.get_rank(src_idx)
                     my_size = placement.subdomain_size(my_idx)
                     r_ext = _C.halo_extent(_vec3(neg), _vec3(my_size), radius).tuple()
-                    if src_rank != rank:  # same-rank handled by the send loop
+                    if src_rank != rank and r_ext[0] * r_ext[1] * r_ext[2] > 0:
+                        # (same-rank recvs are covered by the send loop;
+                        # zero-volume messages skipped as above)
                         key = (src_rank, src_gid, my_gid)
                         if key not in recvs:
                             recvs[key] = WirePlanItem(src_rank, src_gid, my_gid, li)

@@ -37,6 +37,46 @@ def full_region_of(dd, li):
     return flo, fhi
 
 
+def check_valid_regions(dd, handle, scale=1.0):
+    """verify the interior plus every halo side whose direction has a
+    nonzero radius (sides with radius 0 are never exchanged and hold
+    undefined data -- legal for sparse per-direction radii)"""
+    import itertools
+
+    r = dd.radius
+    for li in range(dd.num_local()):
+        lo, hi = dd.local_rect(li)
+        regions = [(lo, hi)]
+        for d in itertools.product((-1, 0, 1), repeat=3):
+            if d == (0, 0, 0) or r.dir(*d) == 0:
+                continue
+            ext = [0, 0, 0]
+            glo = [0, 0, 0]
+            for i in range(3):
+                face = (r.x, r.y, r.z)[i]
+                if d[i] == 0:
+                    glo[i], ext[i] = lo[i], hi[i] - lo[i]
+                elif d[i] < 0:
+                    ext[i] = face(-1)
+                    glo[i] = lo[i] - ext[i]
+                else:
+                    ext[i] = face(1)
+                    glo[i] = hi[i]
+            if ext[0] * ext[1] * ext[2] == 0:
+                continue
+            regions.append((tuple(glo), tuple(glo[i] + ext[i] for i in range(3))))
+        for (glo, ghi) in regions:
+            got = dd.read_global(li, glo, ghi, handle)
+            want = ripple_block(glo, ghi, dd.size, scale).astype(got.dtype)
+            if not np.array_equal(got, want):
+                bad = np.argwhere(got != want)
+                i = tuple(bad[0])
+                raise AssertionError(
+                    f"domain {li} region {glo}..{ghi}: {len(bad)} mismatches; first {i} "
+                    f"got {got[i]} want {want[i]}"
+                )
+
+
 def check_full_regions(dd, handle, scale=1.0):
     """verify every cell of every local domain's full region (incl. halos)"""
     for li in range(dd.num_local()):
