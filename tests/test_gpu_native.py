@@ -192,3 +192,37 @@ def test_halo_multiplier_native_matches_plain():
         )
     for a, b in zip(outs[0], outs[1]):
         np.testing.assert_array_equal(a, b)
+
+
+@pytest.mark.parametrize("seed", [0, 3, 5, 8])
+def test_fuzz_ripple_native(seed):
+    """randomized radii/quantities through the native engine (mirrors the
+    CPU fuzz tier)"""
+    import random
+
+    from test_fuzz_exchange import random_radius
+    from util import check_valid_regions
+
+    rng = random.Random(seed)
+    size = tuple(rng.randint(6, 24) for _ in range(3))
+    n_dom = rng.choice([1, 2, 3, 4])
+    radius = random_radius(rng)
+    max_r = max(
+        radius.dir(x, y, z) for x in (-1, 0, 1) for y in (-1, 0, 1) for z in (-1, 0, 1)
+    )
+    size = tuple(max(s, max_r * n_dom * 2 + n_dom) for s in size)
+    dd = sa.DistributedDomain(*size, backend="native")
+    dd.set_radius(radius)
+    dd.set_gpus([0] * n_dom)
+    handles = []
+    for qi in range(rng.randint(1, 3)):
+        dtype = rng.choice([np.float32, np.float64])
+        handles.append((dd.add_data(dtype, f"q{qi}"), dtype, 1.0 + qi))
+    dd.realize()
+    for h, dtype, scale in handles:
+        for li in range(dd.num_local()):
+            lo, hi = dd.local_rect(li)
+            dd.write_global(li, lo, ripple_block(lo, hi, dd.size, scale).astype(dtype), h)
+    dd.exchange()
+    for h, dtype, scale in handles:
+        check_valid_regions(dd, h, scale)
