@@ -311,6 +311,139 @@ __global__ void __launch_bounds__(256, 3) mhd_momentum_kernel(MhdParams p, MhdSc
   }
 }
 
+// LDS-tiled momentum variant (STENCIL_MHD_LDS=1): stages the 6 vector
+// fields (uu, aa) of a (32,4,2)-cell tile + radius-3 halo in LDS
+// (145.9 KB -> one block per CU) and evaluates j, B, the Lorentz force,
+// advection, viscosity and grad(div u) from LDS lines, replacing BOTH the
+// Lorentz and momentum global-line kernels. lnrho/ss pressure gradients
+// (12 loads/cell) stay on the global path.
+#define MT_X 32
+#define MT_Y 4
+#define MT_Z 2
+#define MH 3 // halo
+
+struct LdsTile {
+  // [field][z][y][x]; +6 halo cells per axis
+  double t[6][MT_Z + 6][MT_Y + 6][MT_X + 6];
+};
+
+__global__ void __launch_bounds__(256) mhd_momentum_lds_kernel(MhdParams p) {
+  __shared__ LdsTile s;
+  const int32_t gx0 = blockIdx.x * MT_X;
+  const int32_t gy0 = blockIdx.y * MT_Y;
+  const int32_t gz0 = blockIdx.z * MT_Z;
+
+  // cooperative load (fields 0..5 = UUX..AAZ, q = 1 + f)
+  const int64_t rawX = p.extX; // region extents (guards below)
+  constexpr int TILE = 6 * (MT_Z + 6) * (MT_Y + 6) * (MT_X + 6);
+  for (int i = threadIdx.x; i < TILE; i += 256) {
+    int r = i;
+    const int x = r % (MT_X + 6);
+    r /= (MT_X + 6);
+    const int y = r % (MT_Y + 6);
+    r /= (MT_Y + 6);
+    const int z = r % (MT_Z + 6);
+    const int f = r / (MT_Z + 6);
+    // global coords of this tile cell (region-local then absolute)
+    const int64_t lx = (int64_t)gx0 + x - MH;
+    const int64_t ly = (int64_t)gy0 + y - MH;
+    const int64_t lz = (int64_t)gz0 + z - MH;
+    // clamp into the full allocation (out-of-range cells are never read
+    // by an in-region output; clamping just keeps the address legal)
+    const int64_t axm = p.loX + lx - p.allocX;
+    const int64_t aym = p.loY + ly - p.allocY;
+    const int64_t azm = p.loZ + lz - p.allocZ;
+    const int64_t ax = axm < 0 ? 0 : axm;
+    const int64_t ay = aym < 0 ? 0 : aym;
+    const int64_t az = azm < 0 ? 0 : azm;
+    const char *base = p.curr[1 + f];
+    s.t[f][z][y][x] = *(const double *)(base + az * p.plane + ay * p.pitch + ax * 8);
+  }
+  __syncthreads();
+
+  const int32_t cx = threadIdx.x % MT_X;
+  const int32_t cy = (threadIdx.x / MT_X) % MT_Y;
+  const int32_t cz = threadIdx.x / (MT_X * MT_Y);
+  const int32_t lx = gx0 + cx, ly = gy0 + cy, lz = gz0 + cz;
+  if (lx >= p.extX || ly >= p.extY || lz >= p.extZ) return;
+
+  const double ix = 1.0 / p.dsx, iy = 1.0 / p.dsy, iz = 1.0 / p.dsz;
+  const int X = cx + MH, Y = cy + MH, Z = cz + MH;
+  enum { FUX = 0, FUY = 1, FUZ = 2, FAX = 3, FAY = 4, FAZ = 5 };
+
+#define LV(f, dxo, dyo, dzo) s.t[f][Z + (dzo)][Y + (dyo)][X + (dxo)]
+#define LDX(f) ((D1[0] * (LV(f, 1, 0, 0) - LV(f, -1, 0, 0)) + D1[1] * (LV(f, 2, 0, 0) - LV(f, -2, 0, 0)) + D1[2] * (LV(f, 3, 0, 0) - LV(f, -3, 0, 0))) * ix)
+#define LDY(f) ((D1[0] * (LV(f, 0, 1, 0) - LV(f, 0, -1, 0)) + D1[1] * (LV(f, 0, 2, 0) - LV(f, 0, -2, 0)) + D1[2] * (LV(f, 0, 3, 0) - LV(f, 0, -3, 0))) * iy)
+#define LDZ(f) ((D1[0] * (LV(f, 0, 0, 1) - LV(f, 0, 0, -1)) + D1[1] * (LV(f, 0, 0, 2) - LV(f, 0, 0, -2)) + D1[2] * (LV(f, 0, 0, 3) - LV(f, 0, 0, -3))) * iz)
+#define LDXX(f) ((D2[0] * LV(f, 0, 0, 0) + D2[1] * (LV(f, 1, 0, 0) + LV(f, -1, 0, 0)) + D2[2] * (LV(f, 2, 0, 0) + LV(f, -2, 0, 0)) + D2[3] * (LV(f, 3, 0, 0) + LV(f, -3, 0, 0))) * (ix * ix))
+#define LDYY(f) ((D2[0] * LV(f, 0, 0, 0) + D2[1] * (LV(f, 0, 1, 0) + LV(f, 0, -1, 0)) + D2[2] * (LV(f, 0, 2, 0) + LV(f, 0, -2, 0)) + D2[3] * (LV(f, 0, 3, 0) + LV(f, 0, -3, 0))) * (iy * iy))
+#define LDZZ(f) ((D2[0] * LV(f, 0, 0, 0) + D2[1] * (LV(f, 0, 0, 1) + LV(f, 0, 0, -1)) + D2[2] * (LV(f, 0, 0, 2) + LV(f, 0, 0, -2)) + D2[3] * (LV(f, 0, 0, 3) + LV(f, 0, 0, -3))) * (iz * iz))
+
+  auto lcross = [&](int f, int a1, int a2, double i1, double i2) {
+    // composed first-derivative quadrant sum along axes a1, a2 (0=x,1=y,2=z)
+    double acc = 0;
+#pragma unroll
+    for (int i = 1; i <= 3; ++i)
+#pragma unroll
+      for (int k = 1; k <= 3; ++k) {
+        const int dx1 = (a1 == 0) ? i : 0, dy1 = (a1 == 1) ? i : 0, dz1 = (a1 == 2) ? i : 0;
+        const int dx2 = (a2 == 0) ? k : 0, dy2 = (a2 == 1) ? k : 0, dz2 = (a2 == 2) ? k : 0;
+        acc += D1[i - 1] * D1[k - 1] *
+               (s.t[f][Z + dz1 + dz2][Y + dy1 + dy2][X + dx1 + dx2] -
+                s.t[f][Z + dz1 - dz2][Y + dy1 - dy2][X + dx1 - dx2] -
+                s.t[f][Z - dz1 + dz2][Y - dy1 + dy2][X - dx1 + dx2] +
+                s.t[f][Z - dz1 - dz2][Y - dy1 - dy2][X - dx1 - dx2]);
+      }
+    return acc * i1 * i2;
+  };
+
+  const MhdCommon c = mhd_setup(p, lx, ly, lz);
+  Stencil st;
+  st.pitch = c.pitch;
+  st.plane = c.plane;
+#pragma unroll
+  for (int q = 0; q < 8; ++q) st.base[q] = c.base[q];
+
+  const Vec3d uu = {LV(FUX, 0, 0, 0), LV(FUY, 0, 0, 0), LV(FUZ, 0, 0, 0)};
+  const double rho_inv = exp(-st.c(LNRHO));
+
+  // current and magnetic field from the A tile
+  const double jx = lcross(FAY, 0, 1, ix, iy) + lcross(FAZ, 0, 2, ix, iz) - LDYY(FAX) - LDZZ(FAX);
+  const double jy = lcross(FAX, 0, 1, ix, iy) + lcross(FAZ, 1, 2, iy, iz) - LDXX(FAY) - LDZZ(FAY);
+  const double jz = lcross(FAX, 0, 2, ix, iz) + lcross(FAY, 1, 2, iy, iz) - LDXX(FAZ) - LDYY(FAZ);
+  const Vec3d B = {LDY(FAZ) - LDZ(FAY), LDZ(FAX) - LDX(FAZ), LDX(FAY) - LDY(FAX)};
+  const Vec3d jxB = cross({jx, jy, jz}, B);
+
+  {
+    const double ugradu = uu.x * LDX(FUX) + uu.y * LDY(FUX) + uu.z * LDZ(FUX);
+    const double press = st.dx(LNRHO, ix) + p.cp_inv * st.dx(SS, ix);
+    const double graddiv = LDXX(FUX) + lcross(FUY, 0, 1, ix, iy) + lcross(FUZ, 0, 2, ix, iz);
+    const double visc = p.nu * (LDXX(FUX) + LDYY(FUX) + LDZZ(FUX) + graddiv / 3.0);
+    write_rk3(p, st, c.out[UUX], UUX, -ugradu - p.cs2 * press + rho_inv * jxB.x + visc);
+  }
+  {
+    const double ugradu = uu.x * LDX(FUY) + uu.y * LDY(FUY) + uu.z * LDZ(FUY);
+    const double press = st.dy(LNRHO, iy) + p.cp_inv * st.dy(SS, iy);
+    const double graddiv = lcross(FUX, 0, 1, ix, iy) + LDYY(FUY) + lcross(FUZ, 1, 2, iy, iz);
+    const double visc = p.nu * (LDXX(FUY) + LDYY(FUY) + LDZZ(FUY) + graddiv / 3.0);
+    write_rk3(p, st, c.out[UUY], UUY, -ugradu - p.cs2 * press + rho_inv * jxB.y + visc);
+  }
+  {
+    const double ugradu = uu.x * LDX(FUZ) + uu.y * LDY(FUZ) + uu.z * LDZ(FUZ);
+    const double press = st.dz(LNRHO, iz) + p.cp_inv * st.dz(SS, iz);
+    const double graddiv = lcross(FUX, 0, 2, ix, iz) + lcross(FUY, 1, 2, iy, iz) + LDZZ(FUZ);
+    const double visc = p.nu * (LDXX(FUZ) + LDYY(FUZ) + LDZZ(FUZ) + graddiv / 3.0);
+    write_rk3(p, st, c.out[UUZ], UUZ, -ugradu - p.cs2 * press + rho_inv * jxB.z + visc);
+  }
+}
+#undef LV
+#undef LDX
+#undef LDY
+#undef LDZ
+#undef LDXX
+#undef LDYY
+#undef LDZZ
+
 } // namespace
 
 void mhd_substep(ExchangeEngine &eng, int dom, const Rect3 &region, int step, double dt,
@@ -371,6 +504,22 @@ void mhd_substep(ExchangeEngine &eng, int dom, const Rect3 &region, int step, do
   dim3 block((uint32_t)bx, (uint32_t)by, (uint32_t)bz);
   dim3 grid((uint32_t)((ext.x + bx - 1) / bx), (uint32_t)((ext.y + by - 1) / by),
             (uint32_t)((ext.z + bz - 1) / bz));
+  static int useLds = -1;
+  if (useLds < 0) {
+    const char *e = getenv("STENCIL_MHD_LDS");
+    useLds = (e && e[0] == '1') ? 1 : 0;
+  }
+  if (useLds) {
+    hipLaunchKernelGGL(mhd_scalar_kernel, grid, block, 0, eng.compute_stream(dom, streamId), p);
+    STENCIL_HIP(hipGetLastError());
+    dim3 lblock(256, 1, 1);
+    dim3 lgrid((uint32_t)((ext.x + 31) / 32), (uint32_t)((ext.y + 3) / 4),
+               (uint32_t)((ext.z + 1) / 2));
+    hipLaunchKernelGGL(mhd_momentum_lds_kernel, lgrid, lblock, 0, eng.compute_stream(dom, streamId),
+                       p);
+    STENCIL_HIP(hipGetLastError());
+    return;
+  }
   MhdScratch sc{};
   sc.rowStride = ext.x * 8;
   sc.planeStride = sc.rowStride * ext.y;
