@@ -87,9 +87,21 @@ int pick_word(int64_t rowBytes, std::initializer_list<int64_t> alignedQuantities
 } // namespace
 
 void CopyBatch::finalize_upload() {
+  // STENCIL_AMD_COPY_DIV=N shrinks the copy grid N-fold (grid-stride
+  // covers the rest): fewer blocks = less CU contention with concurrent
+  // compute kernels at the cost of a longer exchange window
+  static int64_t div = 0;
+  if (!div) {
+    div = 1;
+    if (const char *e = getenv("STENCIL_AMD_COPY_DIV")) {
+      div = atoll(e);
+      if (div < 1) div = 1;
+    }
+  }
   prefix.assign(jobs.size() + 1, 0);
   for (size_t i = 0; i < jobs.size(); ++i) {
-    const int64_t blocks = std::max<int64_t>(1, (jobs[i].nWords + kBlock - 1) / kBlock);
+    const int64_t blocks =
+        std::max<int64_t>(1, (jobs[i].nWords + kBlock * div - 1) / (kBlock * div));
     prefix[i + 1] = prefix[i] + blocks;
   }
   nBlocks = prefix.back();
