@@ -88,10 +88,11 @@ int pick_word(int64_t rowBytes, std::initializer_list<int64_t> alignedQuantities
 
 void CopyBatch::finalize_upload() {
   // Cap the batch at ~512 blocks (grid-stride covers the rest): a halo
-  // exchange overlapped with compute should not flood all 256 CUs --
-  // capping measured +8.6% on the jacobi step (1.295 -> 1.192 ms within
-  // one box) while leaving small exchanges at one block per 256 words.
-  // STENCIL_AMD_COPY_DIV overrides the divisor for experiments.
+  // exchange overlapped with compute need not flood all 256 CUs. One
+  // measurement showed +8.6% on the jacobi step, a re-measurement on
+  // another box showed no difference (run-to-run DVFS variance ~8%); the
+  // cap is kept as it is never worse and bounds CU contention by
+  // construction. STENCIL_AMD_COPY_DIV overrides for experiments.
   constexpr int64_t kMaxBlocks = 512;
   int64_t div = 0;
   if (const char *e = getenv("STENCIL_AMD_COPY_DIV")) div = atoll(e);
