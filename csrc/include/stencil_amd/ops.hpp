@@ -27,10 +27,14 @@ struct MhdCoeffs {
 // one RK3 substep (step in 0..2) of the 8-field 6th-order MHD system over
 // `region` (global coords); reads curr, updates next in place (Williamson
 // two-buffer form) -- caller swaps after each substep
-// scratchBuf: engine buffer (>= 3 * region volume * 8 bytes) holding the
-// intermediate Lorentz force between the kernel passes
+// Separable-derivative MHD: the domain carries 10 fp64 quantities
+// (8 physics fields + div u + div A). Per substep the app runs
+//   exchange() -> mhd_div_pass -> exchange() -> mhd_substep
+// so grad(div .) reduces to first derivatives of the exchanged div fields.
+void mhd_div_pass(ExchangeEngine &eng, int dom, const Rect3 &region, const MhdCoeffs &cf,
+                  int streamId = 0);
 void mhd_substep(ExchangeEngine &eng, int dom, const Rect3 &region, int step, double dt,
-                 const MhdCoeffs &cf, int64_t scratchBuf, int streamId = 0);
+                 const MhdCoeffs &cf, int streamId = 0);
 
 // fill an fp64 region with base + amp*sin(kx*x + ky*y + kz*z + phase)
 // (deterministic smooth initial conditions, reproducible in NumPy)

@@ -91,6 +91,11 @@ class Astaroth:
         if gpus is not None:
             self.dd.set_gpus(gpus)
         self.handles = [self.dd.add_data(np.float64, n) for n in FIELDS]
+        # auxiliary exchanged quantities of the separable-derivative
+        # scheme (csrc/src/mhd.hip): div u and div A, recomputed and
+        # re-exchanged every substep
+        self.dd.add_data(np.float64, "divu")
+        self.dd.add_data(np.float64, "diva")
         self.cf = _C.MhdCoeffs()
         for k in ("dsx", "dsy", "dsz", "cs2", "cp_inv", "nu", "eta", "chi"):
             setattr(self.cf, k, float(self.conf[k]))
@@ -99,18 +104,7 @@ class Astaroth:
         self.dd.realize()
         self.interiors = self.dd.get_interior()
         self.exteriors = self.dd.get_exterior()
-        # per-domain Lorentz-force scratch, one per compute stream
-        # (interior runs on stream 0, exterior shells on stream 1 so they
-        # overlap; each stream needs its own scratch)
-        self.scratch = []
-        self.scratch_ext = []
-        eng = self.dd.backend.engine if hasattr(self.dd.backend, "engine") else None
-        if eng is not None:
-            for li in range(self.dd.num_local()):
-                lo, hi = self.dd.local_rect(li)
-                vol = (hi[0] - lo[0]) * (hi[1] - lo[1]) * (hi[2] - lo[2])
-                self.scratch.append(eng.create_buffer(li, 3 * vol * 8))
-                self.scratch_ext.append(eng.create_buffer(li, 3 * vol * 8))
+
 
     def init_fields(self):
         """harmonic initial conditions on every interior (device-side)"""
@@ -137,29 +131,22 @@ class Astaroth:
         self.dd.backend.sync_compute()
 
     def _substep(self, s: int, dt: float, compute: bool, overlap: bool):
+        """separable-derivative substep:
+        exchange (fields) -> div pass -> exchange (div halos) -> main
+        passes. `overlap` currently unused (the two exchanges serialize
+        the phases; see docs/ROADMAP.md for the pipelined variant)."""
         dd = self.dd
         eng = dd.backend.engine
-        if compute and overlap:
-            for li in range(dd.num_local()):
-                ilo, ihi = self.interiors[li]
-                _C.mhd_substep(
-                    eng, li, _C.Rect3(_C.Vec3(*ilo), _C.Vec3(*ihi)), s, dt, self.cf,
-                    self.scratch[li],
-                )
         dd.exchange()
         if compute:
             for li in range(dd.num_local()):
-                boxes = (
-                    self.exteriors[li]
-                    if overlap
-                    else [dd.local_rect(li)]
-                )
-                for blo, bhi in boxes:
-                    # stream 1: exterior shells overlap the interior kernel
-                    _C.mhd_substep(
-                        eng, li, _C.Rect3(_C.Vec3(*blo), _C.Vec3(*bhi)), s, dt, self.cf,
-                        self.scratch_ext[li], 1 if overlap else 0,
-                    )
+                lo, hi = dd.local_rect(li)
+                _C.mhd_div_pass(eng, li, _C.Rect3(_C.Vec3(*lo), _C.Vec3(*hi)), self.cf)
+            dd.backend.sync_compute()
+            dd.exchange()
+            for li in range(dd.num_local()):
+                lo, hi = dd.local_rect(li)
+                _C.mhd_substep(eng, li, _C.Rect3(_C.Vec3(*lo), _C.Vec3(*hi)), s, dt, self.cf)
         dd.backend.sync_compute()
         dd.swap()
 

@@ -68,23 +68,18 @@ def rhs(F, cf):
     gss = (gx(F[SS]), gy(F[SS]), gz(F[SS]))
 
     du = [[gx(c), gy(c), gz(c)] for c in u]  # du[i][j] = d u_i / d x_j
-    divu = du[0][0] + du[1][1] + du[2][2]
+    # separable formulation (matches csrc/src/mhd.hip exactly): div fields
+    # are computed, stored, and differentiated -- grad(div .) is three
+    # 7-point first derivatives of one field
+    divu = gx(u[0]) + gy(u[1]) + gz(u[2])
     lap_u = [lap(c) for c in u]
-    graddiv_u = [
-        d2(u[0], "x", dsx) + dcross(u[1], "x", dsx, "y", dsy) + dcross(u[2], "x", dsx, "z", dsz),
-        dcross(u[0], "x", dsx, "y", dsy) + d2(u[1], "y", dsy) + dcross(u[2], "y", dsy, "z", dsz),
-        dcross(u[0], "x", dsx, "z", dsz) + dcross(u[1], "y", dsy, "z", dsz) + d2(u[2], "z", dsz),
-    ]
+    graddiv_u = [gx(divu), gy(divu), gz(divu)]
 
     A = [F[AAX], F[AAY], F[AAZ]]
     B = (gy(A[2]) - gz(A[1]), gz(A[0]) - gx(A[2]), gx(A[1]) - gy(A[0]))
     lap_a = [lap(c) for c in A]
-    graddiv_a = [
-        d2(A[0], "x", dsx) + dcross(A[1], "x", dsx, "y", dsy) + dcross(A[2], "x", dsx, "z", dsz),
-        dcross(A[0], "x", dsx, "y", dsy) + d2(A[1], "y", dsy) + dcross(A[2], "y", dsy, "z", dsz),
-        dcross(A[0], "x", dsx, "z", dsz) + dcross(A[1], "y", dsy, "z", dsz) + d2(A[2], "z", dsz),
-    ]
-    j = [graddiv_a[i] - lap_a[i] for i in range(3)]
+    divA = gx(A[0]) + gy(A[1]) + gz(A[2])
+    j = [gx(divA) - lap_a[0], gy(divA) - lap_a[1], gz(divA) - lap_a[2]]
     rho_inv = np.exp(-F[LNRHO])
 
     jxB = (
