@@ -31,12 +31,10 @@ struct MhdCoeffs {
 // (8 physics fields + div u + div A). Per substep the app runs
 //   exchange() -> mhd_div_pass -> exchange() -> mhd_substep
 // so grad(div .) reduces to first derivatives of the exchanged div fields.
-// scratchBuf: engine buffer >= 3 * region volume * 8 bytes holding the
-// pointwise B = curl(A) between the passes (computed once per substep)
 void mhd_div_pass(ExchangeEngine &eng, int dom, const Rect3 &region, const MhdCoeffs &cf,
-                  int64_t scratchBuf, int streamId = 0);
+                  int streamId = 0);
 void mhd_substep(ExchangeEngine &eng, int dom, const Rect3 &region, int step, double dt,
-                 const MhdCoeffs &cf, int64_t scratchBuf, int streamId = 0);
+                 const MhdCoeffs &cf, int streamId = 0);
 
 // fill an fp64 region with base + amp*sin(kx*x + ky*y + kz*z + phase)
 // (deterministic smooth initial conditions, reproducible in NumPy)

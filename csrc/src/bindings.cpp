@@ -270,9 +270,9 @@ PYBIND11_MODULE(_C, m) {
       .def_readwrite("eta", &MhdCoeffs::eta)
       .def_readwrite("chi", &MhdCoeffs::chi);
   m.def("mhd_substep", &mhd_substep, py::arg("eng"), py::arg("dom"), py::arg("region"),
-        py::arg("step"), py::arg("dt"), py::arg("cf"), py::arg("scratch"), py::arg("stream_id") = 0);
+        py::arg("step"), py::arg("dt"), py::arg("cf"), py::arg("stream_id") = 0);
   m.def("mhd_div_pass", &mhd_div_pass, py::arg("eng"), py::arg("dom"), py::arg("region"),
-        py::arg("cf"), py::arg("scratch"), py::arg("stream_id") = 0);
+        py::arg("cf"), py::arg("stream_id") = 0);
   m.def("init_harmonic_f64", &init_harmonic_f64);
   py::class_<FieldStats>(m, "FieldStats")
       .def_readonly("min", &FieldStats::min)

@@ -189,16 +189,9 @@ __device__ __forceinline__ void write_rk3(const MhdParams &p, const Stencil &st,
   *(double *)out = cur + p.beta * (p.alpha_over_beta_prev * (cur - prev) + p.dt * r);
 }
 
-struct MhdB {
-  char *ptr;           // 3 consecutive (z,y,x) fp64 arrays over the region
-  int64_t rowStride;   // extX * 8
-  int64_t planeStride; // rowStride * extY
-  int64_t compStride;  // planeStride * extZ
-};
-
 // kernel 1: continuity + entropy + induction (lnrho, ss, aa). First and
 // second derivatives only, no cross terms.
-__global__ void __launch_bounds__(256) mhd_scalar_kernel(MhdParams p, MhdB sb) {
+__global__ void __launch_bounds__(256) mhd_scalar_kernel(MhdParams p) {
   const int32_t lx = blockIdx.x * blockDim.x + threadIdx.x;
   const int32_t ly = blockIdx.y * blockDim.y + threadIdx.y;
   const int32_t lz = blockIdx.z * blockDim.z + threadIdx.z;
@@ -219,10 +212,8 @@ __global__ void __launch_bounds__(256) mhd_scalar_kernel(MhdParams p, MhdB sb) {
     const Vec3d gss = st.grad(SS, ix, iy, iz);
     write_rk3(p, st, c.out[SS], SS, -dot(uu, gss) + p.chi * st.lap(SS, ix, iy, iz));
   }
-  const char *bp =
-      sb.ptr + (int64_t)lz * sb.planeStride + (int64_t)ly * sb.rowStride + (int64_t)lx * 8;
-  const Vec3d B = {*(const double *)bp, *(const double *)(bp + sb.compStride),
-                   *(const double *)(bp + 2 * sb.compStride)};
+  const Vec3d B = {st.dy(AAZ, iy) - st.dz(AAY, iz), st.dz(AAX, iz) - st.dx(AAZ, ix),
+                   st.dx(AAY, ix) - st.dy(AAX, iy)};
   const Vec3d uxB = cross(uu, B);
   write_rk3(p, st, c.out[AAX], AAX, uxB.x + p.eta * st.lap(AAX, ix, iy, iz));
   write_rk3(p, st, c.out[AAY], AAY, uxB.y + p.eta * st.lap(AAY, ix, iy, iz));
@@ -238,7 +229,7 @@ __global__ void __launch_bounds__(256) mhd_scalar_kernel(MhdParams p, MhdB sb) {
 // composed cross-derivative sums (separable formulation; identical order
 // of accuracy, fp-rounding-level difference mirrored exactly in the NumPy
 // reference).
-__global__ void __launch_bounds__(256) mhd_div_kernel(MhdParams p, MhdB sb) {
+__global__ void __launch_bounds__(256) mhd_div_kernel(MhdParams p) {
   const int32_t lx = blockIdx.x * blockDim.x + threadIdx.x;
   const int32_t ly = blockIdx.y * blockDim.y + threadIdx.y;
   const int32_t lz = blockIdx.z * blockDim.z + threadIdx.z;
@@ -256,17 +247,10 @@ __global__ void __launch_bounds__(256) mhd_div_kernel(MhdParams p, MhdB sb) {
       st.dx(UUX, ix) + st.dy(UUY, iy) + st.dz(UUZ, iz);
   *(double *)(const_cast<char *>(c.base[DIVA])) =
       st.dx(AAX, ix) + st.dy(AAY, iy) + st.dz(AAZ, iz);
-  // B = curl(A) is consumed POINTWISE by induction (u x B) and momentum
-  // (j x B) -- no derivatives of B anywhere -- so it needs no halo and
-  // lives in plain scratch, computed once instead of twice per substep
-  char *bp = sb.ptr + (int64_t)lz * sb.planeStride + (int64_t)ly * sb.rowStride + (int64_t)lx * 8;
-  __builtin_nontemporal_store(st.dy(AAZ, iy) - st.dz(AAY, iz), (double *)bp);
-  __builtin_nontemporal_store(st.dz(AAX, iz) - st.dx(AAZ, ix), (double *)(bp + sb.compStride));
-  __builtin_nontemporal_store(st.dx(AAY, ix) - st.dy(AAX, iy), (double *)(bp + 2 * sb.compStride));
 }
 
 // momentum: j_i = D_i(divA) - lap(A_i); graddiv u = grad(divA... grad(divu)
-__global__ void __launch_bounds__(256) mhd_momentum_kernel(MhdParams p, MhdB sb) {
+__global__ void __launch_bounds__(256) mhd_momentum_kernel(MhdParams p) {
   const int32_t lx = blockIdx.x * blockDim.x + threadIdx.x;
   const int32_t ly = blockIdx.y * blockDim.y + threadIdx.y;
   const int32_t lz = blockIdx.z * blockDim.z + threadIdx.z;
@@ -284,10 +268,8 @@ __global__ void __launch_bounds__(256) mhd_momentum_kernel(MhdParams p, MhdB sb)
   const Vec3d j = {st.dx(DIVA, ix) - st.lap(AAX, ix, iy, iz),
                    st.dy(DIVA, iy) - st.lap(AAY, ix, iy, iz),
                    st.dz(DIVA, iz) - st.lap(AAZ, ix, iy, iz)};
-  const char *bp =
-      sb.ptr + (int64_t)lz * sb.planeStride + (int64_t)ly * sb.rowStride + (int64_t)lx * 8;
-  const Vec3d B = {*(const double *)bp, *(const double *)(bp + sb.compStride),
-                   *(const double *)(bp + 2 * sb.compStride)};
+  const Vec3d B = {st.dy(AAZ, iy) - st.dz(AAY, iz), st.dz(AAX, iz) - st.dx(AAZ, ix),
+                   st.dx(AAY, ix) - st.dy(AAX, iy)};
   const Vec3d jxB = cross(j, B);
 
   {
@@ -361,36 +343,23 @@ static dim3 mhd_block() {
   return dim3((uint32_t)bx, (uint32_t)by, (uint32_t)bz);
 }
 
-static MhdB mhd_b_scratch(ExchangeEngine &eng, const Rect3 &region, int64_t scratchBuf) {
-  const Vec3 ext = region.extent();
-  MhdB sb{};
-  sb.rowStride = ext.x * 8;
-  sb.planeStride = sb.rowStride * ext.y;
-  sb.compStride = sb.planeStride * ext.z;
-  if (eng.buffer_bytes(scratchBuf) < 3 * sb.compStride)
-    throw std::runtime_error("mhd: B scratch buffer too small for region");
-  sb.ptr = (char *)eng.buffer_ptr(scratchBuf);
-  return sb;
-}
-
 void mhd_div_pass(ExchangeEngine &eng, int dom, const Rect3 &region, const MhdCoeffs &cf,
-                  int64_t scratchBuf, int streamId) {
+                  int streamId) {
   LocalDomain &d = eng.domain(dom);
   const Vec3 ext = region.extent();
   if (ext.flatten() <= 0) return;
   MhdParams p{};
   mhd_fill_params(d, region, cf, p);
-  const MhdB sb = mhd_b_scratch(eng, region, scratchBuf);
   STENCIL_HIP(hipSetDevice(d.gpu()));
   dim3 block = mhd_block();
   dim3 grid((uint32_t)((ext.x + block.x - 1) / block.x), (uint32_t)((ext.y + block.y - 1) / block.y),
             (uint32_t)((ext.z + block.z - 1) / block.z));
-  hipLaunchKernelGGL(mhd_div_kernel, grid, block, 0, eng.compute_stream(dom, streamId), p, sb);
+  hipLaunchKernelGGL(mhd_div_kernel, grid, block, 0, eng.compute_stream(dom, streamId), p);
   STENCIL_HIP(hipGetLastError());
 }
 
 void mhd_substep(ExchangeEngine &eng, int dom, const Rect3 &region, int step, double dt,
-                 const MhdCoeffs &cf, int64_t scratchBuf, int streamId) {
+                 const MhdCoeffs &cf, int streamId) {
   LocalDomain &d = eng.domain(dom);
   const Vec3 ext = region.extent();
   if (ext.flatten() <= 0) return;
@@ -406,10 +375,9 @@ void mhd_substep(ExchangeEngine &eng, int dom, const Rect3 &region, int step, do
   dim3 block = mhd_block();
   dim3 grid((uint32_t)((ext.x + block.x - 1) / block.x), (uint32_t)((ext.y + block.y - 1) / block.y),
             (uint32_t)((ext.z + block.z - 1) / block.z));
-  const MhdB sb = mhd_b_scratch(eng, region, scratchBuf);
-  hipLaunchKernelGGL(mhd_scalar_kernel, grid, block, 0, eng.compute_stream(dom, streamId), p, sb);
+  hipLaunchKernelGGL(mhd_scalar_kernel, grid, block, 0, eng.compute_stream(dom, streamId), p);
   STENCIL_HIP(hipGetLastError());
-  hipLaunchKernelGGL(mhd_momentum_kernel, grid, block, 0, eng.compute_stream(dom, streamId), p, sb);
+  hipLaunchKernelGGL(mhd_momentum_kernel, grid, block, 0, eng.compute_stream(dom, streamId), p);
   STENCIL_HIP(hipGetLastError());
 }
 

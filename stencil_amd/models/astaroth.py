@@ -104,14 +104,6 @@ class Astaroth:
         self.dd.realize()
         self.interiors = self.dd.get_interior()
         self.exteriors = self.dd.get_exterior()
-        # per-domain scratch for the pointwise B = curl(A) field
-        self.scratch = []
-        eng = self.dd.backend.engine if hasattr(self.dd.backend, "engine") else None
-        if eng is not None:
-            for li in range(self.dd.num_local()):
-                lo, hi = self.dd.local_rect(li)
-                vol = (hi[0] - lo[0]) * (hi[1] - lo[1]) * (hi[2] - lo[2])
-                self.scratch.append(eng.create_buffer(li, 3 * vol * 8))
 
 
     def init_fields(self):
@@ -149,16 +141,12 @@ class Astaroth:
         if compute:
             for li in range(dd.num_local()):
                 lo, hi = dd.local_rect(li)
-                _C.mhd_div_pass(
-                    eng, li, _C.Rect3(_C.Vec3(*lo), _C.Vec3(*hi)), self.cf, self.scratch[li]
-                )
+                _C.mhd_div_pass(eng, li, _C.Rect3(_C.Vec3(*lo), _C.Vec3(*hi)), self.cf)
             dd.backend.sync_compute()
             dd.exchange()
             for li in range(dd.num_local()):
                 lo, hi = dd.local_rect(li)
-                _C.mhd_substep(
-                    eng, li, _C.Rect3(_C.Vec3(*lo), _C.Vec3(*hi)), s, dt, self.cf, self.scratch[li]
-                )
+                _C.mhd_substep(eng, li, _C.Rect3(_C.Vec3(*lo), _C.Vec3(*hi)), s, dt, self.cf)
         dd.backend.sync_compute()
         dd.swap()
 
