@@ -131,22 +131,51 @@ class Astaroth:
         self.dd.backend.sync_compute()
 
     def _substep(self, s: int, dt: float, compute: bool, overlap: bool):
-        """separable-derivative substep:
-        exchange (fields) -> div pass -> exchange (div halos) -> main
-        passes. `overlap` currently unused (the two exchanges serialize
-        the phases; see docs/ROADMAP.md for the pipelined variant)."""
+        """separable-derivative substep with two exchanges, both hidden
+        under compute when `overlap`:
+
+          div(interior)            || exchange #1 (field halos)
+          div(exterior shells)
+          main(interior)           || exchange #2 (div halos)
+          main(exterior shells)
+
+        Safe because interior cells read only +-3 neighborhoods inside the
+        compute region (never halos), and the cells an exchange sends (the
+        outermost radius-deep shells) lie in the exterior region, which is
+        only written after the exchange that could read it completes."""
         dd = self.dd
         eng = dd.backend.engine
-        dd.exchange()
-        if compute:
+
+        def rect(lo, hi):
+            return _C.Rect3(_C.Vec3(*lo), _C.Vec3(*hi))
+
+        if not compute:
+            dd.exchange()
+            dd.backend.sync_compute()
+            dd.swap()
+            return
+        if overlap:
             for li in range(dd.num_local()):
-                lo, hi = dd.local_rect(li)
-                _C.mhd_div_pass(eng, li, _C.Rect3(_C.Vec3(*lo), _C.Vec3(*hi)), self.cf)
+                _C.mhd_div_pass(eng, li, rect(*self.interiors[li]), self.cf)
+            dd.exchange()  # X1 overlaps div(interior)
+            for li in range(dd.num_local()):
+                for box in self.exteriors[li]:
+                    _C.mhd_div_pass(eng, li, rect(*box), self.cf)
+            dd.backend.sync_compute()  # div complete before X2 reads its edges
+            for li in range(dd.num_local()):
+                _C.mhd_substep(eng, li, rect(*self.interiors[li]), s, dt, self.cf)
+            dd.exchange()  # X2 overlaps main(interior)
+            for li in range(dd.num_local()):
+                for box in self.exteriors[li]:
+                    _C.mhd_substep(eng, li, rect(*box), s, dt, self.cf)
+        else:
+            dd.exchange()
+            for li in range(dd.num_local()):
+                _C.mhd_div_pass(eng, li, rect(*dd.local_rect(li)), self.cf)
             dd.backend.sync_compute()
             dd.exchange()
             for li in range(dd.num_local()):
-                lo, hi = dd.local_rect(li)
-                _C.mhd_substep(eng, li, _C.Rect3(_C.Vec3(*lo), _C.Vec3(*hi)), s, dt, self.cf)
+                _C.mhd_substep(eng, li, rect(*dd.local_rect(li)), s, dt, self.cf)
         dd.backend.sync_compute()
         dd.swap()
 
