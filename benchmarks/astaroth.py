@@ -60,7 +60,8 @@ def main():
 
             dist.barrier()
         t0 = time.perf_counter()
-        app.step(compute=not args.no_compute, overlap=not args.no_overlap)
+        app.step(compute=not args.no_compute,
+                 overlap=False if args.no_overlap else None)
         dt = time.perf_counter() - t0
         if i >= args.warmup:
             stats.insert(dt)

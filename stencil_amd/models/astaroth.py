@@ -179,7 +179,14 @@ class Astaroth:
         dd.backend.sync_compute()
         dd.swap()
 
-    def step(self, dt: Optional[float] = None, compute: bool = True, overlap: bool = True):
+    def step(self, dt: Optional[float] = None, compute: bool = True,
+             overlap: Optional[bool] = None):
+        """overlap=None auto-selects: the interior/exterior split pays only
+        when the exchange is worth hiding (multi-rank); at world=1 the
+        extra thin-slab launches cost more than the 0.2 ms exchange
+        (measured 12.7 vs 10.8 ms/iter at 256^3)"""
+        if overlap is None:
+            overlap = self.dd.comm.world_size > 1
         dt = self.conf["dt"] if dt is None else dt
         for s in range(3):
             self._substep(s, dt, compute, overlap)
