@@ -327,16 +327,17 @@ static void mhd_fill_params(LocalDomain &d, const Rect3 &region, const MhdCoeffs
 }
 
 static dim3 mhd_block() {
-  // 32x4x2 measured best on gfx950 (block sweep in gpurun_out/gpu6.log)
+  // 64x2x2 measured best after the separable-derivative restructure
+  // (sweep in gpurun gpu15 log: 1649 vs 1608 Mcell/s at 32x4x2)
   static int bx = 0, by = 0, bz = 0;
   if (!bx) {
-    bx = 32;
-    by = 4;
+    bx = 64;
+    by = 2;
     bz = 2;
     if (const char *e = getenv("STENCIL_MHD_BLOCK"))
       if (sscanf(e, "%dx%dx%d", &bx, &by, &bz) != 3 || bx * by * bz != 256) {
-        bx = 32;
-        by = 4;
+        bx = 64;
+        by = 2;
         bz = 2;
       }
   }
