@@ -376,9 +376,13 @@ void mhd_substep(ExchangeEngine &eng, int dom, const Rect3 &region, int step, do
   dim3 block = mhd_block();
   dim3 grid((uint32_t)((ext.x + block.x - 1) / block.x), (uint32_t)((ext.y + block.y - 1) / block.y),
             (uint32_t)((ext.z + block.z - 1) / block.z));
+  // scalar (writes lnrho/ss/aa) and momentum (writes uu) touch disjoint
+  // outputs and only read shared inputs: run them CONCURRENTLY on the two
+  // compute streams (the caller's sync_compute joins both)
   hipLaunchKernelGGL(mhd_scalar_kernel, grid, block, 0, eng.compute_stream(dom, streamId), p);
   STENCIL_HIP(hipGetLastError());
-  hipLaunchKernelGGL(mhd_momentum_kernel, grid, block, 0, eng.compute_stream(dom, streamId), p);
+  hipLaunchKernelGGL(mhd_momentum_kernel, grid, block, 0, eng.compute_stream(dom, 1 - streamId),
+                     p);
   STENCIL_HIP(hipGetLastError());
 }
 
