@@ -10,7 +10,7 @@ One rank per GPU (driver): python -m torch.distributed.run --nnodes=1
 import argparse
 import json
 import os
-import sys
+
 import time
 
 

@@ -7,7 +7,7 @@ implementation of standard compressible resistive MHD (csrc/src/mhd.hip).
 from __future__ import annotations
 
 import math
-import time
+
 from typing import Dict, List, Optional
 
 import numpy as np

@@ -11,7 +11,7 @@ from typing import List, Optional
 
 import numpy as np
 
-from ..core import DistributedDomain, Method
+from ..core import DistributedDomain
 from ..parallel.placement import PlacementStrategy
 
 

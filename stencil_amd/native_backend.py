@@ -6,7 +6,7 @@ buffers exported via DLPack and sent with torch.distributed (RCCL) P2P.
 """
 from __future__ import annotations
 
-from typing import Dict, List, Optional, Tuple
+from typing import List, Optional, Tuple
 
 from . import _C
 from .parallel.planning import ExchangePlan, p2p_tag, wire_layout

@@ -8,7 +8,7 @@ Not a performance path.
 """
 from __future__ import annotations
 
-from typing import List, Tuple
+from typing import Tuple
 
 import torch
 
