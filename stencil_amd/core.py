@@ -243,6 +243,26 @@ class DistributedDomain:
         self.time_swap += time.perf_counter() - t0
 
     # ---- geometry queries ----
+    def any_methods(self, m: Method) -> bool:
+        """True if any of the given transport methods are enabled
+        (reference stencil.hpp:150)"""
+        return bool(self.methods & m)
+
+    def get_origin(self, li: int) -> Vec:
+        """global coordinate of local subdomain li's interior origin
+        (reference stencil.hpp:162)"""
+        return self.local_rect(li)[0]
+
+    def get_topology(self):
+        """periodic neighbor topology over the partition grid
+        (reference stencil.hpp:203)"""
+        from .parallel.topology import Topology
+
+        return Topology(self.placement.dim())
+
+    def get_placement(self):
+        return self.placement
+
     def num_local(self) -> int:
         return self.placement.num_local(self.comm.rank)
 

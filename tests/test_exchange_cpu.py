@@ -179,3 +179,24 @@ def test_halo_multiplier_matches_plain(m):
         )
     for a, b in zip(outs[0], outs[1]):
         np.testing.assert_array_equal(a, b)
+
+
+def test_halo_multiplier_with_radius2():
+    from stencil_amd.models.jacobi3d import Jacobi3D
+
+    size = (20, 16, 12)
+    outs = []
+    for mult in (1, 2):
+        app = Jacobi3D(size, backend="torch", gpus=[0, 0], radius=2, halo_multiplier=mult)
+        app.realize()
+        fill_interiors(app.dd, app.h)
+        for _ in range(4):
+            app.step()
+        outs.append(
+            [
+                app.dd.read_global(li, *app.dd.local_rect(li), app.h)
+                for li in range(app.dd.num_local())
+            ]
+        )
+    for a, b in zip(outs[0], outs[1]):
+        np.testing.assert_array_equal(a, b)
