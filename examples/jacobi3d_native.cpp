@@ -49,16 +49,18 @@ int main(int argc, char **argv) {
     jacobi_step(eng, 0, q, interior, compute); // overlaps with...
     eng.launch_translates();                   // ...the halo exchange
     eng.sync_translates();
-    // exterior shells (slide faces in; reference src/stencil.cu:927-977)
+    // exterior shells on the second compute stream: they depend only on
+    // the (synced) exchange, so they overlap the interior kernel
+    // (slide faces in; reference src/stencil.cu:927-977)
     Rect3 c = compute;
     for (int axis = 0; axis < 3; ++axis) {
       Rect3 s = c;
       s.lo[axis] = interior.hi[axis];
-      jacobi_step(eng, 0, q, s, compute);
+      jacobi_step(eng, 0, q, s, compute, /*streamId=*/1);
       c.hi[axis] = interior.hi[axis];
       Rect3 t = c;
       t.hi[axis] = interior.lo[axis];
-      jacobi_step(eng, 0, q, t, compute);
+      jacobi_step(eng, 0, q, t, compute, /*streamId=*/1);
       c.lo[axis] = interior.lo[axis];
     }
     eng.sync_compute();
