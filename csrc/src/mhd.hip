@@ -54,6 +54,8 @@ struct MhdParams {
   double nu, eta, chi;
   double alpha_over_beta_prev; // alpha_s / beta_{s-1} (0 for substep 0)
   double beta;
+  int32_t swizzle; // XCD-aware block remap (see xcd_remap)
+  int32_t pad_;
 };
 
 enum { LNRHO = 0, UUX = 1, UUY = 2, UUZ = 3, AAX = 4, AAY = 5, AAZ = 6, SS = 7, DIVU = 8, DIVA = 9 };
@@ -165,6 +167,23 @@ struct MhdCommon {
   char *out[10];
 };
 
+// XCD-aware block remap (STENCIL_MHD_SWIZZLE=1): gives each XCD a
+// contiguous slab of the block grid. MEASURED 10% SLOWER on the MHD
+// kernels (10.05 -> 11.12 ms/iter, reproducible within-box) despite
+// their 50.8% L2 hit-rate (profiles/): the default round-robin dispatch
+// evidently spreads HBM channel traffic better than slab locality helps.
+// Kept off by default as an experiment hook.
+__device__ __forceinline__ void xcd_remap(int32_t &bx, int32_t &by, int32_t &bz) {
+  const int32_t nwg = gridDim.x * gridDim.y * gridDim.z;
+  const int32_t flat = bx + gridDim.x * (by + gridDim.y * bz);
+  const int32_t q = nwg / 8, r = nwg % 8;
+  const int32_t xcd = flat % 8, idx = flat / 8;
+  const int32_t nf = xcd < r ? xcd * (q + 1) + idx : r * (q + 1) + (xcd - r) * q + idx;
+  bx = nf % gridDim.x;
+  by = (nf / gridDim.x) % gridDim.y;
+  bz = nf / (gridDim.x * gridDim.y);
+}
+
 __device__ __forceinline__ MhdCommon mhd_setup(const MhdParams &p, int32_t lx, int32_t ly,
                                                int32_t lz) {
   const int64_t ax = p.loX + lx - p.allocX;
@@ -192,9 +211,11 @@ __device__ __forceinline__ void write_rk3(const MhdParams &p, const Stencil &st,
 // kernel 1: continuity + entropy + induction (lnrho, ss, aa). First and
 // second derivatives only, no cross terms.
 __global__ void __launch_bounds__(256) mhd_scalar_kernel(MhdParams p) {
-  const int32_t lx = blockIdx.x * blockDim.x + threadIdx.x;
-  const int32_t ly = blockIdx.y * blockDim.y + threadIdx.y;
-  const int32_t lz = blockIdx.z * blockDim.z + threadIdx.z;
+  int32_t bx = blockIdx.x, by = blockIdx.y, bz = blockIdx.z;
+  if (p.swizzle) xcd_remap(bx, by, bz);
+  const int32_t lx = bx * blockDim.x + threadIdx.x;
+  const int32_t ly = by * blockDim.y + threadIdx.y;
+  const int32_t lz = bz * blockDim.z + threadIdx.z;
   if (lx >= p.extX || ly >= p.extY || lz >= p.extZ) return;
   const MhdCommon c = mhd_setup(p, lx, ly, lz);
   Stencil st;
@@ -230,9 +251,11 @@ __global__ void __launch_bounds__(256) mhd_scalar_kernel(MhdParams p) {
 // of accuracy, fp-rounding-level difference mirrored exactly in the NumPy
 // reference).
 __global__ void __launch_bounds__(256) mhd_div_kernel(MhdParams p) {
-  const int32_t lx = blockIdx.x * blockDim.x + threadIdx.x;
-  const int32_t ly = blockIdx.y * blockDim.y + threadIdx.y;
-  const int32_t lz = blockIdx.z * blockDim.z + threadIdx.z;
+  int32_t bx = blockIdx.x, by = blockIdx.y, bz = blockIdx.z;
+  if (p.swizzle) xcd_remap(bx, by, bz);
+  const int32_t lx = bx * blockDim.x + threadIdx.x;
+  const int32_t ly = by * blockDim.y + threadIdx.y;
+  const int32_t lz = bz * blockDim.z + threadIdx.z;
   if (lx >= p.extX || ly >= p.extY || lz >= p.extZ) return;
   const MhdCommon c = mhd_setup(p, lx, ly, lz);
   Stencil st;
@@ -251,9 +274,11 @@ __global__ void __launch_bounds__(256) mhd_div_kernel(MhdParams p) {
 
 // momentum: j_i = D_i(divA) - lap(A_i); graddiv u = grad(divA... grad(divu)
 __global__ void __launch_bounds__(256) mhd_momentum_kernel(MhdParams p) {
-  const int32_t lx = blockIdx.x * blockDim.x + threadIdx.x;
-  const int32_t ly = blockIdx.y * blockDim.y + threadIdx.y;
-  const int32_t lz = blockIdx.z * blockDim.z + threadIdx.z;
+  int32_t bx = blockIdx.x, by = blockIdx.y, bz = blockIdx.z;
+  if (p.swizzle) xcd_remap(bx, by, bz);
+  const int32_t lx = bx * blockDim.x + threadIdx.x;
+  const int32_t ly = by * blockDim.y + threadIdx.y;
+  const int32_t lz = bz * blockDim.z + threadIdx.z;
   if (lx >= p.extX || ly >= p.extY || lz >= p.extZ) return;
   const MhdCommon c = mhd_setup(p, lx, ly, lz);
   Stencil st;
@@ -324,6 +349,12 @@ static void mhd_fill_params(LocalDomain &d, const Rect3 &region, const MhdCoeffs
   p.nu = cf.nu;
   p.eta = cf.eta;
   p.chi = cf.chi;
+  static int swz = -1;
+  if (swz < 0) {
+    const char *e = getenv("STENCIL_MHD_SWIZZLE");
+    swz = (e && e[0] == '1') ? 1 : 0;
+  }
+  p.swizzle = swz;
 }
 
 static dim3 mhd_block() {
