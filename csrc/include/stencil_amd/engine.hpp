@@ -74,9 +74,11 @@ public:
 
   //// plan-time registration (positions in allocation coordinates)
 
-  // direct-write copy of one region for ALL quantities:
-  // srcDom's curr -> dstDom's curr (possibly a peer GPU over xGMI)
-  void add_translate(int srcDom, int dstDom, const Vec3 &srcPos, const Vec3 &dstPos, const Vec3 &ext);
+  // direct-write copy of one region: srcDom's curr -> dstDom's curr
+  // (possibly a peer GPU over xGMI). `qis` selects quantities (empty =
+  // all); `group` selects the launch group (quantity-group exchanges).
+  void add_translate(int srcDom, int dstDom, const Vec3 &srcPos, const Vec3 &dstPos, const Vec3 &ext,
+                     int group = 0, const std::vector<int64_t> &qis = {});
 
   //// cross-process direct writes (HIP IPC over xGMI; the MI355X analog of
   //// the reference's tx_colocated direct-access senders)
@@ -88,9 +90,9 @@ public:
                              const std::vector<int64_t> &pitches, const std::vector<int64_t> &ysizes,
                              const std::vector<int64_t> &elemSizes,
                              const std::vector<int64_t> &pads);
-  // direct-write one region (all quantities) into a remote view's curr
+  // direct-write one region into a remote view's curr
   void add_translate_view(int srcDom, int64_t view, const Vec3 &srcPos, const Vec3 &dstPos,
-                          const Vec3 &ext);
+                          const Vec3 &ext, int group = 0, const std::vector<int64_t> &qis = {});
   // swap every view's curr/next pointer sets (call in lockstep with
   // LocalDomain::swap on every rank)
   void flip_views();
@@ -114,7 +116,7 @@ public:
   void finalize();
 
   //// per-exchange execution (stream-ordered; host-sync via the sync_* calls)
-  void launch_translates();
+  void launch_translates(int group = 0);
   void launch_packs(int group = 0);
   void launch_unpacks(int group = 0);
   void sync_translates();
@@ -147,6 +149,8 @@ private:
   struct TranslateSpec {
     int srcDom, dstDom; // dstDom: local domain index, or remote view id
     bool dstIsView = false;
+    int group = 0;
+    std::vector<int64_t> qis; // empty = all quantities
     Vec3 srcPos, dstPos, ext;
   };
   struct RemoteView {
@@ -180,8 +184,10 @@ private:
   std::vector<hipStream_t> computeStreams_;  // per domain, stream 0
   std::vector<hipStream_t> computeStreams2_; // per domain, stream 1
 
-  std::vector<CopyBatch> translateBatches_; // one per device with jobs
-  static constexpr int kGroups = 3; // 0 = wire, 1/2 = colo staging parity
+  // pack/unpack launch-group encoding for exchange-group g:
+  //   wire = 3g, colo staging parity 0/1 = 3g+1 / 3g+2
+  static constexpr int kGroups = 12; // supports 4 exchange groups
+  std::vector<CopyBatch> translateBatches_[kGroups];
   std::vector<CopyBatch> packBatches_[kGroups];
   std::vector<CopyBatch> unpackBatches_[kGroups];
   bool finalized_ = false;

@@ -199,7 +199,9 @@ PYBIND11_MODULE(_C, m) {
       .def(py::init<std::vector<std::shared_ptr<LocalDomain>>>())
       .def("enable_peer_all", &ExchangeEngine::enable_peer_all)
       .def_static("can_access_peer", &ExchangeEngine::can_access_peer)
-      .def("add_translate", &ExchangeEngine::add_translate)
+      .def("add_translate", &ExchangeEngine::add_translate, py::arg("src"), py::arg("dst"),
+           py::arg("src_pos"), py::arg("dst_pos"), py::arg("ext"), py::arg("group") = 0,
+           py::arg("qis") = std::vector<int64_t>{})
       .def("create_remote_view",
            [](ExchangeEngine &e, int openDev, const std::vector<py::bytes> &cur,
               const std::vector<py::bytes> &nxt, const std::vector<int64_t> &pitches,
@@ -210,7 +212,9 @@ PYBIND11_MODULE(_C, m) {
              for (auto &b : nxt) n.push_back(b);
              return e.create_remote_view(openDev, c, n, pitches, ysizes, es, pads);
            })
-      .def("add_translate_view", &ExchangeEngine::add_translate_view)
+      .def("add_translate_view", &ExchangeEngine::add_translate_view, py::arg("src"),
+           py::arg("view"), py::arg("src_pos"), py::arg("dst_pos"), py::arg("ext"),
+           py::arg("group") = 0, py::arg("qis") = std::vector<int64_t>{})
       .def("flip_views", &ExchangeEngine::flip_views)
       .def("create_buffer", &ExchangeEngine::create_buffer)
       .def("add_pack", &ExchangeEngine::add_pack, py::arg("dom"), py::arg("buf"),
@@ -224,7 +228,7 @@ PYBIND11_MODULE(_C, m) {
              return e.open_remote_buffer(openDev, std::string(handle), bytes);
            })
       .def("finalize", &ExchangeEngine::finalize)
-      .def("launch_translates", &ExchangeEngine::launch_translates)
+      .def("launch_translates", &ExchangeEngine::launch_translates, py::arg("group") = 0)
       .def("launch_packs", &ExchangeEngine::launch_packs, py::arg("group") = 0)
       .def("launch_unpacks", &ExchangeEngine::launch_unpacks, py::arg("group") = 0)
       .def("sync_translates", &ExchangeEngine::sync_translates)

@@ -164,8 +164,8 @@ ExchangeEngine::ExchangeEngine(std::vector<std::shared_ptr<LocalDomain>> domains
 }
 
 ExchangeEngine::~ExchangeEngine() {
-  for (auto &b : translateBatches_) b.destroy();
   for (int g = 0; g < kGroups; ++g) {
+    for (auto &b : translateBatches_[g]) b.destroy();
     for (auto &b : packBatches_[g]) b.destroy();
     for (auto &b : unpackBatches_[g]) b.destroy();
   }
@@ -218,8 +218,8 @@ void ExchangeEngine::enable_peer_all() {
 }
 
 void ExchangeEngine::add_translate(int srcDom, int dstDom, const Vec3 &srcPos, const Vec3 &dstPos,
-                                   const Vec3 &ext) {
-  translateSpecs_.push_back({srcDom, dstDom, false, srcPos, dstPos, ext});
+                                   const Vec3 &ext, int group, const std::vector<int64_t> &qis) {
+  translateSpecs_.push_back({srcDom, dstDom, false, group, qis, srcPos, dstPos, ext});
 }
 
 int64_t ExchangeEngine::create_remote_view(int openDev, const std::vector<std::string> &currHandles,
@@ -262,8 +262,9 @@ int64_t ExchangeEngine::create_remote_view(int openDev, const std::vector<std::s
 }
 
 void ExchangeEngine::add_translate_view(int srcDom, int64_t view, const Vec3 &srcPos,
-                                        const Vec3 &dstPos, const Vec3 &ext) {
-  translateSpecs_.push_back({srcDom, (int)view, true, srcPos, dstPos, ext});
+                                        const Vec3 &dstPos, const Vec3 &ext, int group,
+                                        const std::vector<int64_t> &qis) {
+  translateSpecs_.push_back({srcDom, (int)view, true, group, qis, srcPos, dstPos, ext});
 }
 
 void ExchangeEngine::flip_views() {
@@ -324,8 +325,9 @@ void ExchangeEngine::finalize() {
   build_batches_(translateSpecs_, packSpecs_);
   const char *g = getenv("STENCIL_AMD_GRAPHS");
   const bool graphs = g && g[0] == '1';
-  std::vector<std::vector<CopyBatch> *> sets = {&translateBatches_};
+  std::vector<std::vector<CopyBatch> *> sets;
   for (int gr = 0; gr < kGroups; ++gr) {
+    sets.push_back(&translateBatches_[gr]);
     sets.push_back(&packBatches_[gr]);
     sets.push_back(&unpackBatches_[gr]);
   }
@@ -339,12 +341,14 @@ void ExchangeEngine::finalize() {
 
 void ExchangeEngine::build_batches_(const std::vector<TranslateSpec> &ts,
                                     const std::vector<PackSpec> &ps) {
-  std::map<int, CopyBatch> tb;
-  std::map<std::pair<int, int>, CopyBatch> pb, ub; // (group, gpu)
+  std::map<std::pair<int, int>, CopyBatch> tb, pb, ub; // (group, gpu)
 
   for (const auto &t : ts) {
     LocalDomain &s = *domains_[t.srcDom];
-    for (int64_t qi = 0; qi < s.num_data(); ++qi) {
+    std::vector<int64_t> qis = t.qis;
+    if (qis.empty())
+      for (int64_t qi = 0; qi < s.num_data(); ++qi) qis.push_back(qi);
+    for (int64_t qi : qis) {
       const int64_t es = s.elem_size(qi);
       const Pitched &sp = s.curr(qi); // pitch/ysize only; base via slot
       int64_t dPitch, dPlane;
@@ -377,7 +381,7 @@ void ExchangeEngine::build_batches_(const std::vector<TranslateSpec> &ts,
       j.extXw = (int32_t)(rowBytes / w);
       j.extY = (int32_t)t.ext.y;
       j.nWords = (int64_t)j.extXw * t.ext.y * t.ext.z;
-      auto &batch = tb[s.gpu()];
+      auto &batch = tb[{t.group, s.gpu()}];
       batch.dev = s.gpu();
       batch.jobs.push_back(j);
     }
@@ -420,7 +424,7 @@ void ExchangeEngine::build_batches_(const std::vector<TranslateSpec> &ts,
     batch.jobs.push_back(j);
   }
 
-  for (auto &kv : tb) translateBatches_.push_back(std::move(kv.second));
+  for (auto &kv : tb) translateBatches_[kv.first.first].push_back(std::move(kv.second));
   for (auto &kv : pb) packBatches_[kv.first.first].push_back(std::move(kv.second));
   for (auto &kv : ub) unpackBatches_[kv.first.first].push_back(std::move(kv.second));
 }
@@ -461,9 +465,9 @@ hipStream_t ExchangeEngine::compute_stream(int dom, int which) {
 
 // roctx ranges give rocprof-sys/rocprofv3 timelines the same phase
 // annotations the reference had via NVTX (tracing)
-void ExchangeEngine::launch_translates() {
+void ExchangeEngine::launch_translates(int group) {
   roctxRangePush("stencil::translate");
-  for (auto &b : translateBatches_) b.launch(comm_stream_(b.dev));
+  for (auto &b : translateBatches_[group]) b.launch(comm_stream_(b.dev));
   roctxRangePop();
 }
 void ExchangeEngine::launch_packs(int group) {
@@ -479,9 +483,9 @@ void ExchangeEngine::launch_unpacks(int group) {
 
 void ExchangeEngine::sync_translates() {
   roctxRangePush("stencil::sync_translates");
-  for (auto &b : translateBatches_) {
-    STENCIL_HIP(hipSetDevice(b.dev));
-    STENCIL_HIP(hipStreamSynchronize(comm_stream_(b.dev)));
+  for (auto &kv : commStreams_) {
+    STENCIL_HIP(hipSetDevice(kv.first));
+    STENCIL_HIP(hipStreamSynchronize(kv.second));
   }
   roctxRangePop();
 }

@@ -72,6 +72,7 @@ class DistributedDomain:
         self.methods = Method.DEFAULT
         self.strategy = PlacementStrategy.NodeAware
         self.gpus: Optional[List[int]] = None
+        self.exchange_groups: Optional[List[List[int]]] = None
         self.output_prefix = os.environ.get("STENCIL_OUTPUT_PREFIX", "")
         self.comm = None
         self.placement = None
@@ -105,6 +106,16 @@ class DistributedDomain:
 
     def set_gpus(self, gpus: List[int]):
         self.gpus = list(gpus)
+
+    def set_exchange_groups(self, groups: List[List[int]]):
+        """partition quantities into independently exchangeable groups;
+        exchange(group=i) then moves only groups[i]'s quantities (e.g. the
+        MHD solver exchanges its 8 physics fields and its 2 div fields at
+        different points of a substep). Call before realize()."""
+        seen = [q for g in groups for q in g]
+        if sorted(seen) != sorted(set(seen)):
+            raise ValueError("exchange groups must be disjoint")
+        self.exchange_groups = [sorted(g) for g in groups]
 
     def set_output_prefix(self, p: str):
         self.output_prefix = p
@@ -158,11 +169,13 @@ class DistributedDomain:
         if self.backend_kind == "native":
             from .native_backend import NativeBackend
 
-            self.backend = NativeBackend(specs, self._data, self.radius)
+            self.backend = NativeBackend(specs, self._data, self.radius, self.exchange_groups)
         else:
             from .torch_backend import TorchBackend
 
-            self.backend = TorchBackend(specs, self._data, self.radius, self.torch_device)
+            self.backend = TorchBackend(
+                specs, self._data, self.radius, self.torch_device, self.exchange_groups
+            )
         self.setup_times["realize"] = time.perf_counter() - t0
 
         t0 = time.perf_counter()
@@ -217,24 +230,24 @@ class DistributedDomain:
         return total
 
     # ---- iteration ----
-    def exchange(self):
+    def exchange(self, group: int = 0):
         t0 = time.perf_counter()
-        self.backend.exchange()
+        self.backend.exchange(group)
         self.time_exchange += time.perf_counter() - t0
 
-    def exchange_begin(self):
+    def exchange_begin(self, group: int = 0):
         """asynchronous exchange start (native backend): all device work is
         enqueued; call exchange_end() before reading halos"""
         if hasattr(self.backend, "exchange_begin"):
-            self.backend.exchange_begin()
+            self.backend.exchange_begin(group)
         # torch backend has no async path; everything happens in end()
 
-    def exchange_end(self):
+    def exchange_end(self, group: int = 0):
         t0 = time.perf_counter()
         if hasattr(self.backend, "exchange_end"):
-            self.backend.exchange_end()
+            self.backend.exchange_end(group)
         else:
-            self.backend.exchange()
+            self.backend.exchange(group)
         self.time_exchange += time.perf_counter() - t0
 
     def swap(self):

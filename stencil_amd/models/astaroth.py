@@ -87,6 +87,10 @@ class Astaroth:
         self.size = tuple(size)
         self.dd = DistributedDomain(*size, backend=backend)
         self.dd.set_radius(3)
+        # exchange group 0 = the 8 physics fields (before the div pass),
+        # group 1 = div u / div A (after it) -- each exchange moves only
+        # what changed
+        self.dd.set_exchange_groups([list(range(8)), [8, 9]])
         self.dd.set_placement(placement)
         if gpus is not None:
             self.dd.set_gpus(gpus)
@@ -150,30 +154,31 @@ class Astaroth:
             return _C.Rect3(_C.Vec3(*lo), _C.Vec3(*hi))
 
         if not compute:
-            dd.exchange()
+            dd.exchange(group=0)
+            dd.exchange(group=1)
             dd.backend.sync_compute()
             dd.swap()
             return
         if overlap:
             for li in range(dd.num_local()):
                 _C.mhd_div_pass(eng, li, rect(*self.interiors[li]), self.cf)
-            dd.exchange()  # X1 overlaps div(interior)
+            dd.exchange(group=0)  # field halos; overlaps div(interior)
             for li in range(dd.num_local()):
                 for box in self.exteriors[li]:
                     _C.mhd_div_pass(eng, li, rect(*box), self.cf)
             dd.backend.sync_compute()  # div complete before X2 reads its edges
             for li in range(dd.num_local()):
                 _C.mhd_substep(eng, li, rect(*self.interiors[li]), s, dt, self.cf)
-            dd.exchange()  # X2 overlaps main(interior)
+            dd.exchange(group=1)  # div halos only; overlaps main(interior)
             for li in range(dd.num_local()):
                 for box in self.exteriors[li]:
                     _C.mhd_substep(eng, li, rect(*box), s, dt, self.cf)
         else:
-            dd.exchange()
+            dd.exchange(group=0)
             for li in range(dd.num_local()):
                 _C.mhd_div_pass(eng, li, rect(*dd.local_rect(li)), self.cf)
             dd.backend.sync_compute()
-            dd.exchange()
+            dd.exchange(group=1)
             for li in range(dd.num_local()):
                 _C.mhd_substep(eng, li, rect(*dd.local_rect(li)), s, dt, self.cf)
         dd.backend.sync_compute()

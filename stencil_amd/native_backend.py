@@ -24,10 +24,16 @@ def _rect3(lo, hi) -> "_C.Rect3":
 
 class NativeBackend:
     def __init__(self, domain_specs: List[Tuple[Vec, Vec, int]], data_defs: List[Tuple[int, str]],
-                 radius: "_C.Radius"):
-        """domain_specs: (size, origin, cuda) per local domain"""
+                 radius: "_C.Radius", groups: Optional[List[List[int]]] = None):
+        """domain_specs: (size, origin, cuda) per local domain.
+        groups: quantity-index lists defining independent exchange groups
+        (None = one group with every quantity); exchange(g) then moves
+        only that group's quantities."""
         self.radius = radius
         self.data_defs = list(data_defs)
+        self.groups = groups if groups is not None else [list(range(len(data_defs)))]
+        if len(self.groups) > 4:
+            raise ValueError("at most 4 exchange groups (engine launch-group encoding)")
         self.domains = []
         for size, origin, cuda in domain_specs:
             d = _C.LocalDomain(_vec3(size), _vec3(origin), int(cuda))
@@ -39,15 +45,16 @@ class NativeBackend:
         self.engine = _C.ExchangeEngine(self.domains)
         if len({c for _, _, c in domain_specs}) > 1:
             self.engine.enable_peer_all()
-        self._send_ops: List[Tuple[object, int, int]] = []  # (tensor, peer, tag)
-        self._recv_ops: List[Tuple[object, int, int]] = []
-        self._has_wire = False
+        ng = len(self.groups)
+        self._send_ops = [[] for _ in range(ng)]  # per group: (tensor, peer, tag)
+        self._recv_ops = [[] for _ in range(ng)]
+        self._has_wire = [False] * ng
         self._wire_via_cpu = False
         self._cpu_mirror = {}
         self._ipc_active = False
         self._colo_group = None
         self._ipc_error = None
-        self._colo_parity = 0
+        self._colo_parity = [0] * ng
         self._staging_recv = {}
 
     # ---- plan registration ----
@@ -60,7 +67,7 @@ class NativeBackend:
         self._ipc_active = False
         self._colo_group = None
         self._ipc_error = None
-        self._colo_parity = 0
+        self._colo_parity = [0] * len(self.groups)
         self._staging_recv = {}
         if ctx and ctx.get("ipc", True) and os.environ.get("STENCIL_AMD_IPC", "1") != "0":
             comm = ctx["comm"]
@@ -85,43 +92,50 @@ class NativeBackend:
                     ipc_sends = []
         self._make_colo_groups(ctx)
         elem_sizes = [es for es, _ in self.data_defs]
-        for t in plan.translates:
-            src = self.domains[t.src_local]
-            dst = self.domains[t.dst_local]
-            d = _vec3(t.dir)
-            nd = _vec3(tuple(-c for c in t.dir))
-            src_pos = src.halo_pos(d, False)
-            dst_pos = dst.halo_pos(nd, True)
-            self.engine.add_translate(t.src_local, t.dst_local, src_pos, dst_pos, _vec3(t.ext))
+        for g, qis in enumerate(self.groups):
+            for t in plan.translates:
+                src = self.domains[t.src_local]
+                dst = self.domains[t.dst_local]
+                d = _vec3(t.dir)
+                nd = _vec3(tuple(-c for c in t.dir))
+                src_pos = src.halo_pos(d, False)
+                dst_pos = dst.halo_pos(nd, True)
+                self.engine.add_translate(
+                    t.src_local, t.dst_local, src_pos, dst_pos, _vec3(t.ext), g, sorted(qis)
+                )
 
         import torch
 
-        for item, is_send in [(s, True) for s in plan.sends] + [(r, False) for r in plan.recvs]:
-            total, chunks = wire_layout(item.messages, elem_sizes)
-            buf = self.engine.create_buffer(item.local_id, total)
-            dom = self.domains[item.local_id]
-            for mi, qi, off, nbytes in chunks:
-                m = item.messages[mi]
+        for g, qis in enumerate(self.groups):
+            for item, is_send in [(s, True) for s in plan.sends] + [(r, False) for r in plan.recvs]:
+                total, chunks = wire_layout(item.messages, elem_sizes, qis)
+                buf = self.engine.create_buffer(item.local_id, total)
+                dom = self.domains[item.local_id]
+                for mi, qi, off, nbytes in chunks:
+                    m = item.messages[mi]
+                    if is_send:
+                        pos = dom.halo_pos(_vec3(m.dir), False)
+                        self.engine.add_pack(item.local_id, buf, off, pos, _vec3(m.ext), qi,
+                                             group=3 * g)
+                    else:
+                        nd = _vec3(tuple(-c for c in m.dir))
+                        pos = dom.halo_pos(nd, True)
+                        self.engine.add_unpack(item.local_id, buf, off, pos, _vec3(m.ext), qi,
+                                               group=3 * g)
+                tensor = torch.from_dlpack(self.engine.buffer_dlpack(buf))
+                tag = p2p_tag(item.src_gid, item.dst_gid)
                 if is_send:
-                    pos = dom.halo_pos(_vec3(m.dir), False)
-                    self.engine.add_pack(item.local_id, buf, off, pos, _vec3(m.ext), qi)
+                    self._send_ops[g].append((tensor, item.peer_rank, tag))
                 else:
-                    nd = _vec3(tuple(-c for c in m.dir))
-                    pos = dom.halo_pos(nd, True)
-                    self.engine.add_unpack(item.local_id, buf, off, pos, _vec3(m.ext), qi)
-            tensor = torch.from_dlpack(self.engine.buffer_dlpack(buf))
-            tag = p2p_tag(item.src_gid, item.dst_gid)
-            if is_send:
-                self._send_ops.append((tensor, item.peer_rank, tag))
-            else:
-                self._recv_ops.append((tensor, item.peer_rank, tag))
-        self._has_wire = bool(self._send_ops or self._recv_ops)
-        if self._has_wire:
+                    self._recv_ops[g].append((tensor, item.peer_rank, tag))
+            self._has_wire[g] = bool(self._send_ops[g] or self._recv_ops[g])
+        if any(self._has_wire):
             self._wire_via_cpu = self._detect_cpu_wire()
             if self._wire_via_cpu:
                 self._cpu_mirror = {
                     id(t): torch.empty(t.shape, dtype=torch.uint8, device="cpu")
-                    for t, _, _ in self._send_ops + self._recv_ops
+                    for g in range(len(self.groups))
+                    for t, _, _ in self._send_ops[g] + self._recv_ops[g]
                 }
         self.engine.finalize()
 
@@ -162,31 +176,32 @@ class NativeBackend:
                         "next": [d.ipc_handle(qi, True) for qi in range(nq)],
                     }
                 )
-            # receiver side of the staged path: allocate a double-buffered
-            # staging buffer per incoming (src,dst) pair's thin messages,
-            # register the parity-0/1 unpack jobs, export the IPC handle
+            # receiver side of the staged path: per exchange group, a
+            # double-buffered staging buffer per incoming (src,dst) pair's
+            # thin messages, with parity unpack jobs in groups 3g+1/3g+2
             staging = {}
-            for r in ipc_recvs:
-                thin = [m for m in r.messages if self._is_thin(m)]
-                if not thin:
-                    continue
-                total, chunks = wire_layout(thin, elem_sizes)
-                buf = self.engine.create_buffer(r.local_id, 2 * total)
-                dom = self.domains[r.local_id]
-                for mi, qi, off, nbytes in chunks:
-                    m = thin[mi]
-                    nd = _vec3(tuple(-c for c in m.dir))
-                    pos = dom.halo_pos(nd, True)
-                    for parity in (0, 1):
-                        self.engine.add_unpack(
-                            r.local_id, buf, off + parity * total, pos, _vec3(m.ext), qi,
-                            group=1 + parity,
-                        )
-                staging[(r.src_gid, r.dst_gid)] = (
-                    self.engine.buffer_ipc_handle(buf),
-                    total,
-                )
-                self._staging_recv[(r.src_gid, r.dst_gid)] = (buf, total)
+            for g, qis in enumerate(self.groups):
+                for r in ipc_recvs:
+                    thin = [m for m in r.messages if self._is_thin(m)]
+                    if not thin:
+                        continue
+                    total, chunks = wire_layout(thin, elem_sizes, qis)
+                    buf = self.engine.create_buffer(r.local_id, 2 * total)
+                    dom = self.domains[r.local_id]
+                    for mi, qi, off, nbytes in chunks:
+                        m = thin[mi]
+                        nd = _vec3(tuple(-c for c in m.dir))
+                        pos = dom.halo_pos(nd, True)
+                        for parity in (0, 1):
+                            self.engine.add_unpack(
+                                r.local_id, buf, off + parity * total, pos, _vec3(m.ext), qi,
+                                group=3 * g + 1 + parity,
+                            )
+                    staging[(r.src_gid, r.dst_gid, g)] = (
+                        self.engine.buffer_ipc_handle(buf),
+                        total,
+                    )
+                    self._staging_recv[(r.src_gid, r.dst_gid, g)] = (buf, total)
             export = {"domains": export, "staging": staging}
         except Exception as e:
             export = {"error": str(e)}
@@ -213,12 +228,13 @@ class NativeBackend:
                         src_gpu, info["curr"], info["next"], info["pitch"], info["ysize"],
                         info["es"], info["pad"]
                     )
-                skey = (s.src_gid, s.dst_gid)
-                stg = infos[s.peer_rank]["staging"].get(skey)
-                if stg is not None and skey not in remote_staging:
-                    handle, total = stg
-                    rb = self.engine.open_remote_buffer(src_gpu, handle, 2 * total)
-                    remote_staging[skey] = (rb, total)
+                for g in range(len(self.groups)):
+                    skey = (s.src_gid, s.dst_gid, g)
+                    stg = infos[s.peer_rank]["staging"].get(skey)
+                    if stg is not None and skey not in remote_staging:
+                        handle, total = stg
+                        rb = self.engine.open_remote_buffer(src_gpu, handle, 2 * total)
+                        remote_staging[skey] = (rb, total)
         except Exception as e:
             open_err = str(e)
         # consensus: either EVERY rank uses IPC or none does (a mixed state
@@ -227,32 +243,35 @@ class NativeBackend:
         bad = [v for v in votes if v is not None]
         if bad:
             raise RuntimeError(f"IPC open failed on some rank: {bad[0]}")
-        for s in ipc_sends:
-            dst_idx = placement.dimensionize(s.dst_gid)
-            dst_li = placement.get_subdomain_id(dst_idx)
-            dst_size = placement.subdomain_size(dst_idx)
-            src_gpu = self.domains[s.local_id].gpu()
-            key = (s.peer_rank, dst_li, src_gpu)
-            dom = self.domains[s.local_id]
-            fat = [m for m in s.messages if not self._is_thin(m)]
-            thin = [m for m in s.messages if self._is_thin(m)]
-            for m in fat:
-                nd = _vec3(tuple(-c for c in m.dir))
-                src_pos = dom.halo_pos(_vec3(m.dir), False)
-                dst_pos = _C.halo_pos(nd, _vec3(dst_size), radius, True)
-                self.engine.add_translate_view(s.local_id, views[key], src_pos, dst_pos, _vec3(m.ext))
-            if thin:
-                rb, total = remote_staging[(s.src_gid, s.dst_gid)]
-                _t2, chunks = wire_layout(thin, elem_sizes)
-                assert _t2 == total, "staged wire layout mismatch"
-                for mi, qi, off, nbytes in chunks:
-                    m = thin[mi]
-                    pos = dom.halo_pos(_vec3(m.dir), False)
-                    for parity in (0, 1):
-                        self.engine.add_pack(
-                            s.local_id, rb, off + parity * total, pos, _vec3(m.ext), qi,
-                            group=1 + parity,
-                        )
+        for g, qis in enumerate(self.groups):
+            for s in ipc_sends:
+                dst_idx = placement.dimensionize(s.dst_gid)
+                dst_li = placement.get_subdomain_id(dst_idx)
+                dst_size = placement.subdomain_size(dst_idx)
+                src_gpu = self.domains[s.local_id].gpu()
+                key = (s.peer_rank, dst_li, src_gpu)
+                dom = self.domains[s.local_id]
+                fat = [m for m in s.messages if not self._is_thin(m)]
+                thin = [m for m in s.messages if self._is_thin(m)]
+                for m in fat:
+                    nd = _vec3(tuple(-c for c in m.dir))
+                    src_pos = dom.halo_pos(_vec3(m.dir), False)
+                    dst_pos = _C.halo_pos(nd, _vec3(dst_size), radius, True)
+                    self.engine.add_translate_view(
+                        s.local_id, views[key], src_pos, dst_pos, _vec3(m.ext), g, sorted(qis)
+                    )
+                if thin:
+                    rb, total = remote_staging[(s.src_gid, s.dst_gid, g)]
+                    _t2, chunks = wire_layout(thin, elem_sizes, qis)
+                    assert _t2 == total, "staged wire layout mismatch"
+                    for mi, qi, off, nbytes in chunks:
+                        m = thin[mi]
+                        pos = dom.halo_pos(_vec3(m.dir), False)
+                        for parity in (0, 1):
+                            self.engine.add_pack(
+                                s.local_id, rb, off + parity * total, pos, _vec3(m.ext), qi,
+                                group=3 * g + 1 + parity,
+                            )
 
     def _make_colo_groups(self, ctx):
         """per-node gloo subgroup for the post-translate IPC barrier.
@@ -286,62 +305,66 @@ class NativeBackend:
         return "nccl" not in backend
 
     # ---- per-iteration ----
-    def exchange_begin(self):
-        """stream-ordered first half: launch all local/IPC translates and
-        (if cross-rank) the pack kernels. Returns immediately; GPU work
-        overlaps whatever the app runs on its compute streams."""
-        self.engine.launch_translates()
+    def exchange_begin(self, group: int = 0):
+        """stream-ordered first half for one exchange group: launch all
+        local/IPC translates and (if cross-rank) the pack kernels. Returns
+        immediately; GPU work overlaps the app's compute streams."""
+        g = group
+        self.engine.launch_translates(g)
         if self._ipc_active:
             # staged thin messages: coalesced pack straight into the
             # receiver's staging buffer (parity-selected half)
-            self.engine.launch_packs(1 + self._colo_parity)
-        if self._has_wire:
-            self.engine.launch_packs()
+            self.engine.launch_packs(3 * g + 1 + self._colo_parity[g])
+        if self._has_wire[g]:
+            self.engine.launch_packs(3 * g)
 
-    def exchange_end(self):
-        """second half: move wire buffers, unpack, and block until every
-        halo is in place."""
-        if self._has_wire:
+    def exchange_end(self, group: int = 0):
+        """second half: move wire buffers, unpack, and block until the
+        group's halos are in place."""
+        g = group
+        if self._has_wire[g]:
             import torch.distributed as dist
 
             self.engine.sync_packs()
             if self._wire_via_cpu:
                 ops = []
-                for t, peer, tag in self._send_ops:
+                for t, peer, tag in self._send_ops[g]:
                     m = self._cpu_mirror[id(t)]
                     m.copy_(t)
                     ops.append(dist.P2POp(dist.isend, m, peer, tag=tag))
-                for t, peer, tag in self._recv_ops:
+                for t, peer, tag in self._recv_ops[g]:
                     ops.append(dist.P2POp(dist.irecv, self._cpu_mirror[id(t)], peer, tag=tag))
                 for w in dist.batch_isend_irecv(ops):
                     w.wait()
-                for t, _, _ in self._recv_ops:
+                for t, _, _ in self._recv_ops[g]:
                     t.copy_(self._cpu_mirror[id(t)])
             else:
-                ops = [dist.P2POp(dist.isend, t, peer, tag=tag) for t, peer, tag in self._send_ops]
+                ops = [
+                    dist.P2POp(dist.isend, t, peer, tag=tag) for t, peer, tag in self._send_ops[g]
+                ]
                 ops += [
-                    dist.P2POp(dist.irecv, t, peer, tag=tag) for t, peer, tag in self._recv_ops
+                    dist.P2POp(dist.irecv, t, peer, tag=tag) for t, peer, tag in self._recv_ops[g]
                 ]
                 for w in dist.batch_isend_irecv(ops):
                     w.wait()
-            self.engine.launch_unpacks()
+            self.engine.launch_unpacks(3 * g)
         self.engine.sync_all()
         if self._ipc_active:
             # all colocated ranks' direct writes and staged packs are
             # complete after the barrier (each rank synced its own streams
             # above); then unpack this exchange's staging parity locally.
-            # The barrier also keeps senders at most one exchange ahead,
-            # which is what makes the two staging parities sufficient.
+            # The barrier also keeps senders at most one exchange (of this
+            # group) ahead, which makes two staging parities sufficient.
             import torch.distributed as dist
 
             dist.barrier(group=self._colo_group)
-            self.engine.launch_unpacks(1 + self._colo_parity)
+            self.engine.launch_unpacks(3 * g + 1 + self._colo_parity[g])
             self.engine.sync_packs()
-            self._colo_parity ^= 1
+            self._colo_parity[g] ^= 1
 
-    def exchange(self):
-        self.exchange_begin()
-        self.exchange_end()
+    def exchange(self, group: int = 0):
+        self.exchange_begin(group)
+        self.exchange_end(group)
 
     def swap(self):
         for d in self.domains:

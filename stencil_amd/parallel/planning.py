@@ -140,17 +140,21 @@ def plan_exchange(placement, radius: "_C.Radius", rank: int) -> ExchangePlan:
     return plan
 
 
-def wire_layout(messages: List[Message], elem_sizes: List[int]) -> Tuple[int, List[Tuple[int, int, int, int]]]:
+def wire_layout(messages: List[Message], elem_sizes: List[int], qis=None
+                ) -> Tuple[int, List[Tuple[int, int, int, int]]]:
     """byte layout of one packed buffer: per message (sorted by direction),
-    per quantity, a 16 B-aligned chunk. Returns (total_bytes, chunks) with
-    chunks = [(msg_index, qi, offset, nbytes)]. Both ranks compute this
+    per quantity of the exchange group (sorted indices; None = all), a
+    16 B-aligned chunk. Returns (total_bytes, chunks) with chunks =
+    [(msg_index, qi, offset, nbytes)]. Both ranks compute this
     identically, so it is the wire format."""
+    if qis is None:
+        qis = range(len(elem_sizes))
     chunks = []
     off = 0
     for mi, m in enumerate(messages):
-        for qi, es in enumerate(elem_sizes):
+        for qi in sorted(qis):
             off = (off + 15) // 16 * 16
-            nbytes = es * m.volume()
+            nbytes = elem_sizes[qi] * m.volume()
             chunks.append((mi, qi, off, nbytes))
             off += nbytes
     return (off + 15) // 16 * 16, chunks

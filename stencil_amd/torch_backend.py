@@ -75,9 +75,10 @@ class _TorchDomain:
 
 
 class TorchBackend:
-    def __init__(self, domain_specs, data_defs, radius, device: str = "cpu"):
+    def __init__(self, domain_specs, data_defs, radius, device: str = "cpu", groups=None):
         self.radius = radius
         self.data_defs = list(data_defs)
+        self.groups = groups if groups is not None else [list(range(len(data_defs)))]
         if device.startswith("cuda") and len({c for _, _, c in domain_specs}) > 1:
             # one tensor device per local domain
             self.domains = [
@@ -87,68 +88,71 @@ class TorchBackend:
             self.domains = [
                 _TorchDomain(s, o, device, data_defs, radius) for s, o, c in domain_specs
             ]
-        self._translates = []
-        self._send_bufs = []  # (buffer, peer, tag, [(chunk, dom, pos, ext, qi)])
-        self._recv_bufs = []
+        ng = len(self.groups)
+        self._translates = [[] for _ in range(ng)]
+        self._send_bufs = [[] for _ in range(ng)]  # per group
+        self._recv_bufs = [[] for _ in range(ng)]
 
     def register_plan(self, plan: ExchangePlan, ctx=None):
         elem_sizes = [es for es, _ in self.data_defs]
-        for t in plan.translates:
-            src = self.domains[t.src_local]
-            dst = self.domains[t.dst_local]
-            nd = tuple(-c for c in t.dir)
-            self._translates.append(
-                (t.src_local, t.dst_local, src.halo_pos(t.dir, False), dst.halo_pos(nd, True), t.ext)
-            )
-        for item, is_send in [(s, True) for s in plan.sends] + [(r, False) for r in plan.recvs]:
-            total, chunks = wire_layout(item.messages, elem_sizes)
-            dom = self.domains[item.local_id]
-            buf = torch.zeros(total, dtype=torch.uint8, device=dom.device)
-            entries = []
-            for mi, qi, off, nbytes in chunks:
-                m = item.messages[mi]
-                nd = tuple(-c for c in m.dir)
-                pos = dom.halo_pos(m.dir, False) if is_send else dom.halo_pos(nd, True)
-                entries.append((off, nbytes, item.local_id, pos, m.ext, qi))
-            rec = (buf, item.peer_rank, p2p_tag(item.src_gid, item.dst_gid), entries)
-            (self._send_bufs if is_send else self._recv_bufs).append(rec)
+        for g, qis in enumerate(self.groups):
+            for t in plan.translates:
+                src = self.domains[t.src_local]
+                dst = self.domains[t.dst_local]
+                nd = tuple(-c for c in t.dir)
+                self._translates[g].append(
+                    (t.src_local, t.dst_local, src.halo_pos(t.dir, False),
+                     dst.halo_pos(nd, True), t.ext, sorted(qis))
+                )
+            for item, is_send in [(s, True) for s in plan.sends] + [(r, False) for r in plan.recvs]:
+                total, chunks = wire_layout(item.messages, elem_sizes, qis)
+                dom = self.domains[item.local_id]
+                buf = torch.zeros(total, dtype=torch.uint8, device=dom.device)
+                entries = []
+                for mi, qi, off, nbytes in chunks:
+                    m = item.messages[mi]
+                    nd = tuple(-c for c in m.dir)
+                    pos = dom.halo_pos(m.dir, False) if is_send else dom.halo_pos(nd, True)
+                    entries.append((off, nbytes, item.local_id, pos, m.ext, qi))
+                rec = (buf, item.peer_rank, p2p_tag(item.src_gid, item.dst_gid), entries)
+                (self._send_bufs[g] if is_send else self._recv_bufs[g]).append(rec)
 
-    def _pack(self):
-        for buf, _peer, _tag, entries in self._send_bufs:
+    def _pack(self, g):
+        for buf, _peer, _tag, entries in self._send_bufs[g]:
             for off, nbytes, li, pos, ext, qi in entries:
                 reg = self.domains[li].region(qi, pos, ext).contiguous()
                 buf[off : off + nbytes] = reg.view(-1).view(torch.uint8)
 
-    def _unpack(self):
-        for buf, _peer, _tag, entries in self._recv_bufs:
+    def _unpack(self, g):
+        for buf, _peer, _tag, entries in self._recv_bufs[g]:
             for off, nbytes, li, pos, ext, qi in entries:
                 dom = self.domains[li]
                 dtype = dom.curr[qi].dtype
                 reg = buf[off : off + nbytes].view(dtype).reshape(ext[2], ext[1], ext[0])
                 dom.region(qi, pos, ext).copy_(reg)
 
-    def exchange(self):
-        for sl, dl, spos, dpos, ext in self._translates:
+    def exchange(self, group: int = 0):
+        g = group
+        for sl, dl, spos, dpos, ext, qis in self._translates[g]:
             src = self.domains[sl]
             dst = self.domains[dl]
-            dst.region(0, dpos, ext)  # noop guard
-            for qi in range(len(self.data_defs)):
+            for qi in qis:
                 dst.region(qi, dpos, ext).copy_(src.region(qi, spos, ext))
-        if self._send_bufs or self._recv_bufs:
+        if self._send_bufs[g] or self._recv_bufs[g]:
             import torch.distributed as dist
 
-            self._pack()
+            self._pack(g)
             ops = [
                 dist.P2POp(dist.isend, buf, peer, tag=tag)
-                for buf, peer, tag, _ in self._send_bufs
+                for buf, peer, tag, _ in self._send_bufs[g]
             ]
             ops += [
                 dist.P2POp(dist.irecv, buf, peer, tag=tag)
-                for buf, peer, tag, _ in self._recv_bufs
+                for buf, peer, tag, _ in self._recv_bufs[g]
             ]
             for w in dist.batch_isend_irecv(ops):
                 w.wait()
-            self._unpack()
+            self._unpack(g)
 
     def swap(self):
         for d in self.domains:

@@ -200,3 +200,27 @@ def test_halo_multiplier_with_radius2():
         )
     for a, b in zip(outs[0], outs[1]):
         np.testing.assert_array_equal(a, b)
+
+
+def test_exchange_groups_selective():
+    """exchange(group) must move only that group's quantities"""
+    import stencil_amd as sa
+
+    dd = make_dd((10, 8, 8), 1, 2)
+    dd.set_exchange_groups([[0], [1]])
+    ha = dd.add_data(np.float32, "a")
+    hb = dd.add_data(np.float32, "b")
+    dd.realize()
+    fill_interiors(dd, ha, scale=1.0)
+    fill_interiors(dd, hb, scale=2.0)
+    dd.exchange(group=0)
+    check_full_regions(dd, ha, scale=1.0)  # a's halos are fresh
+    # b's halos are still zero (only the interior was written)
+    from util import full_region_of
+
+    lo, hi = dd.local_rect(0)
+    flo, fhi = full_region_of(dd, 0)
+    got = dd.read_global(0, flo, fhi, hb)
+    assert got[0, 0, 0] == 0.0  # a corner halo cell untouched
+    dd.exchange(group=1)
+    check_full_regions(dd, hb, scale=2.0)
