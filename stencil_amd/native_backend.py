@@ -157,7 +157,6 @@ class NativeBackend:
         comm, placement = ctx["comm"], ctx["placement"]
         radius = self.radius
         elem_sizes = [es for es, _ in self.data_defs]
-        self._colo_parity = 0
         # phase 1: export handles. A failing rank must still reach the
         # allgather (its peers would otherwise hang), so errors are
         # exported as data and re-raised on EVERY rank afterwards.
