@@ -55,3 +55,37 @@ def test_fuzz_ripple(seed):
     dd.exchange()
     for h, dtype, scale in handles:
         check_valid_regions(dd, h, scale)
+
+
+@pytest.mark.parametrize("seed", range(6))
+def test_fuzz_exchange_groups(seed):
+    """random quantity partitions into 2 exchange groups; exchanging both
+    groups must fill every fillable halo of every quantity"""
+    rng = random.Random(1000 + seed)
+    size = tuple(rng.randint(8, 20) for _ in range(3))
+    n_dom = rng.choice([1, 2, 3])
+    radius = random_radius(rng)
+    max_r = max(
+        radius.dir(x, y, z) for x in (-1, 0, 1) for y in (-1, 0, 1) for z in (-1, 0, 1)
+    )
+    size = tuple(max(s, max_r * n_dom * 2 + n_dom) for s in size)
+    nq = rng.randint(2, 4)
+    qs = list(range(nq))
+    rng.shuffle(qs)
+    cut = rng.randint(1, nq - 1)
+    groups = [sorted(qs[:cut]), sorted(qs[cut:])]
+
+    dd = sa.DistributedDomain(*size, backend="torch")
+    dd.set_radius(radius)
+    dd.set_gpus([0] * n_dom)
+    dd.set_exchange_groups(groups)
+    handles = [(dd.add_data(np.float32, f"q{qi}"), 1.0 + qi) for qi in range(nq)]
+    dd.realize()
+    for h, scale in handles:
+        for li in range(dd.num_local()):
+            lo, hi = dd.local_rect(li)
+            dd.write_global(li, lo, ripple_block(lo, hi, dd.size, scale), h)
+    dd.exchange(group=0)
+    dd.exchange(group=1)
+    for h, scale in handles:
+        check_valid_regions(dd, h, scale)
