@@ -34,7 +34,7 @@ def main():
     import torch.distributed as dist
 
     local_rank = int(os.environ.get("LOCAL_RANK", rank))
-    torch.cuda.set_device(local_rank)
+    torch.cuda.set_device(local_rank % max(1, torch.cuda.device_count()))
     dist.init_process_group(backend="cpu:gloo,cuda:nccl")
 
     # plan only (no allocation of domains)

@@ -40,7 +40,9 @@ def main():
         import torch
         import torch.distributed as dist
 
-        torch.cuda.set_device(int(os.environ.get("LOCAL_RANK", rank)))
+        torch.cuda.set_device(
+            int(os.environ.get("LOCAL_RANK", rank)) % max(1, torch.cuda.device_count())
+        )
         dist.init_process_group(backend="cpu:gloo,cuda:nccl")
         n, gpus = world, None
     else:
