@@ -224,3 +224,33 @@ def test_exchange_groups_selective():
     assert got[0, 0, 0] == 0.0  # a corner halo cell untouched
     dd.exchange(group=1)
     check_full_regions(dd, hb, scale=2.0)
+
+
+def test_plan_files_and_timers(tmp_path):
+    """Aux-subsystem parity (reference SETUP_STATS/EXCHANGE_STATS +
+    --plan-file machinery, src/stencil.cu realize): plan/matrix files are
+    written under the output prefix and the always-on timers accumulate."""
+    dd = make_dd((12, 10, 8), 1, 2)
+    dd.set_output_prefix(str(tmp_path / "run_"))
+    h = dd.add_data(np.float32, "q")
+    dd.realize()
+    plans = list(tmp_path.glob("run_plan_*.txt"))
+    assert len(plans) == 1, plans
+    txt = plans[0].read_text()
+    assert "q" in txt or "quantit" in txt or len(txt) > 0
+    assert "direct_kernel" in txt and "bytes=" in txt
+    # rank x rank matrix: single process -> 1x1 (all traffic is direct)
+    mats = list(tmp_path.glob("run_mat*.txt"))
+    assert len(mats) == 1
+    mat = np.atleast_2d(np.loadtxt(str(mats[0])))
+    assert mat.shape == (1, 1)
+
+    for k in ("topo", "placement", "realize", "plan", "create"):
+        assert k in dd.setup_times and dd.setup_times[k] >= 0.0
+
+    fill_interiors(dd, h)
+    assert dd.time_exchange == 0.0
+    dd.exchange()
+    dd.swap()
+    assert dd.time_exchange > 0.0
+    assert dd.time_swap > 0.0
