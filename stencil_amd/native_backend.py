@@ -53,6 +53,7 @@ class NativeBackend:
         self._cpu_mirror = {}
         self._ipc_active = False
         self._colo_group = None
+        self._colo_nccl_group = None
         self._ipc_error = None
         self._colo_parity = [0] * ng
         self._staging_recv = {}
@@ -66,6 +67,7 @@ class NativeBackend:
 
         self._ipc_active = False
         self._colo_group = None
+        self._colo_nccl_group = None
         self._ipc_error = None
         self._colo_parity = [0] * len(self.groups)
         self._staging_recv = {}
@@ -273,10 +275,25 @@ class NativeBackend:
                             )
 
     def _make_colo_groups(self, ctx):
-        """per-node gloo subgroup for the post-translate IPC barrier.
-        new_group is collective: every rank creates every node's group."""
+        """per-node subgroups for the post-translate IPC barrier.
+        new_group is collective: every rank creates every node's group.
+
+        Two barrier flavors: a gloo host barrier (always created; the
+        fallback and the only option when ranks share a device — RCCL
+        refuses communicators with two ranks on one GPU), and a device
+        barrier (1-element RCCL all_reduce, waited via event so the
+        engine's in-flight compute streams are NOT synced) used when every
+        colocated rank drives a distinct GPU. A host gloo barrier costs
+        O(100us..ms) per exchange at world 8, which is material against a
+        ~1.3 ms jacobi step; the RCCL barrier tracks stream completion at
+        collective cost (~20-30us). STENCIL_AMD_BARRIER=gloo forces the
+        host barrier; the decision is allgathered so it is identical on
+        every rank (a mixed choice would deadlock)."""
         if not self._ipc_active or ctx is None:
             return
+        import os
+
+        import torch
         import torch.distributed as dist
 
         comm = ctx["comm"]
@@ -284,10 +301,44 @@ class NativeBackend:
         by_node = {}
         for r, node in enumerate(nodes):
             by_node.setdefault(node, []).append(r)
+        my_dev = self.domains[0].gpu() if self.domains else -1
+        want_dev = (
+            os.environ.get("STENCIL_AMD_BARRIER", "auto") != "gloo"
+            and torch.cuda.is_available()
+            and "nccl" in dist.get_backend_config()
+        )
+        infos = comm.allgather_object((nodes[comm.rank], my_dev, bool(want_dev)))
+        use_dev = all(i[2] for i in infos)
+        if use_dev:
+            for node, ranks in by_node.items():
+                devs = [infos[r][1] for r in ranks]
+                if len(set(devs)) != len(devs):
+                    use_dev = False  # shared device somewhere -> gloo everywhere
         for node in sorted(by_node):
             g = dist.new_group(ranks=by_node[node], backend="gloo")
             if comm.rank in by_node[node]:
                 self._colo_group = g
+        self._colo_nccl_group = None
+        if use_dev:
+            for node in sorted(by_node):
+                g = dist.new_group(ranks=by_node[node], backend="nccl")
+                if comm.rank in by_node[node]:
+                    self._colo_nccl_group = g
+            self._barrier_t = torch.ones(1, device="cuda")
+            self._barrier_ev = torch.cuda.Event()
+
+    def _colo_barrier(self):
+        """block the host until every colocated rank has passed its
+        translate sync (their xGMI writes into our buffers are complete)."""
+        import torch.distributed as dist
+
+        if self._colo_nccl_group is not None:
+            w = dist.all_reduce(self._barrier_t, group=self._colo_nccl_group, async_op=True)
+            w.wait()  # torch's current (default) stream waits on the collective
+            self._barrier_ev.record()  # default stream is otherwise empty:
+            self._barrier_ev.synchronize()  # host waits for the collective only
+        else:
+            dist.barrier(group=self._colo_group)
 
     @staticmethod
     def _detect_cpu_wire() -> bool:
@@ -354,9 +405,7 @@ class NativeBackend:
             # above); then unpack this exchange's staging parity locally.
             # The barrier also keeps senders at most one exchange (of this
             # group) ahead, which makes two staging parities sufficient.
-            import torch.distributed as dist
-
-            dist.barrier(group=self._colo_group)
+            self._colo_barrier()
             self.engine.launch_unpacks(3 * g + 1 + self._colo_parity[g])
             self.engine.sync_packs()
             self._colo_parity[g] ^= 1
