@@ -43,6 +43,9 @@ public:
   void realize();
   // swap curr/next and refresh the device pointer tables in place
   void swap();
+  // swap only the HOST-side mirrors: used by the whole-step hipGraph
+  // path, whose in-graph swap_tables kernel flips the device tables
+  void swap_host_only() { std::swap(curr_, next_); }
 
   //// geometry (all positions in "allocation coordinates": element offsets
   //// from the first allocated element, which sits at global coordinate

@@ -60,6 +60,8 @@ struct CopyBatch {
   void finalize_upload();
   void capture_graph(); // capture the batch launch into a hipGraph
   void launch(hipStream_t stream);
+  // plain kernel launch, never via graphExec (safe inside stream capture)
+  void launch_plain(hipStream_t stream);
   void destroy();
 };
 
@@ -117,6 +119,10 @@ public:
 
   //// per-exchange execution (stream-ordered; host-sync via the sync_* calls)
   void launch_translates(int group = 0);
+  // launch a group's translate batches onto a caller-owned stream with
+  // plain kernel launches (used by whole-step hipGraph capture, where a
+  // nested hipGraphLaunch would be illegal)
+  void launch_translates_plain_on(uintptr_t stream, int group = 0);
   void launch_packs(int group = 0);
   void launch_unpacks(int group = 0);
   void sync_translates();

@@ -16,6 +16,16 @@ void jacobi_step(ExchangeEngine &eng, int dom, int64_t qi, const Rect3 &region,
 void fill_f32(ExchangeEngine &eng, int dom, int64_t qi, const Rect3 &region, float value,
               bool nextBuf);
 
+// Whole-step hipGraph for the single-process single-domain jacobi path:
+// [translate copy_batch -> full-region jacobi -> device-side table swap]
+// captured once per buffer parity, replayed per step (~5 us host cost vs
+// ~0.25 ms of per-step orchestration). launch() enqueues nSteps replays
+// (alternating parities, host mirrors flipped); sync() blocks on them.
+int64_t jacobi_graph_create(ExchangeEngine &eng, int dom, int64_t qi, const Rect3 &region,
+                            const Rect3 &computeRegion);
+void jacobi_graph_launch(int64_t handle, int64_t nSteps);
+void jacobi_graph_sync(int64_t handle);
+
 // physical coefficients of the MHD solver (see csrc/src/mhd.hip)
 struct MhdCoeffs {
   double dsx = 1.0, dsy = 1.0, dsz = 1.0;

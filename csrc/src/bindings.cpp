@@ -262,6 +262,10 @@ PYBIND11_MODULE(_C, m) {
   m.def("jacobi_step", &jacobi_step, py::arg("eng"), py::arg("dom"), py::arg("qi"),
         py::arg("region"), py::arg("compute_region"), py::arg("stream_id") = 0);
   m.def("fill_f32", &fill_f32);
+  m.def("jacobi_graph_create", &jacobi_graph_create, py::arg("eng"), py::arg("dom"),
+        py::arg("qi"), py::arg("region"), py::arg("compute_region"));
+  m.def("jacobi_graph_launch", &jacobi_graph_launch, py::arg("handle"), py::arg("n_steps") = 1);
+  m.def("jacobi_graph_sync", &jacobi_graph_sync);
 
   py::class_<MhdCoeffs>(m, "MhdCoeffs")
       .def(py::init<>())

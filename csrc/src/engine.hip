@@ -148,6 +148,13 @@ void CopyBatch::launch(hipStream_t stream) {
   STENCIL_HIP(hipGetLastError());
 }
 
+void CopyBatch::launch_plain(hipStream_t stream) {
+  if (jobs.empty()) return;
+  hipLaunchKernelGGL(copy_batch_kernel, dim3((uint32_t)nBlocks), dim3(kBlock), 0, stream, dJobs,
+                     dPrefix, (int)jobs.size());
+  STENCIL_HIP(hipGetLastError());
+}
+
 void CopyBatch::destroy() {
   if (graphExec) (void)hipGraphExecDestroy(graphExec);
   graphExec = nullptr;
@@ -469,6 +476,9 @@ void ExchangeEngine::launch_translates(int group) {
   roctxRangePush("stencil::translate");
   for (auto &b : translateBatches_[group]) b.launch(comm_stream_(b.dev));
   roctxRangePop();
+}
+void ExchangeEngine::launch_translates_plain_on(uintptr_t stream, int group) {
+  for (auto &b : translateBatches_[group]) b.launch_plain((hipStream_t)stream);
 }
 void ExchangeEngine::launch_packs(int group) {
   roctxRangePush("stencil::pack");

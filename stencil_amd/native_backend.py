@@ -439,5 +439,16 @@ class NativeBackend:
         _C.jacobi_step(self.engine, li, qi, _rect3(region_lo, region_hi), _rect3(c_lo, c_hi),
                        stream_id)
 
+    def jacobi_graph_create(self, li: int, qi: int, region_lo: Vec, region_hi: Vec,
+                            c_lo: Vec, c_hi: Vec) -> int:
+        """whole-step replay graph (single-process single-domain path);
+        see csrc/src/jacobi.hip jacobi_graph_create"""
+        return _C.jacobi_graph_create(self.engine, li, qi, _rect3(region_lo, region_hi),
+                                      _rect3(c_lo, c_hi))
+
+    def jacobi_graph_step(self, handle: int, n: int = 1):
+        _C.jacobi_graph_launch(handle, n)
+        _C.jacobi_graph_sync(handle)
+
     def sync_compute(self):
         self.engine.sync_compute()
