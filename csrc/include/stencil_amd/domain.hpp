@@ -46,6 +46,10 @@ public:
   // swap only the HOST-side mirrors: used by the whole-step hipGraph
   // path, whose in-graph swap_tables kernel flips the device tables
   void swap_host_only() { std::swap(curr_, next_); }
+  // enqueue a tiny kernel that flips the CONTENTS of the device pointer
+  // tables (curr <-> next) on `stream` -- the graph-node counterpart of
+  // swap(): slot-indirect copy jobs stay valid across replays
+  void enqueue_table_swap(hipStream_t stream);
 
   //// geometry (all positions in "allocation coordinates": element offsets
   //// from the first allocated element, which sits at global coordinate

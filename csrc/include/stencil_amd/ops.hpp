@@ -43,6 +43,15 @@ struct MhdCoeffs {
 // so grad(div .) reduces to first derivatives of the exchanged div fields.
 void mhd_div_pass(ExchangeEngine &eng, int dom, const Rect3 &region, const MhdCoeffs &cf,
                   int streamId = 0);
+// Whole-substep hipGraphs for the world=1 single-domain astaroth shape:
+// per RK3 substep and buffer parity, [X1 translates -> div -> X2
+// translates -> scalar || momentum -> table swap] captured once;
+// mhd_graph_iter replays 3*nIters substep graphs (host mirrors flipped).
+int64_t mhd_graph_create(ExchangeEngine &eng, int dom, const Rect3 &region, double dt,
+                         const MhdCoeffs &cf);
+void mhd_graph_iter(int64_t handle, int64_t nIters = 1);
+void mhd_graph_sync(int64_t handle);
+
 void mhd_substep(ExchangeEngine &eng, int dom, const Rect3 &region, int step, double dt,
                  const MhdCoeffs &cf, int streamId = 0);
 

@@ -266,6 +266,10 @@ PYBIND11_MODULE(_C, m) {
         py::arg("qi"), py::arg("region"), py::arg("compute_region"));
   m.def("jacobi_graph_launch", &jacobi_graph_launch, py::arg("handle"), py::arg("n_steps") = 1);
   m.def("jacobi_graph_sync", &jacobi_graph_sync);
+  m.def("mhd_graph_create", &mhd_graph_create, py::arg("eng"), py::arg("dom"), py::arg("region"),
+        py::arg("dt"), py::arg("cf"));
+  m.def("mhd_graph_iter", &mhd_graph_iter, py::arg("handle"), py::arg("n_iters") = 1);
+  m.def("mhd_graph_sync", &mhd_graph_sync);
 
   py::class_<MhdCoeffs>(m, "MhdCoeffs")
       .def(py::init<>())

@@ -72,6 +72,23 @@ void LocalDomain::swap() {
   swapUpload_();
 }
 
+namespace {
+__global__ void swap_tables_kernel(char **a, char **b, int n) {
+  const int i = threadIdx.x;
+  if (i < n) {
+    char *t = a[i];
+    a[i] = b[i];
+    b[i] = t;
+  }
+}
+} // namespace
+
+void LocalDomain::enqueue_table_swap(hipStream_t stream) {
+  hipLaunchKernelGGL(swap_tables_kernel, dim3(1), dim3(256), 0, stream, devCurrRaw_, devNextRaw_,
+                     (int)num_data());
+  STENCIL_HIP(hipGetLastError());
+}
+
 void LocalDomain::swapUpload_() {
   STENCIL_HIP(hipSetDevice(dev_));
   const int64_t n = num_data();

@@ -301,20 +301,10 @@ void jacobi_step(ExchangeEngine &eng, int dom, int64_t qi, const Rect3 &region,
 
 namespace {
 
-// flips the contents of the fixed device pointer tables (curr <-> next)
-// as a graph node, so a captured step graph stays valid forever: the
-// translate jobs read through these slots and the jacobi kernargs are
-// baked per parity (one graph per parity avoids the measured ~18%
-// slot-indirection cost on the jacobi kernel itself).
-__global__ void swap_tables_kernel(char **a, char **b, int n) {
-  const int i = threadIdx.x;
-  if (i < n) {
-    char *t = a[i];
-    a[i] = b[i];
-    b[i] = t;
-  }
-}
-
+// One graph per buffer parity: the jacobi kernargs are baked per parity
+// (slot-indirecting the jacobi kernel itself was measured ~18% slower),
+// while the translate jobs stay slot-indirect and an in-graph
+// LocalDomain::enqueue_table_swap node flips the device tables.
 struct StepGraph {
   hipStream_t stream = nullptr;
   hipGraphExec_t exec[2] = {nullptr, nullptr};
@@ -344,8 +334,7 @@ int64_t jacobi_graph_create(ExchangeEngine &eng, int dom, int64_t qi, const Rect
     STENCIL_HIP(hipStreamBeginCapture(sg->stream, hipStreamCaptureModeThreadLocal));
     eng.launch_translates_plain_on((uintptr_t)sg->stream, 0);
     launch_jacobi_on(d, qi, region, computeRegion, sg->stream);
-    hipLaunchKernelGGL(swap_tables_kernel, dim3(1), dim3(256), 0, sg->stream,
-                       d.dev_curr_slots(), d.dev_next_slots(), (int)d.num_data());
+    d.enqueue_table_swap(sg->stream);
     hipGraph_t g = nullptr;
     STENCIL_HIP(hipStreamEndCapture(sg->stream, &g));
     STENCIL_HIP(hipGraphInstantiate(&sg->exec[par], g, nullptr, nullptr, 0));

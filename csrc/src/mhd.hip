@@ -375,24 +375,27 @@ static dim3 mhd_block() {
   return dim3((uint32_t)bx, (uint32_t)by, (uint32_t)bz);
 }
 
-void mhd_div_pass(ExchangeEngine &eng, int dom, const Rect3 &region, const MhdCoeffs &cf,
-                  int streamId) {
-  LocalDomain &d = eng.domain(dom);
+namespace {
+
+dim3 mhd_grid(const Vec3 &ext, const dim3 &block) {
+  return dim3((uint32_t)((ext.x + block.x - 1) / block.x),
+              (uint32_t)((ext.y + block.y - 1) / block.y),
+              (uint32_t)((ext.z + block.z - 1) / block.z));
+}
+
+void mhd_div_launch_on(LocalDomain &d, const Rect3 &region, const MhdCoeffs &cf,
+                       hipStream_t stream) {
   const Vec3 ext = region.extent();
   if (ext.flatten() <= 0) return;
   MhdParams p{};
   mhd_fill_params(d, region, cf, p);
-  STENCIL_HIP(hipSetDevice(d.gpu()));
   dim3 block = mhd_block();
-  dim3 grid((uint32_t)((ext.x + block.x - 1) / block.x), (uint32_t)((ext.y + block.y - 1) / block.y),
-            (uint32_t)((ext.z + block.z - 1) / block.z));
-  hipLaunchKernelGGL(mhd_div_kernel, grid, block, 0, eng.compute_stream(dom, streamId), p);
+  hipLaunchKernelGGL(mhd_div_kernel, mhd_grid(ext, block), block, 0, stream, p);
   STENCIL_HIP(hipGetLastError());
 }
 
-void mhd_substep(ExchangeEngine &eng, int dom, const Rect3 &region, int step, double dt,
-                 const MhdCoeffs &cf, int streamId) {
-  LocalDomain &d = eng.domain(dom);
+void mhd_substep_launch_on(LocalDomain &d, const Rect3 &region, int step, double dt,
+                           const MhdCoeffs &cf, hipStream_t sScalar, hipStream_t sMomentum) {
   const Vec3 ext = region.extent();
   if (ext.flatten() <= 0) return;
   // Williamson (1980) coefficients
@@ -403,18 +406,105 @@ void mhd_substep(ExchangeEngine &eng, int dom, const Rect3 &region, int step, do
   p.dt = dt;
   p.alpha_over_beta_prev = (step == 0) ? 0.0 : ALPHA[step] / BETA[step - 1];
   p.beta = BETA[step];
-  STENCIL_HIP(hipSetDevice(d.gpu()));
   dim3 block = mhd_block();
-  dim3 grid((uint32_t)((ext.x + block.x - 1) / block.x), (uint32_t)((ext.y + block.y - 1) / block.y),
-            (uint32_t)((ext.z + block.z - 1) / block.z));
   // scalar (writes lnrho/ss/aa) and momentum (writes uu) touch disjoint
   // outputs and only read shared inputs: run them CONCURRENTLY on the two
-  // compute streams (the caller's sync_compute joins both)
-  hipLaunchKernelGGL(mhd_scalar_kernel, grid, block, 0, eng.compute_stream(dom, streamId), p);
+  // streams (the caller joins them)
+  hipLaunchKernelGGL(mhd_scalar_kernel, mhd_grid(ext, block), block, 0, sScalar, p);
   STENCIL_HIP(hipGetLastError());
-  hipLaunchKernelGGL(mhd_momentum_kernel, grid, block, 0, eng.compute_stream(dom, 1 - streamId),
-                     p);
+  hipLaunchKernelGGL(mhd_momentum_kernel, mhd_grid(ext, block), block, 0, sMomentum, p);
   STENCIL_HIP(hipGetLastError());
+}
+
+} // namespace
+
+void mhd_div_pass(ExchangeEngine &eng, int dom, const Rect3 &region, const MhdCoeffs &cf,
+                  int streamId) {
+  LocalDomain &d = eng.domain(dom);
+  STENCIL_HIP(hipSetDevice(d.gpu()));
+  mhd_div_launch_on(d, region, cf, eng.compute_stream(dom, streamId));
+}
+
+void mhd_substep(ExchangeEngine &eng, int dom, const Rect3 &region, int step, double dt,
+                 const MhdCoeffs &cf, int streamId) {
+  LocalDomain &d = eng.domain(dom);
+  STENCIL_HIP(hipSetDevice(d.gpu()));
+  mhd_substep_launch_on(d, region, step, dt, cf, eng.compute_stream(dom, streamId),
+                        eng.compute_stream(dom, 1 - streamId));
+}
+
+namespace {
+
+// Whole-substep replay graphs for the single-process single-domain MHD
+// path (the world=1 astaroth shape). Each RK3 substep s captures, per
+// buffer parity: [X1 translates (field halos, group 0) -> div pass ->
+// X2 translates (div halos, group 1) -> scalar || momentum (event-forked
+// second stream) -> device-side table swap]. The host gap measured
+// ~0.45 ms per substep at 256^3 (profiles/astaroth_256_kernel_stats.csv
+// GPU-busy vs wall); replay costs one hipGraphLaunch.
+struct MhdStepGraph {
+  hipStream_t stream = nullptr;  // capture/replay stream
+  hipStream_t stream2 = nullptr; // momentum fork
+  hipEvent_t evFork = nullptr, evJoin = nullptr;
+  hipGraphExec_t exec[3][2] = {{nullptr, nullptr}, {nullptr, nullptr}, {nullptr, nullptr}};
+  int parity = 0;
+  ExchangeEngine *eng = nullptr;
+  int dom = 0;
+};
+std::vector<std::unique_ptr<MhdStepGraph>> g_mhdGraphs;
+
+} // namespace
+
+int64_t mhd_graph_create(ExchangeEngine &eng, int dom, const Rect3 &region, double dt,
+                         const MhdCoeffs &cf) {
+  LocalDomain &d = eng.domain(dom);
+  STENCIL_HIP(hipSetDevice(d.gpu()));
+  auto sg = std::make_unique<MhdStepGraph>();
+  sg->eng = &eng;
+  sg->dom = dom;
+  STENCIL_HIP(hipStreamCreateWithFlags(&sg->stream, hipStreamNonBlocking));
+  STENCIL_HIP(hipStreamCreateWithFlags(&sg->stream2, hipStreamNonBlocking));
+  STENCIL_HIP(hipEventCreateWithFlags(&sg->evFork, hipEventDisableTiming));
+  STENCIL_HIP(hipEventCreateWithFlags(&sg->evJoin, hipEventDisableTiming));
+  for (int s = 0; s < 3; ++s) {
+    for (int par = 0; par < 2; ++par) {
+      STENCIL_HIP(hipStreamBeginCapture(sg->stream, hipStreamCaptureModeThreadLocal));
+      eng.launch_translates_plain_on((uintptr_t)sg->stream, 0); // X1: field halos
+      mhd_div_launch_on(d, region, cf, sg->stream);
+      eng.launch_translates_plain_on((uintptr_t)sg->stream, 1); // X2: div halos
+      STENCIL_HIP(hipEventRecord(sg->evFork, sg->stream));
+      STENCIL_HIP(hipStreamWaitEvent(sg->stream2, sg->evFork, 0));
+      mhd_substep_launch_on(d, region, s, dt, cf, sg->stream, sg->stream2);
+      STENCIL_HIP(hipEventRecord(sg->evJoin, sg->stream2));
+      STENCIL_HIP(hipStreamWaitEvent(sg->stream, sg->evJoin, 0));
+      d.enqueue_table_swap(sg->stream);
+      hipGraph_t g = nullptr;
+      STENCIL_HIP(hipStreamEndCapture(sg->stream, &g));
+      STENCIL_HIP(hipGraphInstantiate(&sg->exec[s][par], g, nullptr, nullptr, 0));
+      STENCIL_HIP(hipGraphDestroy(g));
+      d.swap(); // bake the other parity next
+    }
+  }
+  g_mhdGraphs.push_back(std::move(sg));
+  return (int64_t)g_mhdGraphs.size() - 1;
+}
+
+void mhd_graph_iter(int64_t handle, int64_t nIters) {
+  MhdStepGraph &sg = *g_mhdGraphs.at(handle);
+  LocalDomain &d = sg.eng->domain(sg.dom);
+  STENCIL_HIP(hipSetDevice(d.gpu()));
+  for (int64_t i = 0; i < nIters; ++i)
+    for (int s = 0; s < 3; ++s) {
+      STENCIL_HIP(hipGraphLaunch(sg.exec[s][sg.parity], sg.stream));
+      sg.parity ^= 1;
+      d.swap_host_only();
+    }
+}
+
+void mhd_graph_sync(int64_t handle) {
+  MhdStepGraph &sg = *g_mhdGraphs.at(handle);
+  STENCIL_HIP(hipSetDevice(sg.eng->domain(sg.dom).gpu()));
+  STENCIL_HIP(hipStreamSynchronize(sg.stream));
 }
 
 } // namespace stencil_amd

@@ -240,6 +240,7 @@ def test_jacobi_step_graph_matches_eager(monkeypatch):
         monkeypatch.setenv("STENCIL_AMD_STEP_GRAPH", "1" if mode == "graph" else "0")
         app = Jacobi3D(size, backend="native", gpus=[0])
         app.realize()
+        fill_interiors(app.dd, app.h)  # match _run_jacobi's initial state
         if mode == "graph":
             assert app._graph is not None, "graph path did not activate"
         else:
