@@ -168,6 +168,17 @@ ExchangeEngine::ExchangeEngine(std::vector<std::shared_ptr<LocalDomain>> domains
     : domains_(std::move(domains)) {
   computeStreams_.resize(domains_.size(), nullptr);
   computeStreams2_.resize(domains_.size(), nullptr);
+  // spin-wait host syncs: interrupt-based stream-sync wakeups cost
+  // 20-100 us each on ROCm; an HPC bench prefers burning the core.
+  // Tolerant: the flag may be rejected once a device context exists.
+  const char *spin = getenv("STENCIL_AMD_SPIN");
+  if (!spin || spin[0] != '0') {
+    for (auto &d : domains_) {
+      if (hipSetDevice(d->gpu()) != hipSuccess) continue;
+      (void)hipSetDeviceFlags(hipDeviceScheduleSpin);
+      (void)hipGetLastError(); // clear hipErrorSetOnActiveProcess
+    }
+  }
 }
 
 ExchangeEngine::~ExchangeEngine() {

@@ -226,6 +226,16 @@ class Astaroth:
         for s in range(3):
             self._substep(s, dt, compute, overlap)
 
+    def run(self, n: int):
+        """n full RK3 iterations with one host sync at the end (graph
+        mode queues 3n substep replays back-to-back)"""
+        if self._graph is not None and self.dd.comm.world_size == 1:
+            _C.mhd_graph_iter(self._graph, n)
+            _C.mhd_graph_sync(self._graph)
+            return
+        for _ in range(n):
+            self.step()
+
     def read_field(self, li: int, name: str) -> np.ndarray:
         lo, hi = self.dd.local_rect(li)
         return self.dd.read_global(li, lo, hi, self.handles[FIELDS.index(name)])
