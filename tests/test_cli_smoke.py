@@ -64,3 +64,24 @@ def test_bench_qap_cli():
     out = run_cli("bench_qap.py", "--n", "6")
     assert out.returncode == 0, out.stderr
     assert out.stdout.startswith("case,")
+
+
+def test_model_run_equals_steps_torch():
+    """Jacobi3D.run(n) (pipelined API) must equal n x step() on the torch
+    fallback path (graph mode is GPU-only; see test_gpu_native for that)"""
+    import numpy as np
+
+    from stencil_amd.models.jacobi3d import Jacobi3D
+
+    outs = []
+    for mode in ("run", "step"):
+        app = Jacobi3D((14, 12, 10), backend="torch", gpus=[0])
+        app.realize()
+        if mode == "run":
+            app.run(3)
+        else:
+            for _ in range(3):
+                app.step()
+        lo, hi = app.dd.local_rect(0)
+        outs.append(app.dd.read_global(0, lo, hi, app.h))
+    np.testing.assert_array_equal(outs[0], outs[1])
