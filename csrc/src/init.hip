@@ -3,6 +3,7 @@
 // give deterministic smooth initial conditions reproducible in NumPy).
 #include <hip/hip_runtime.h>
 
+#include "stencil_amd/accessor.hpp"
 #include "stencil_amd/device_util.hpp"
 #include "stencil_amd/domain.hpp"
 #include "stencil_amd/engine.hpp"
@@ -14,9 +15,7 @@ namespace stencil_amd {
 namespace {
 
 struct InitParams {
-  char *base;
-  int64_t pitch, plane;
-  int64_t allocX, allocY, allocZ;
+  Accessor<double> acc; // grid-coordinate accessor (reference accessor.hpp)
   int64_t loX, loY, loZ;
   int32_t extX, extY, extZ;
   double base0, amp;
@@ -29,10 +28,8 @@ __global__ void __launch_bounds__(256) init_harmonic_f64_kernel(InitParams p) {
   const int32_t ly = blockIdx.y * blockDim.y + threadIdx.y;
   const int32_t lz = blockIdx.z;
   if (lx >= p.extX || ly >= p.extY || lz >= p.extZ) return;
-  const int64_t gx = p.loX + lx, gy = p.loY + ly, gz = p.loZ + lz;
-  const double v = p.base0 + p.amp * sin(p.kx * gx + p.ky * gy + p.kz * gz + p.phase);
-  *(double *)(p.base + (gz - p.allocZ) * p.plane + (gy - p.allocY) * p.pitch +
-              (gx - p.allocX) * 8) = v;
+  const Vec3 g(p.loX + lx, p.loY + ly, p.loZ + lz);
+  p.acc[g] = p.base0 + p.amp * sin(p.kx * g.x + p.ky * g.y + p.kz * g.z + p.phase);
 }
 
 } // namespace
@@ -44,13 +41,7 @@ void init_harmonic_f64(ExchangeEngine &eng, int dom, int64_t qi, const Rect3 &re
   const Vec3 ext = region.extent();
   if (ext.flatten() <= 0) return;
   InitParams p{};
-  p.base = nextBuf ? d.next(qi).ptr : d.curr(qi).ptr;
-  p.pitch = d.curr(qi).pitch;
-  p.plane = d.curr(qi).plane();
-  const Rect3 full = d.full_region();
-  p.allocX = full.lo.x;
-  p.allocY = full.lo.y;
-  p.allocZ = full.lo.z;
+  p.acc = nextBuf ? next_accessor<double>(d, qi) : curr_accessor<double>(d, qi);
   p.loX = region.lo.x;
   p.loY = region.lo.y;
   p.loZ = region.lo.z;
