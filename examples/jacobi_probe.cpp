@@ -63,11 +63,57 @@ __global__ void __launch_bounds__(256) probe(P p) {
       cm = cc;
       cc = cp;
     }
-    vfloat4 ov = {out.x, out.y, out.z, out.w};
-    __builtin_nontemporal_store(ov, (vfloat4 *)dcol);
+    if (VARIANT == 3) { // full stencil, PLAIN store (NT ablation)
+      *(float4 *)dcol = out;
+    } else {
+      vfloat4 ov = {out.x, out.y, out.z, out.w};
+      __builtin_nontemporal_store(ov, (vfloat4 *)dcol);
+    }
     if (VARIANT == 0) {
       cc = *(const float4 *)(col + p.plane);
     }
+    col += p.plane;
+    dcol += p.plane;
+  }
+}
+
+// absolute bandwidth ceiling: flat linear float4 stream over the same bytes
+__global__ void __launch_bounds__(256) probe_lin(const char *src, char *dst, int64_t n16) {
+  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n16; i += stride) {
+    vfloat4 v = *(const vfloat4 *)(src + i * 16);
+    __builtin_nontemporal_store(v, (vfloat4 *)(dst + i * 16));
+  }
+}
+
+// full stencil with the DISPATCH ORDER swapped: z-chunks advance fastest
+// across blockIdx.y, y-rows across blockIdx.z (does walk order engage
+// DRAM channels differently?)
+__global__ void __launch_bounds__(256) probe_swap(P p) {
+  const int32_t u = blockIdx.x * blockDim.x + threadIdx.x;
+  const int32_t y = blockIdx.z * blockDim.y + threadIdx.y;
+  const int32_t z0 = blockIdx.y * ZCH;
+  if (u >= p.nx4 || y >= p.ny) return;
+  const int32_t zEnd = min(z0 + ZCH, p.nz);
+  const char *col = p.src + (int64_t)(z0 + 3) * p.plane + (int64_t)(y + 3) * p.pitch + 16 + u * 16;
+  char *dcol = p.dst + (int64_t)(z0 + 3) * p.plane + (int64_t)(y + 3) * p.pitch + 16 + u * 16;
+  float4 cm = *(const float4 *)(col - p.plane);
+  float4 cc = *(const float4 *)(col);
+  for (int32_t z = z0; z < zEnd; ++z) {
+    const float4 cp = *(const float4 *)(col + p.plane);
+    const float left = *(const float *)(col - 4);
+    const float right = *(const float *)(col + 16);
+    const float4 py = *(const float4 *)(col + p.pitch);
+    const float4 my = *(const float4 *)(col - p.pitch);
+    float4 out;
+    out.x = (cc.y + left + py.x + my.x + cp.x + cm.x) / 6.0f;
+    out.y = (cc.z + cc.x + py.y + my.y + cp.y + cm.y) / 6.0f;
+    out.z = (cc.w + cc.y + py.z + my.z + cp.z + cm.z) / 6.0f;
+    out.w = (right + cc.z + py.w + my.w + cp.w + cm.w) / 6.0f;
+    vfloat4 ov = {out.x, out.y, out.z, out.w};
+    __builtin_nontemporal_store(ov, (vfloat4 *)dcol);
+    cm = cc;
+    cc = cp;
     col += p.plane;
     dcol += p.plane;
   }
@@ -206,9 +252,13 @@ int main(int argc, char **argv) {
   CHECK(hipMemcpy(slots + 1, &b, sizeof(char *), hipMemcpyHostToDevice));
   dim3 grdp((p.nx4 + 2 + 63) / 64, (p.ny + 3) / 4, (p.nz + ZCH - 1) / ZCH);
 
-  const int NV = 7;
-  const char *names[NV] = {"copy", "xz", "full", "full8", "prod-notail", "prod+sph+tail", "prod+tail"};
-  double best[NV] = {1e30, 1e30, 1e30, 1e30, 1e30, 1e30, 1e30};
+  const int64_t n16 = p.plane * n / 16;
+  dim3 grdsw(grd.x, grd.z, grd.y);
+  const int NV = 10;
+  const char *names[NV] = {"copy", "xz",   "full",    "full8",   "prod-notail",
+                           "prod+sph+tail", "prod+tail", "linstream", "full-plainst", "full-dispswap"};
+  double best[NV];
+  for (int v = 0; v < NV; ++v) best[v] = 1e30;
   for (int r = 0; r < rounds; ++r) {
     for (int v = 0; v < NV; ++v) {
       CHECK(hipEventRecord(e0));
@@ -216,6 +266,9 @@ int main(int argc, char **argv) {
       case 0: hipLaunchKernelGGL(probe<0>, grd, blk, 0, 0, p); break;
       case 1: hipLaunchKernelGGL(probe<1>, grd, blk, 0, 0, p); break;
       case 2: hipLaunchKernelGGL(probe<2>, grd, blk, 0, 0, p); break;
+      case 7: hipLaunchKernelGGL(probe_lin, dim3(256 * 16), dim3(256), 0, 0, a, b, n16); break;
+      case 8: hipLaunchKernelGGL(probe<3>, grd, blk, 0, 0, p); break;
+      case 9: hipLaunchKernelGGL(probe_swap, grdsw, blk, 0, 0, p); break;
       case 3: hipLaunchKernelGGL(probe8, grd8, blk, 0, 0, p); break;
       case 4:
         hipLaunchKernelGGL(probe_prod, grdp, blk, 0, 0, p, (const char *const *)slots,
@@ -237,8 +290,10 @@ int main(int argc, char **argv) {
       best[v] = std::min(best[v], (double)ms);
     }
   }
-  for (int v = 0; v < NV; ++v)
-    printf("%-6s %8.3f ms  %7.1f Gcell/s  %6.2f TB/s(8B/cell)\n", names[v], best[v],
-           cells / best[v] / 1e6, cells * 8 / best[v] / 1e9);
+  for (int v = 0; v < NV; ++v) {
+    const double bytes = (v == 7) ? 2.0 * n16 * 16 : cells * 8; // linstream: full alloc R+W
+    printf("%-14s %8.3f ms  %7.1f Gcell/s  %6.2f TB/s\n", names[v], best[v],
+           cells / best[v] / 1e6, bytes / best[v] / 1e9);
+  }
   return 0;
 }
