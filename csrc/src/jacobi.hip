@@ -41,6 +41,7 @@ struct JacobiParams {
   // region to compute, global coords
   int64_t loX, loY, loZ;
   int32_t extX, extY, extZ;
+  int32_t vecAll; // 1 = pure-vector launch (see launch_jacobi_on)
   // whole compute region (for the hot/cold spheres)
   int64_t cLoX, cLoY, cLoZ, cHiX, cHiY, cHiZ;
 };
@@ -133,7 +134,11 @@ __global__ void __launch_bounds__(256) jacobi_kernel_v4(JacobiParams p) {
   };
 
   const int64_t a0 = p.loX - p.allocX;
-  const int32_t head = (int32_t)(((a0 + 3) & ~3LL) - a0) > p.extX
+  // vecAll: host pre-adjusted loX/extX (16B-aligned start, extX%4==0,
+  // writes beyond the true region land only in halo/row-slack bytes that
+  // the next exchange refreshes before any read) -- no scalar lanes
+  const int32_t head = p.vecAll ? 0
+                       : (int32_t)(((a0 + 3) & ~3LL) - a0) > p.extX
                            ? p.extX
                            : (int32_t)(((a0 + 3) & ~3LL) - a0);
   const int32_t body4 = (p.extX - head) / 4;
@@ -238,9 +243,9 @@ namespace {
 // parity and enqueues the right kernel variant onto `stream`. Also used
 // by the whole-step hipGraph capture (pointers get baked per parity).
 void launch_jacobi_on(LocalDomain &d, int64_t qi, const Rect3 &region,
-                      const Rect3 &computeRegion, hipStream_t stream) {
+                      const Rect3 &computeRegion, hipStream_t stream, bool fullRectVec = false) {
   if (d.elem_size(qi) != 4) throw std::runtime_error("jacobi_step: quantity must be fp32");
-  const Vec3 ext = region.extent();
+  Vec3 ext = region.extent();
   if (ext.flatten() <= 0) return;
   JacobiParams p{};
   p.src = d.curr(qi).ptr;
@@ -263,11 +268,31 @@ void launch_jacobi_on(LocalDomain &d, int64_t qi, const Rect3 &region,
   p.cHiX = computeRegion.hi.x;
   p.cHiY = computeRegion.hi.y;
   p.cHiZ = computeRegion.hi.z;
+  // fullRectVec: the region is the whole interior rect (radius >= 1 on
+  // both x sides). Extend it left to the previous 16 B-aligned ADDRESS
+  // and right to a multiple of 4, eliminating the scalar head/tail lanes
+  // (the probe measured them at ~10% of the kernel). The extended cells
+  // are x-halo cells or pitch-slack bytes of the adjacent row: every one
+  // is rewritten by the next exchange (or is padding) before any read,
+  // and never a translate SOURCE (sources are interior cells).
+  if (fullRectVec && ext.x >= 8) {
+    const int64_t a0 = region.lo.x - full.lo.x;
+    const uintptr_t addr0 = (uintptr_t)p.src + a0 * 4;
+    const int64_t under = (int64_t)((addr0 & 15) >> 2);
+    int64_t e2 = ext.x + under;
+    const int64_t over = (4 - (e2 & 3)) & 3;
+    e2 += over;
+    p.loX = region.lo.x - under;
+    p.extX = (int32_t)e2;
+    p.vecAll = 1;
+    ext.x = e2;
+  }
   if (ext.x >= 8 && ext.y <= 0x7fffffff) {
     // vectorized row-mapped kernel; block shape tunable via env
-    const int64_t a0 = region.lo.x - full.lo.x;
-    const int64_t head = std::min<int64_t>(((a0 + 3) & ~3LL) - a0, ext.x);
-    const int64_t units = (ext.x - head) / 4 + 2;
+    const int64_t a0 = p.loX - full.lo.x;
+    const int64_t head =
+        p.vecAll ? 0 : std::min<int64_t>(((a0 + 3) & ~3LL) - a0, ext.x);
+    const int64_t units = (ext.x - head) / 4 + (p.vecAll ? 0 : 2);
     static int bx = 0, by = 0;
     if (!bx) {
       bx = 64;
@@ -333,7 +358,9 @@ int64_t jacobi_graph_create(ExchangeEngine &eng, int dom, int64_t qi, const Rect
   for (int par = 0; par < 2; ++par) {
     STENCIL_HIP(hipStreamBeginCapture(sg->stream, hipStreamCaptureModeThreadLocal));
     eng.launch_translates_plain_on((uintptr_t)sg->stream, 0);
-    launch_jacobi_on(d, qi, region, computeRegion, sg->stream);
+    // the graph's region is the whole interior rect: pure-vector launch
+    // (scalar tail lanes measured ~10% of the kernel in the probe)
+    launch_jacobi_on(d, qi, region, computeRegion, sg->stream, /*fullRectVec=*/true);
     d.enqueue_table_swap(sg->stream);
     hipGraph_t g = nullptr;
     STENCIL_HIP(hipStreamEndCapture(sg->stream, &g));
