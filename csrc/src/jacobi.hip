@@ -136,11 +136,14 @@ __global__ void __launch_bounds__(256) jacobi_kernel_v4(JacobiParams p) {
   const int64_t a0 = p.loX - p.allocX;
   // vecAll: host pre-adjusted loX/extX (16B-aligned start, extX%4==0,
   // writes beyond the true region land only in halo/row-slack bytes that
-  // the next exchange refreshes before any read) -- no scalar lanes
-  const int32_t head = p.vecAll ? 0
-                       : (int32_t)(((a0 + 3) & ~3LL) - a0) > p.extX
-                           ? p.extX
-                           : (int32_t)(((a0 + 3) & ~3LL) - a0);
+  // the next exchange refreshes before any read) -- no scalar lanes.
+  // Otherwise head = elements until the next 16 B-aligned ADDRESS (the
+  // allocation pad shifts the base, so index-aligned units would issue
+  // misaligned dwordx4 -- measured ~10% on gfx950); must match the
+  // launcher's units formula.
+  const int32_t headAl =
+      (int32_t)((16 - (((uintptr_t)srcBase + (uintptr_t)(a0 * 4)) & 15)) & 15) >> 2;
+  const int32_t head = p.vecAll ? 0 : (headAl > p.extX ? p.extX : headAl);
   const int32_t body4 = (p.extX - head) / 4;
   const int32_t tail = p.extX - head - body4 * 4;
 
@@ -290,8 +293,9 @@ void launch_jacobi_on(LocalDomain &d, int64_t qi, const Rect3 &region,
   if (ext.x >= 8 && ext.y <= 0x7fffffff) {
     // vectorized row-mapped kernel; block shape tunable via env
     const int64_t a0 = p.loX - full.lo.x;
-    const int64_t head =
-        p.vecAll ? 0 : std::min<int64_t>(((a0 + 3) & ~3LL) - a0, ext.x);
+    const int64_t headAl =
+        (int64_t)((16 - (((uintptr_t)p.src + (uintptr_t)(a0 * 4)) & 15)) & 15) >> 2;
+    const int64_t head = p.vecAll ? 0 : std::min<int64_t>(headAl, ext.x);
     const int64_t units = (ext.x - head) / 4 + (p.vecAll ? 0 : 2);
     static int bx = 0, by = 0;
     if (!bx) {
