@@ -12,6 +12,8 @@ echo "=== smoke() ===" >> $LOG
 timeout 300 python -c "import __graft_entry__; __graft_entry__.smoke()" >> $LOG 2>&1
 echo "=== bench.py final (default flags) ===" >> $LOG
 timeout 200 python bench.py 2>&1 | grep -E '^\{' >> $LOG
+echo "=== eager-path bench (address-aligned head fix) ===" >> $LOG
+STENCIL_AMD_STEP_GRAPH=0 timeout 200 python bench.py --gpus 1 --steps 20 --warmup 3 2>&1 | grep -E '^\{' >> $LOG
 echo "=== astaroth final ===" >> $LOG
 timeout 200 python benchmarks/astaroth.py --gpus 1 --per-gpu 256 --iters 10 --warmup 2 2>&1 | grep astaroth, >> $LOG
 echo "=== soak: jacobi graph 500 iters + finite check ===" >> $LOG
