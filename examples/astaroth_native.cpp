@@ -47,7 +47,8 @@ int main(int argc, char **argv) {
   eng.finalize();
 
   const Rect3 r = dom->compute_region();
-  MhdCoeffs cf; // defaults match stencil_amd/models/astaroth.py DEFAULT_CONF
+  MhdCoeffs cf; // astaroth.conf-style spacing (models/astaroth.py DEFAULT_CONF)
+  cf.dsx = cf.dsy = cf.dsz = 0.04908738521; // 2*pi/128
   for (int qi = 0; qi < 8; ++qi) {
     const double amp = qi == 0 ? 0.01 : 1e-3;
     init_harmonic_f64(eng, 0, qi, r, 0.0, amp, 2 * M_PI * (1 + qi % 3) / n,
