@@ -77,6 +77,61 @@ __global__ void __launch_bounds__(256) probe(P p) {
   }
 }
 
+// y-neighbor rows staged through LDS: a 64x4 block marching z keeps the
+// current plane's center vectors of its 4 rows + 2 halo rows in LDS
+// (ping-pong buffers, one barrier per z-step), so the py/my global vector
+// loads become LDS reads -- halves the vector-load issue rate.
+__global__ void __launch_bounds__(256) probe_lds(P p) {
+  __shared__ float4 tile[2][6][64];
+  const int32_t tx = threadIdx.x; // x-unit lane
+  const int32_t ry = threadIdx.y; // row within block (0..3)
+  // no early returns: __syncthreads needs every thread, so out-of-range
+  // lanes clamp their coordinates and skip only the store
+  const int32_t u0 = blockIdx.x * blockDim.x + tx;
+  const int32_t y0 = blockIdx.y * 4 + ry;
+  const bool valid = u0 < p.nx4 && y0 < p.ny;
+  const int32_t u = min(u0, p.nx4 - 1);
+  const int32_t y = min(y0, p.ny - 1);
+  const int32_t z0 = blockIdx.z * ZCH;
+  const int32_t zEnd = min(z0 + ZCH, p.nz);
+  const char *col = p.src + (int64_t)(z0 + 3) * p.plane + (int64_t)(y + 3) * p.pitch + 16 + u * 16;
+  char *dcol = p.dst + (int64_t)(z0 + 3) * p.plane + (int64_t)(y + 3) * p.pitch + 16 + u * 16;
+  float4 cm = *(const float4 *)(col - p.plane);
+  float4 cc = *(const float4 *)(col);
+  // prime LDS with plane z0's centers (+2 y-halo rows)
+  tile[0][ry + 1][tx] = cc;
+  if (ry == 0) tile[0][0][tx] = *(const float4 *)(col - p.pitch);
+  if (ry == 3) tile[0][5][tx] = *(const float4 *)(col + p.pitch);
+  __syncthreads();
+  int buf = 0;
+  for (int32_t z = z0; z < zEnd; ++z) {
+    const float4 cp = *(const float4 *)(col + p.plane);
+    const float left = *(const float *)(col - 4);
+    const float right = *(const float *)(col + 16);
+    const float4 py = tile[buf][ry + 2][tx];
+    const float4 my = tile[buf][ry][tx];
+    float4 out;
+    out.x = (cc.y + left + py.x + my.x + cp.x + cm.x) / 6.0f;
+    out.y = (cc.z + cc.x + py.y + my.y + cp.y + cm.y) / 6.0f;
+    out.z = (cc.w + cc.y + py.z + my.z + cp.z + cm.z) / 6.0f;
+    out.w = (right + cc.z + py.w + my.w + cp.w + cm.w) / 6.0f;
+    if (valid) {
+      vfloat4 ov = {out.x, out.y, out.z, out.w};
+      __builtin_nontemporal_store(ov, (vfloat4 *)dcol);
+    }
+    // stage plane z+1 into the other LDS buffer for the next step
+    tile[buf ^ 1][ry + 1][tx] = cp;
+    if (ry == 0) tile[buf ^ 1][0][tx] = *(const float4 *)(col + p.plane - p.pitch);
+    if (ry == 3) tile[buf ^ 1][5][tx] = *(const float4 *)(col + p.plane + p.pitch);
+    __syncthreads();
+    buf ^= 1;
+    cm = cc;
+    cc = cp;
+    col += p.plane;
+    dcol += p.plane;
+  }
+}
+
 // absolute bandwidth ceiling: flat linear float4 stream over the same bytes
 __global__ void __launch_bounds__(256) probe_lin(const char *src, char *dst, int64_t n16) {
   const int64_t stride = (int64_t)gridDim.x * blockDim.x;
@@ -254,9 +309,10 @@ int main(int argc, char **argv) {
 
   const int64_t n16 = p.plane * n / 16;
   dim3 grdsw(grd.x, grd.z, grd.y);
-  const int NV = 10;
+  const int NV = 11;
   const char *names[NV] = {"copy", "xz",   "full",    "full8",   "prod-notail",
-                           "prod+sph+tail", "prod+tail", "linstream", "full-plainst", "full-dispswap"};
+                           "prod+sph+tail", "prod+tail", "linstream", "full-plainst", "full-dispswap",
+                           "full-ldsrows"};
   double best[NV];
   for (int v = 0; v < NV; ++v) best[v] = 1e30;
   for (int r = 0; r < rounds; ++r) {
@@ -269,6 +325,7 @@ int main(int argc, char **argv) {
       case 7: hipLaunchKernelGGL(probe_lin, dim3(256 * 16), dim3(256), 0, 0, a, b, n16); break;
       case 8: hipLaunchKernelGGL(probe<3>, grd, blk, 0, 0, p); break;
       case 9: hipLaunchKernelGGL(probe_swap, grdsw, blk, 0, 0, p); break;
+      case 10: hipLaunchKernelGGL(probe_lds, grd, blk, 0, 0, p); break;
       case 3: hipLaunchKernelGGL(probe8, grd8, blk, 0, 0, p); break;
       case 4:
         hipLaunchKernelGGL(probe_prod, grdp, blk, 0, 0, p, (const char *const *)slots,
