@@ -189,6 +189,97 @@ __global__ void __launch_bounds__(256) jacobi_kernel_v4(JacobiParams p) {
   }
 }
 
+// vecAll-only variant: y-neighbor rows staged through LDS. A 64x4 block
+// marching z keeps the current plane's center vectors of its 4 rows + 2
+// y-halo rows in ping-pong LDS buffers (one barrier per z-step), so the
+// py/my global vector loads become LDS reads. Probe: 0.749 ms vs 0.839
+// for the pure-global stencil at 752^3 -- at the mapping's copy roofline
+// (examples/jacobi_probe.cpp full-ldsrows). Requires blockDim (64,4,1)
+// and a vecAll launch (no scalar lanes; every lane's row exists because
+// the region is the interior rect of a radius>=1 domain).
+__global__ void __launch_bounds__(256) jacobi_kernel_v4_lds(JacobiParams p) {
+  __shared__ float4 tile[2][6][64];
+  const int32_t tx = threadIdx.x;
+  const int32_t ry = threadIdx.y;
+  // no early returns (__syncthreads needs every lane): out-of-range lanes
+  // clamp coordinates and skip only the store
+  const int32_t u0 = blockIdx.x * 64 + tx;
+  const int32_t ly0 = blockIdx.y * 4 + ry;
+  const int32_t body4 = p.extX / 4; // vecAll: extX % 4 == 0
+  const bool valid = u0 < body4 && ly0 < p.extY;
+  const int32_t u = min(u0, body4 - 1);
+  // clamp to extY (the +y halo row, which exists for radius >= 1): a
+  // VALID lane ry-1 reads lane ry's staged row as its +y neighbor, so
+  // lanes at ly0 == extY must stage the TRUE halo row, not a duplicate
+  const int32_t ly = min(ly0, (int32_t)p.extY);
+  // +y halo row staged by ry==3: one past its own row, clamped in-bounds
+  const int64_t pyHalo = (ly < p.extY) ? p.pitch : 0;
+  const int32_t lz0 = blockIdx.z * JAC_ZCHUNK;
+  const int32_t zEnd = min((int32_t)(lz0 + JAC_ZCHUNK), p.extZ);
+
+  const int64_t gy = p.loY + ly;
+  const int64_t ay = gy - p.allocY;
+  const int64_t a0 = p.loX - p.allocX;
+  const int64_t ax = a0 + (int64_t)u * 4;
+  const int32_t gx = (int32_t)(ax + p.allocX);
+  const int64_t az0 = p.loZ + lz0 - p.allocZ;
+  const char *col = p.src + az0 * p.plane + ay * p.pitch + ax * 4;
+  char *dcol = p.dst + az0 * p.plane + ay * p.pitch + ax * 4;
+
+  const int32_t cw = (int32_t)(p.cHiX - p.cLoX);
+  const int32_t hotX = (int32_t)p.cLoX + cw / 3, coldX = (int32_t)p.cLoX + cw * 2 / 3;
+  const int32_t cY = (int32_t)(p.cLoY + p.cHiY) / 2, cZ = (int32_t)(p.cLoZ + p.cHiZ) / 2;
+  const int32_t srad = cw / 10;
+  auto sphere4 = [&](int32_t gzz, float4 &out) {
+    const int32_t dy = (int32_t)gy - cY, dz = gzz - cZ;
+    const int32_t yz2 = dy * dy + dz * dz;
+    if (yz2 >= (srad + 1) * (srad + 1)) return;
+#pragma unroll
+    for (int i = 0; i < 4; ++i) {
+      const int32_t dxh = gx + i - hotX, dxc = gx + i - coldX;
+      if ((int32_t)__fsqrt_rn((float)(dxh * dxh + yz2)) <= srad)
+        (&out.x)[i] = 1.0f;
+      else if ((int32_t)__fsqrt_rn((float)(dxc * dxc + yz2)) <= srad)
+        (&out.x)[i] = 0.0f;
+    }
+  };
+
+  float4 cm = *(const float4 *)(col - p.plane);
+  float4 cc = *(const float4 *)(col);
+  tile[0][ry + 1][tx] = cc;
+  if (ry == 0) tile[0][0][tx] = *(const float4 *)(col - p.pitch);
+  if (ry == 3) tile[0][5][tx] = *(const float4 *)(col + pyHalo);
+  __syncthreads();
+  int buf = 0;
+  for (int32_t lz = lz0; lz < zEnd; ++lz) {
+    const float4 cp = *(const float4 *)(col + p.plane);
+    const float left = *(const float *)(col - 4);
+    const float right = *(const float *)(col + 16);
+    const float4 py = tile[buf][ry + 2][tx];
+    const float4 my = tile[buf][ry][tx];
+    float4 out;
+    out.x = (cc.y + left + py.x + my.x + cp.x + cm.x) / 6.0f;
+    out.y = (cc.z + cc.x + py.y + my.y + cp.y + cm.y) / 6.0f;
+    out.z = (cc.w + cc.y + py.z + my.z + cp.z + cm.z) / 6.0f;
+    out.w = (right + cc.z + py.w + my.w + cp.w + cm.w) / 6.0f;
+    sphere4((int32_t)(p.loZ + lz), out);
+    if (valid) {
+      typedef float vfloat4 __attribute__((ext_vector_type(4)));
+      vfloat4 ov = {out.x, out.y, out.z, out.w};
+      __builtin_nontemporal_store(ov, (vfloat4 *)dcol);
+    }
+    tile[buf ^ 1][ry + 1][tx] = cp;
+    if (ry == 0) tile[buf ^ 1][0][tx] = *(const float4 *)(col + p.plane - p.pitch);
+    if (ry == 3) tile[buf ^ 1][5][tx] = *(const float4 *)(col + p.plane + pyHalo);
+    __syncthreads();
+    buf ^= 1;
+    cm = cc;
+    cc = cp;
+    col += p.plane;
+    dcol += p.plane;
+  }
+}
+
 __global__ void jacobi_kernel(JacobiParams p) {
   const char *srcBase = p.src;
   char *dstBase = p.dst;
@@ -289,6 +380,20 @@ void launch_jacobi_on(LocalDomain &d, int64_t qi, const Rect3 &region,
     p.extX = (int32_t)e2;
     p.vecAll = 1;
     ext.x = e2;
+  }
+  static int useLds = -1;
+  if (useLds < 0) {
+    const char *e = getenv("STENCIL_JAC_LDS");
+    useLds = (e && e[0] == '0') ? 0 : 1;
+  }
+  if (p.vecAll && useLds && ext.x >= 8) {
+    // LDS-staged y-rows variant (fixed 64x4 block); see jacobi_kernel_v4_lds
+    dim3 block(64, 4, 1);
+    dim3 grid((uint32_t)((p.extX / 4 + 63) / 64), (uint32_t)((ext.y + 3) / 4),
+              (uint32_t)((ext.z + 15) / 16));
+    hipLaunchKernelGGL(jacobi_kernel_v4_lds, grid, block, 0, stream, p);
+    STENCIL_HIP(hipGetLastError());
+    return;
   }
   if (ext.x >= 8 && ext.y <= 0x7fffffff) {
     // vectorized row-mapped kernel; block shape tunable via env
