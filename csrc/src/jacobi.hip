@@ -3,12 +3,18 @@
 //
 // The kernel is HBM-bandwidth-bound: per cell it reads the 6 face neighbors
 // + writes one value; with y/z-neighbor rows served from L2/L3 the traffic
-// floor is 4 B read + 4 B write per cell. Design choices for CDNA4:
-//   - linearized thread mapping with x fastest (any region shape stays
-//     coalesced, including 1-element-thick exterior slabs),
-//   - float4 vector path along x when the region is 4-aligned in x
-//     (Guideline 13: vectorize to 16 B/lane),
-//   - grid-stride with a capped grid (Guideline 11).
+// floor is 4 B read + 4 B write per cell (PMC-verified: 1.64 GB fetched /
+// 1.75 GB written per 750^3 step, profiles/jacobi_750_*_size.csv).
+// Design choices for CDNA4, each A/B-measured via examples/jacobi_probe:
+//   - float4 x-strips marching z with register-rolled center planes,
+//     nontemporal stores (plain stores 9% slower),
+//   - jacobi_kernel_v4_lds (the vecAll/graph path): y-neighbor rows
+//     staged through 12 KB of ping-pong LDS, full 8 waves/SIMD — runs at
+//     the mapping's copy roofline (527 Gcell/s @750^3),
+//   - pure-vector full-rect launches (tail lanes cost 10%) with
+//     address-based 16 B alignment,
+//   - linearized scalar kernel for thin exterior slabs (any shape stays
+//     coalesced), grid-stride with a capped grid.
 // Hot/cold sphere sources match the reference's behavior (truncated-int
 // sqrtf distance, HOT=1/COLD=0 fixed cells) so results are comparable.
 #include <hip/hip_runtime.h>
