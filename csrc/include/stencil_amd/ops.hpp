@@ -9,8 +9,15 @@ namespace stencil_amd {
 // One Jacobi 7-point step of quantity qi over `region` (global coords):
 // next = avg of 6 face neighbors of curr, with the reference's hot/cold
 // sphere sources fixed inside `computeRegion` (bin/jacobi3d.cu:40-85).
+// extendVec enables the pure-vector + LDS-rows fast path (1 = free
+// extension, 2 = strict/no-extension for IPC ranks) by
+// extending the row to aligned bounds: ONLY valid when every extended
+// cell (x-halo cells, pitch-slack bytes, or exterior-shell cells that a
+// LATER kernel on the SAME stream rewrites) is discard-safe and the
+// domain has radius >= 1 -- the caller asserts this (the overlap
+// interior and full-rect launches qualify).
 void jacobi_step(ExchangeEngine &eng, int dom, int64_t qi, const Rect3 &region,
-                 const Rect3 &computeRegion, int streamId = 0);
+                 const Rect3 &computeRegion, int streamId = 0, int extendVec = 0);
 
 // fill an fp32 region with `value` (curr or next buffer)
 void fill_f32(ExchangeEngine &eng, int dom, int64_t qi, const Rect3 &region, float value,

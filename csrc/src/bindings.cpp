@@ -260,7 +260,8 @@ PYBIND11_MODULE(_C, m) {
       });
 
   m.def("jacobi_step", &jacobi_step, py::arg("eng"), py::arg("dom"), py::arg("qi"),
-        py::arg("region"), py::arg("compute_region"), py::arg("stream_id") = 0);
+        py::arg("region"), py::arg("compute_region"), py::arg("stream_id") = 0,
+        py::arg("extend_vec") = 0);
   m.def("fill_f32", &fill_f32);
   m.def("jacobi_graph_create", &jacobi_graph_create, py::arg("eng"), py::arg("dom"),
         py::arg("qi"), py::arg("region"), py::arg("compute_region"));

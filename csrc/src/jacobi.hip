@@ -342,8 +342,15 @@ namespace {
 // shared launch logic: builds params from the domain's CURRENT buffer
 // parity and enqueues the right kernel variant onto `stream`. Also used
 // by the whole-step hipGraph capture (pointers get baked per parity).
+// fullRectVec: 0 = off; 1 = free extension (single-process or RCCL-wire
+// ranks: extended writes land in refresh-before-read halo/slack bytes or
+// exterior cells a later same-stream kernel rewrites); 2 = strict (IPC
+// ranks: a fast peer writes our NEXT-buffer halos during our compute, so
+// no extension is allowed -- the fast path engages only when the region
+// is already aligned, which the allocation pad arranges for the overlap
+// interior).
 void launch_jacobi_on(LocalDomain &d, int64_t qi, const Rect3 &region,
-                      const Rect3 &computeRegion, hipStream_t stream, bool fullRectVec = false) {
+                      const Rect3 &computeRegion, hipStream_t stream, int fullRectVec = 0) {
   if (d.elem_size(qi) != 4) throw std::runtime_error("jacobi_step: quantity must be fp32");
   Vec3 ext = region.extent();
   if (ext.flatten() <= 0) return;
@@ -382,10 +389,12 @@ void launch_jacobi_on(LocalDomain &d, int64_t qi, const Rect3 &region,
     int64_t e2 = ext.x + under;
     const int64_t over = (4 - (e2 & 3)) & 3;
     e2 += over;
-    p.loX = region.lo.x - under;
-    p.extX = (int32_t)e2;
-    p.vecAll = 1;
-    ext.x = e2;
+    if (fullRectVec != 2 || (under == 0 && over == 0)) {
+      p.loX = region.lo.x - under;
+      p.extX = (int32_t)e2;
+      p.vecAll = 1;
+      ext.x = e2;
+    }
   }
   static int useLds = -1;
   if (useLds < 0) {
@@ -433,10 +442,11 @@ void launch_jacobi_on(LocalDomain &d, int64_t qi, const Rect3 &region,
 } // namespace
 
 void jacobi_step(ExchangeEngine &eng, int dom, int64_t qi, const Rect3 &region,
-                 const Rect3 &computeRegion, int streamId) {
+                 const Rect3 &computeRegion, int streamId, int extendVec) {
   LocalDomain &d = eng.domain(dom);
   STENCIL_HIP(hipSetDevice(d.gpu()));
-  launch_jacobi_on(d, qi, region, computeRegion, eng.compute_stream(dom, streamId));
+  launch_jacobi_on(d, qi, region, computeRegion, eng.compute_stream(dom, streamId),
+                   extendVec);
 }
 
 namespace {
@@ -475,7 +485,7 @@ int64_t jacobi_graph_create(ExchangeEngine &eng, int dom, int64_t qi, const Rect
     eng.launch_translates_plain_on((uintptr_t)sg->stream, 0);
     // the graph's region is the whole interior rect: pure-vector launch
     // (scalar tail lanes measured ~10% of the kernel in the probe)
-    launch_jacobi_on(d, qi, region, computeRegion, sg->stream, /*fullRectVec=*/true);
+    launch_jacobi_on(d, qi, region, computeRegion, sg->stream, /*fullRectVec=*/1);
     d.enqueue_table_swap(sg->stream);
     hipGraph_t g = nullptr;
     STENCIL_HIP(hipStreamEndCapture(sg->stream, &g));

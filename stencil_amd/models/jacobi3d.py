@@ -105,25 +105,43 @@ class Jacobi3D:
 
     def _eager_step(self, overlap: bool = True):
         dd = self.dd
+        # pure-vector + LDS-rows fast path (see csrc/src/jacobi.hip
+        # launch_jacobi_on). Mode 1 (free row extension) is valid when the
+        # extended cells are discard-safe LOCALLY: single process, or
+        # multi-rank without direct IPC writes (an IPC peer running one
+        # exchange ahead writes our next-buffer halos during our compute,
+        # which the extension would race). IPC ranks use mode 2: the fast
+        # kernel engages only when the region is already aligned -- which
+        # the allocation pad arranges for the overlap interior. m>1 is
+        # excluded: temporal blocking keeps halo data live across phases.
+        ext_mode = 0
+        if dd.backend_kind == "native" and self.m == 1:
+            ext_mode = 2 if getattr(dd.backend, "_ipc_active", False) else 1
         if overlap:
             # interior compute (on compute streams) overlaps the exchange
             for li in range(dd.num_local()):
                 ilo, ihi = self.interiors[li]
-                dd.backend.jacobi_step(li, self.h.index, ilo, ihi, self.compute_lo, self.compute_hi)
+                dd.backend.jacobi_step(li, self.h.index, ilo, ihi, self.compute_lo,
+                                       self.compute_hi, extend_vec=ext_mode)
             dd.exchange()
-            # exterior shells go on the second compute stream: they only
-            # depend on the exchange (synced above), so they overlap the
-            # still-running interior kernel
+            # exterior shells: in mode 1 the +-x slabs intersect the
+            # interior's row extension, so they go on stream 0 ORDERED
+            # AFTER the interior kernel (their correct values must land
+            # last); y/z slabs only touch rows the interior never writes
+            # and stay concurrent on the second compute stream
             for li in range(dd.num_local()):
                 for blo, bhi in self.exteriors[li]:
+                    x_slab = ext_mode == 1 and (bhi[0] - blo[0]) <= 2
                     dd.backend.jacobi_step(
-                        li, self.h.index, blo, bhi, self.compute_lo, self.compute_hi, stream_id=1
+                        li, self.h.index, blo, bhi, self.compute_lo, self.compute_hi,
+                        stream_id=0 if x_slab else 1,
                     )
         else:
             dd.exchange()
             for li in range(dd.num_local()):
                 lo, hi = dd.local_rect(li)
-                dd.backend.jacobi_step(li, self.h.index, lo, hi, self.compute_lo, self.compute_hi)
+                dd.backend.jacobi_step(li, self.h.index, lo, hi, self.compute_lo,
+                                       self.compute_hi, extend_vec=ext_mode)
         dd.backend.sync_compute()
         dd.swap()
 

@@ -435,9 +435,9 @@ class NativeBackend:
         _C.fill_f32(self.engine, li, qi, _rect3(region_lo, region_hi), value, next_buf)
 
     def jacobi_step(self, li: int, qi: int, region_lo: Vec, region_hi: Vec,
-                    c_lo: Vec, c_hi: Vec, stream_id: int = 0):
+                    c_lo: Vec, c_hi: Vec, stream_id: int = 0, extend_vec: bool = False):
         _C.jacobi_step(self.engine, li, qi, _rect3(region_lo, region_hi), _rect3(c_lo, c_hi),
-                       stream_id)
+                       stream_id, extend_vec)
 
     def jacobi_graph_create(self, li: int, qi: int, region_lo: Vec, region_hi: Vec,
                             c_lo: Vec, c_hi: Vec) -> int:

@@ -180,7 +180,8 @@ class TorchBackend:
         ext = tuple(region_hi[i] - region_lo[i] for i in range(3))
         dom.region(qi, pos, ext, next_buf).fill_(value)
 
-    def jacobi_step(self, li, qi, region_lo, region_hi, c_lo, c_hi, stream_id=0):
+    def jacobi_step(self, li, qi, region_lo, region_hi, c_lo, c_hi, stream_id=0,
+                    extend_vec=0):  # extend_vec: native-only fast-path hint, ignored here
         """reference 7-point Jacobi with hot/cold spheres (fp32)"""
         dom = self.domains[li]
         flo = dom.full_lo()
