@@ -142,7 +142,6 @@ def test_multi_rank_jacobi_matches_torch(use_ipc, port):
     for rank, status, out in results:
         assert status == "ok", f"rank {rank}: {status}"
         for lo, hi, arr in out:
-            want = np.zeros_like(arr)
             # read the same global box from the torch reference
             got_ref = None
             for li in range(ref.dd.num_local()):
