@@ -55,6 +55,7 @@ struct MhdParams {
   double alpha_over_beta_prev; // alpha_s / beta_{s-1} (0 for substep 0)
   double beta;
   int32_t swizzle; // XCD-aware block remap (see xcd_remap)
+  int32_t ychunk;  // y-chunked dispatch order in BLOCKS (see ychunk_remap)
   int32_t pad_;
 };
 
@@ -184,6 +185,22 @@ __device__ __forceinline__ void xcd_remap(int32_t &bx, int32_t &by, int32_t &bz)
   bz = nf / (gridDim.x * gridDim.y);
 }
 
+// y-chunked dispatch (STENCIL_MHD_YCHUNK=<blocks>): iterate the block
+// grid chunk-of-y-rows-major (for each y-chunk: all z, then y-in-chunk,
+// then x) so a z-slab's working set per XCD shrinks by nby/C -- the
+// momentum/scalar kernels refetch z-star planes ~3.4x because a 2-thick
+// slab of 10 fp64 fields (~5 MB/XCD at 256^3) thrashes the 4 MB XCD L2.
+// C must divide gridDim.y (host guarantees).
+__device__ __forceinline__ void ychunk_remap(int32_t C, int32_t &bx, int32_t &by, int32_t &bz) {
+  const int32_t flat = bx + gridDim.x * (by + gridDim.y * bz);
+  const int32_t per = gridDim.x * C * gridDim.z;
+  const int32_t chunk = flat / per, rem = flat % per;
+  bz = rem / (gridDim.x * C);
+  const int32_t rem2 = rem % (gridDim.x * C);
+  by = chunk * C + rem2 / gridDim.x;
+  bx = rem2 % gridDim.x;
+}
+
 __device__ __forceinline__ MhdCommon mhd_setup(const MhdParams &p, int32_t lx, int32_t ly,
                                                int32_t lz) {
   const int64_t ax = p.loX + lx - p.allocX;
@@ -213,6 +230,7 @@ __device__ __forceinline__ void write_rk3(const MhdParams &p, const Stencil &st,
 __global__ void __launch_bounds__(256) mhd_scalar_kernel(MhdParams p) {
   int32_t bx = blockIdx.x, by = blockIdx.y, bz = blockIdx.z;
   if (p.swizzle) xcd_remap(bx, by, bz);
+  if (p.ychunk) ychunk_remap(p.ychunk, bx, by, bz);
   const int32_t lx = bx * blockDim.x + threadIdx.x;
   const int32_t ly = by * blockDim.y + threadIdx.y;
   const int32_t lz = bz * blockDim.z + threadIdx.z;
@@ -253,6 +271,7 @@ __global__ void __launch_bounds__(256) mhd_scalar_kernel(MhdParams p) {
 __global__ void __launch_bounds__(256) mhd_div_kernel(MhdParams p) {
   int32_t bx = blockIdx.x, by = blockIdx.y, bz = blockIdx.z;
   if (p.swizzle) xcd_remap(bx, by, bz);
+  if (p.ychunk) ychunk_remap(p.ychunk, bx, by, bz);
   const int32_t lx = bx * blockDim.x + threadIdx.x;
   const int32_t ly = by * blockDim.y + threadIdx.y;
   const int32_t lz = bz * blockDim.z + threadIdx.z;
@@ -276,6 +295,7 @@ __global__ void __launch_bounds__(256) mhd_div_kernel(MhdParams p) {
 __global__ void __launch_bounds__(256) mhd_momentum_kernel(MhdParams p) {
   int32_t bx = blockIdx.x, by = blockIdx.y, bz = blockIdx.z;
   if (p.swizzle) xcd_remap(bx, by, bz);
+  if (p.ychunk) ychunk_remap(p.ychunk, bx, by, bz);
   const int32_t lx = bx * blockDim.x + threadIdx.x;
   const int32_t ly = by * blockDim.y + threadIdx.y;
   const int32_t lz = bz * blockDim.z + threadIdx.z;
@@ -355,6 +375,21 @@ static void mhd_fill_params(LocalDomain &d, const Rect3 &region, const MhdCoeffs
     swz = (e && e[0] == '1') ? 1 : 0;
   }
   p.swizzle = swz;
+  p.ychunk = 0; // set per-launch (needs the real grid) in the launchers
+}
+
+// STENCIL_MHD_YCHUNK=<blocks>: largest divisor of the y-block count that
+// is <= the request (C must divide gridDim.y for the in-kernel decode)
+static int32_t mhd_ychunk(int32_t nby) {
+  static int ych = -1;
+  if (ych < 0) {
+    const char *e = getenv("STENCIL_MHD_YCHUNK");
+    ych = e ? atoi(e) : 0;
+  }
+  if (ych <= 0) return 0;
+  for (int32_t c = ych < nby ? ych : nby; c >= 1; --c)
+    if (nby % c == 0) return c == nby ? 0 : c;
+  return 0;
 }
 
 static dim3 mhd_block() {
@@ -390,7 +425,9 @@ void mhd_div_launch_on(LocalDomain &d, const Rect3 &region, const MhdCoeffs &cf,
   MhdParams p{};
   mhd_fill_params(d, region, cf, p);
   dim3 block = mhd_block();
-  hipLaunchKernelGGL(mhd_div_kernel, mhd_grid(ext, block), block, 0, stream, p);
+  dim3 grid = mhd_grid(ext, block);
+  p.ychunk = mhd_ychunk((int32_t)grid.y);
+  hipLaunchKernelGGL(mhd_div_kernel, grid, block, 0, stream, p);
   STENCIL_HIP(hipGetLastError());
 }
 
@@ -407,12 +444,14 @@ void mhd_substep_launch_on(LocalDomain &d, const Rect3 &region, int step, double
   p.alpha_over_beta_prev = (step == 0) ? 0.0 : ALPHA[step] / BETA[step - 1];
   p.beta = BETA[step];
   dim3 block = mhd_block();
+  dim3 grid = mhd_grid(ext, block);
+  p.ychunk = mhd_ychunk((int32_t)grid.y);
   // scalar (writes lnrho/ss/aa) and momentum (writes uu) touch disjoint
   // outputs and only read shared inputs: run them CONCURRENTLY on the two
   // streams (the caller joins them)
-  hipLaunchKernelGGL(mhd_scalar_kernel, mhd_grid(ext, block), block, 0, sScalar, p);
+  hipLaunchKernelGGL(mhd_scalar_kernel, grid, block, 0, sScalar, p);
   STENCIL_HIP(hipGetLastError());
-  hipLaunchKernelGGL(mhd_momentum_kernel, mhd_grid(ext, block), block, 0, sMomentum, p);
+  hipLaunchKernelGGL(mhd_momentum_kernel, grid, block, 0, sMomentum, p);
   STENCIL_HIP(hipGetLastError());
 }
 
