@@ -203,6 +203,7 @@ __global__ void __launch_bounds__(256) jacobi_kernel_v4(JacobiParams p) {
 // (examples/jacobi_probe.cpp full-ldsrows). Requires blockDim (64,4,1)
 // and a vecAll launch (no scalar lanes; every lane's row exists because
 // the region is the interior rect of a radius>=1 domain).
+template <int ZC>
 __global__ void __launch_bounds__(256) jacobi_kernel_v4_lds(JacobiParams p) {
   __shared__ float4 tile[2][6][64];
   const int32_t tx = threadIdx.x;
@@ -220,8 +221,8 @@ __global__ void __launch_bounds__(256) jacobi_kernel_v4_lds(JacobiParams p) {
   const int32_t ly = min(ly0, (int32_t)p.extY);
   // +y halo row staged by ry==3: one past its own row, clamped in-bounds
   const int64_t pyHalo = (ly < p.extY) ? p.pitch : 0;
-  const int32_t lz0 = blockIdx.z * JAC_ZCHUNK;
-  const int32_t zEnd = min((int32_t)(lz0 + JAC_ZCHUNK), p.extZ);
+  const int32_t lz0 = blockIdx.z * ZC;
+  const int32_t zEnd = min((int32_t)(lz0 + ZC), p.extZ);
 
   const int64_t gy = p.loY + ly;
   const int64_t ay = gy - p.allocY;
@@ -403,10 +404,18 @@ void launch_jacobi_on(LocalDomain &d, int64_t qi, const Rect3 &region,
   }
   if (p.vecAll && useLds && ext.x >= 8) {
     // LDS-staged y-rows variant (fixed 64x4 block); see jacobi_kernel_v4_lds
+    static int zc = 0;
+    if (!zc) {
+      const char *e = getenv("STENCIL_JAC_ZC");
+      zc = (e && atoi(e) == 32) ? 32 : 16;
+    }
     dim3 block(64, 4, 1);
     dim3 grid((uint32_t)((p.extX / 4 + 63) / 64), (uint32_t)((ext.y + 3) / 4),
-              (uint32_t)((ext.z + 15) / 16));
-    hipLaunchKernelGGL(jacobi_kernel_v4_lds, grid, block, 0, stream, p);
+              (uint32_t)((ext.z + zc - 1) / zc));
+    if (zc == 32)
+      hipLaunchKernelGGL(jacobi_kernel_v4_lds<32>, grid, block, 0, stream, p);
+    else
+      hipLaunchKernelGGL(jacobi_kernel_v4_lds<16>, grid, block, 0, stream, p);
     STENCIL_HIP(hipGetLastError());
     return;
   }
