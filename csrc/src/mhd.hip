@@ -292,7 +292,7 @@ __global__ void __launch_bounds__(256) mhd_div_kernel(MhdParams p) {
 }
 
 // momentum: j_i = D_i(divA) - lap(A_i); graddiv u = grad(divA... grad(divu)
-__global__ void __launch_bounds__(256) mhd_momentum_kernel(MhdParams p) {
+__device__ __forceinline__ void mhd_momentum_body(const MhdParams &p) {
   int32_t bx = blockIdx.x, by = blockIdx.y, bz = blockIdx.z;
   if (p.swizzle) xcd_remap(bx, by, bz);
   if (p.ychunk) ychunk_remap(p.ychunk, bx, by, bz);
@@ -335,6 +335,13 @@ __global__ void __launch_bounds__(256) mhd_momentum_kernel(MhdParams p) {
     const double visc = p.nu * (st.lap(UUZ, ix, iy, iz) + st.dz(DIVU, iz) / 3.0);
     write_rk3(p, st, c.out[UUZ], UUZ, -ugradu - p.cs2 * press + rho_inv * jxB.z + visc);
   }
+}
+
+__global__ void __launch_bounds__(256) mhd_momentum_kernel(MhdParams p) { mhd_momentum_body(p); }
+// occupancy experiment (STENCIL_MHD_MOM5=1): force 5 waves/SIMD -- the
+// compiler must fit ~102 VGPR, possibly spilling; measured A/B decides
+__global__ void __launch_bounds__(256, 5) mhd_momentum_kernel_w5(MhdParams p) {
+  mhd_momentum_body(p);
 }
 
 } // namespace
@@ -456,7 +463,15 @@ void mhd_substep_launch_on(LocalDomain &d, const Rect3 &region, int step, double
   // streams (the caller joins them)
   hipLaunchKernelGGL(mhd_scalar_kernel, grid, block, 0, sScalar, p);
   STENCIL_HIP(hipGetLastError());
-  hipLaunchKernelGGL(mhd_momentum_kernel, grid, block, 0, sMomentum, p);
+  static int mom5 = -1;
+  if (mom5 < 0) {
+    const char *e = getenv("STENCIL_MHD_MOM5");
+    mom5 = (e && e[0] == '1') ? 1 : 0;
+  }
+  if (mom5)
+    hipLaunchKernelGGL(mhd_momentum_kernel_w5, grid, block, 0, sMomentum, p);
+  else
+    hipLaunchKernelGGL(mhd_momentum_kernel, grid, block, 0, sMomentum, p);
   STENCIL_HIP(hipGetLastError());
 }
 
