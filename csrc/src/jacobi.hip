@@ -10,7 +10,7 @@
 //     nontemporal stores (plain stores 9% slower),
 //   - jacobi_kernel_v4_lds (the vecAll/graph path): y-neighbor rows
 //     staged through 12 KB of ping-pong LDS, full 8 waves/SIMD — runs at
-//     the mapping's copy roofline (527 Gcell/s @750^3),
+//     the mapping's copy roofline (540 Gcell/s @750^3),
 //   - pure-vector full-rect launches (tail lanes cost 10%) with
 //     address-based 16 B alignment,
 //   - linearized scalar kernel for thin exterior slabs (any shape stays
