@@ -91,3 +91,22 @@ def test_wire_layout_alignment():
     # chunks must not overlap
     for (a, b) in zip(chunks, chunks[1:]):
         assert a[2] + a[3] <= b[2]
+
+
+def test_wire_layout_group_subset_deterministic():
+    """group exchanges: a qis subset must produce identical layouts on
+    both sides regardless of the subset's given order, and cover only the
+    subset's quantities (the RCCL wire matches by order, not tags)"""
+    from stencil_amd.parallel.planning import Message
+
+    msgs = [Message((1, 0, 0), 0, 1, (2, 3, 5)), Message((0, 0, 1), 0, 1, (7, 6, 1))]
+    es = [4, 8, 4, 8]
+    t1, c1 = wire_layout(msgs, es, qis=[3, 1])
+    t2, c2 = wire_layout(msgs, es, qis=[1, 3])
+    assert t1 == t2 and c1 == c2
+    qset = {qi for _, qi, _, _ in c1}
+    assert qset == {1, 3}
+    # full-set layout differs (more bytes) but stays consistent
+    tf, cf = wire_layout(msgs, es, qis=None)
+    assert tf > t1
+    assert {qi for _, qi, _, _ in cf} == {0, 1, 2, 3}
