@@ -254,3 +254,32 @@ def test_plan_files_and_timers(tmp_path):
     dd.swap()
     assert dd.time_exchange > 0.0
     assert dd.time_swap > 0.0
+
+
+def test_exchange_groups_validation_and_omission():
+    import stencil_amd as sa
+
+    dd = make_dd((10, 8, 8), 1, 1)
+    with pytest.raises(ValueError):
+        dd.set_exchange_groups([[0, 1], [1]])  # overlapping
+
+    # a quantity left out of every group never moves; grouped ones do
+    dd2 = make_dd((10, 8, 8), 1, 1)
+    ha = dd2.add_data(np.float32, "a")
+    hb = dd2.add_data(np.float32, "b")
+    dd2.set_exchange_groups([[ha.index]])
+    dd2.realize()
+    fill_interiors(dd2, ha)
+    fill_interiors(dd2, hb)
+    dd2.exchange(group=0)
+    from util import check_valid_regions
+
+    check_valid_regions(dd2, ha)
+    lo, hi = dd2.local_rect(0)
+    flo, fhi, full_b = (
+        tuple(c - 1 for c in lo),
+        tuple(c + 1 for c in hi),
+        None,
+    )
+    full_b = dd2.read_global(0, flo, fhi, hb)
+    assert (full_b[0, :, :] == 0).all()  # b's halo untouched (zero-init)
