@@ -89,3 +89,37 @@ def test_fuzz_exchange_groups(seed):
     dd.exchange(group=1)
     for h, scale in handles:
         check_valid_regions(dd, h, scale)
+
+
+@pytest.mark.parametrize("seed", range(20, 28))
+def test_fuzz_multi_step_swap(seed):
+    """randomized MULTI-STEP campaigns: exchange / swap / re-fill cycles
+    with mixed dtypes and asymmetric radii — catches swap-parity and
+    stale-pointer bugs that single-exchange fuzz cannot"""
+    rng = random.Random(seed)
+    n_dom = rng.choice([1, 2, 4])
+    radius = random_radius(rng)
+    max_r = max(
+        radius.dir(x, y, z) for x in (-1, 0, 1) for y in (-1, 0, 1) for z in (-1, 0, 1)
+    )
+    size = tuple(max(rng.randint(8, 20), max_r * n_dom * 2 + n_dom) for _ in range(3))
+
+    dd = sa.DistributedDomain(*size, backend="torch")
+    dd.set_radius(radius)
+    dd.set_gpus([0] * n_dom)
+    handles = []
+    for qi in range(rng.randint(1, 3)):
+        dtype = rng.choice([np.float32, np.float64])
+        handles.append((dd.add_data(dtype, f"q{qi}"), dtype))
+    dd.realize()
+
+    for step in range(rng.randint(2, 4)):
+        scale = 1.0 + step
+        for h, dtype in handles:
+            for li in range(dd.num_local()):
+                lo, hi = dd.local_rect(li)
+                dd.write_global(li, lo, ripple_block(lo, hi, dd.size, scale).astype(dtype), h)
+        dd.exchange()
+        for h, dtype in handles:
+            check_valid_regions(dd, h, scale)
+        dd.swap()
