@@ -89,19 +89,22 @@ class Jacobi3D:
             return
         self._eager_step(overlap)
 
-    def run(self, n: int):
-        """n steps with one host sync at the end: in graph mode the n
-        replays queue back-to-back, hiding the ~0.1 ms/step of
-        hipGraphLaunch + stream-sync wake latency (measured gpu23:
-        1.132 ms GPU-busy vs 1.263 ms walled per step at 750^3)"""
+    def run(self, n: int) -> float:
+        """n steps with one host sync at the end, returning elapsed
+        seconds: in graph mode the n replays queue back-to-back, hiding
+        the ~0.1 ms/step of hipGraphLaunch + stream-sync wake latency
+        (measured gpu23: 1.132 ms GPU-busy vs 1.263 ms walled per step
+        at 750^3)"""
+        t0 = time.perf_counter()
         if self._graph is not None and self.m == 1:
             from .. import _C
 
             _C.jacobi_graph_launch(self._graph, n)
             _C.jacobi_graph_sync(self._graph)
-            return
-        for _ in range(n):
-            self.step()
+        else:
+            for _ in range(n):
+                self.step()
+        return time.perf_counter() - t0
 
     def _eager_step(self, overlap: bool = True):
         dd = self.dd
@@ -162,9 +165,3 @@ class Jacobi3D:
         dd.backend.sync_compute()
         dd.swap()
         self._phase = (self._phase + 1) % self.m
-
-    def run(self, iters: int) -> float:
-        t0 = time.perf_counter()
-        for _ in range(iters):
-            self.step()
-        return time.perf_counter() - t0
