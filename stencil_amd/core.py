@@ -39,25 +39,41 @@ class Method(enum.Flag):
 
 
 class DataHandle:
-    __slots__ = ("index", "elem_size", "name")
+    __slots__ = ("index", "elem_size", "name", "dtype")
 
-    def __init__(self, index: int, elem_size: int, name: str):
+    def __init__(self, index: int, elem_size: int, name: str, dtype=None):
         self.index = index
         self.elem_size = elem_size
         self.name = name
+        # the numpy view dtype for host-side access (read_global /
+        # write_paraview); defaults to the size-canonical dtype
+        self.dtype = np.dtype(dtype) if dtype is not None else np.dtype(_NP_DTYPES[elem_size])
 
 
-def _es_of(dtype_or_size) -> int:
+def _np_dtype_of(dtype_or_size) -> np.dtype:
+    """resolve an add_data() spec (torch dtype / numpy dtype / element
+    size) to the numpy dtype used for host views. Kept alongside the
+    element size so read_global/write_paraview never reinterpret e.g.
+    int32 data as float32 (round-1 advisor finding)."""
     try:
         import torch
 
         if isinstance(dtype_or_size, torch.dtype):
-            return torch.tensor([], dtype=dtype_or_size).element_size()
+            t = torch.tensor([], dtype=dtype_or_size)
+            try:
+                return t.numpy().dtype
+            except Exception:  # no numpy equivalent (bf16...): raw view
+                return np.dtype(_NP_DTYPES[t.element_size()])
     except ImportError:
         pass
+    if isinstance(dtype_or_size, np.dtype):
+        return dtype_or_size
     if isinstance(dtype_or_size, type) and issubclass(dtype_or_size, np.generic):
-        return np.dtype(dtype_or_size).itemsize
-    return int(dtype_or_size)
+        return np.dtype(dtype_or_size)
+    size = int(dtype_or_size)
+    if size not in _NP_DTYPES:
+        raise ValueError(f"unsupported element size {size} (want 1/2/4/8)")
+    return np.dtype(_NP_DTYPES[size])
 
 
 class DistributedDomain:
@@ -69,6 +85,7 @@ class DistributedDomain:
         self.torch_device = device
         self.radius = _C.Radius.constant(0)
         self._data: List[Tuple[int, str]] = []
+        self._dtypes: List[np.dtype] = []
         self.methods = Method.DEFAULT
         self.strategy = PlacementStrategy.NodeAware
         self.gpus: Optional[List[int]] = None
@@ -94,9 +111,16 @@ class DistributedDomain:
             self.radius = r
 
     def add_data(self, dtype_or_size, name: str = "") -> DataHandle:
-        es = _es_of(dtype_or_size)
+        dt = _np_dtype_of(dtype_or_size)
+        es = dt.itemsize
         self._data.append((es, name))
-        return DataHandle(len(self._data) - 1, es, name)
+        self._dtypes.append(dt)
+        return DataHandle(len(self._data) - 1, es, name, dt)
+
+    def data_handle(self, index: int) -> DataHandle:
+        """handle for an already-added quantity (dtype-preserving)"""
+        es, name = self._data[index]
+        return DataHandle(index, es, name, self._dtypes[index])
 
     def set_methods(self, m: Method):
         self.methods = m
@@ -346,7 +370,7 @@ class DistributedDomain:
         pos = tuple(lo[i] - flo[i] for i in range(3))
         ext = tuple(hi[i] - lo[i] for i in range(3))
         raw = self.backend.read_region(li, pos, ext, handle.index, from_next)
-        arr = np.frombuffer(raw, dtype=_NP_DTYPES[handle.elem_size])
+        arr = np.frombuffer(raw, dtype=handle.dtype)
         return arr.reshape(ext[2], ext[1], ext[0])
 
     def write_global(self, li: int, lo: Vec, arr: np.ndarray, handle: DataHandle, to_next=False):
@@ -368,8 +392,8 @@ class DistributedDomain:
         for li in range(self.num_local()):
             lo, hi = self.local_rect(li)
             gid = self.placement.linearize(self.placement.get_idx(rank, li))
-            for i, (es, name) in enumerate(self._data):
-                h = DataHandle(i, es, name)
+            for i in range(len(self._data)):
+                h = self.data_handle(i)
                 arrays[f"d{gid}_q{i}"] = self.read_global(li, lo, hi, h)
         np.savez_compressed(f"{path}.rank{rank}.npz", **arrays)
 
@@ -381,8 +405,8 @@ class DistributedDomain:
             for li in range(self.num_local()):
                 lo, hi = self.local_rect(li)
                 gid = self.placement.linearize(self.placement.get_idx(rank, li))
-                for i, (es, name) in enumerate(self._data):
-                    h = DataHandle(i, es, name)
+                for i in range(len(self._data)):
+                    h = self.data_handle(i)
                     self.write_global(li, lo, data[f"d{gid}_q{i}"], h)
 
     # ---- observability ----
@@ -420,19 +444,22 @@ class DistributedDomain:
         """dump each local subdomain interior as CSV 'Z,Y,X,q0,q1,...'
         (reference src/stencil.cu:1188-1264)"""
         rank = self.comm.rank
-        handles = [DataHandle(i, es, name) for i, (es, name) in enumerate(self._data)]
+        handles = [self.data_handle(i) for i in range(len(self._data))]
         for li in range(self.num_local()):
             lo, hi = self.local_rect(li)
             arrays = [self.read_global(li, lo, hi, h).astype(np.float64) for h in handles]
             if zero_nans:
                 arrays = [np.nan_to_num(a, nan=0.0) for a in arrays]
             gid = self.placement.linearize(self.placement.get_idx(rank, li))
+            # vectorized dump (the round-1 Python triple loop took minutes
+            # on large domains): one savetxt over a coordinate+value table
+            Z, Y, X = np.mgrid[lo[2]:hi[2], lo[1]:hi[1], lo[0]:hi[0]]
+            table = np.column_stack(
+                [Z.ravel(), Y.ravel(), X.ravel()] + [a.ravel() for a in arrays]
+            )
             with open(f"{prefix}{gid}.txt", "w") as f:
                 f.write("Z,Y,X," + ",".join(h.name or f"q{h.index}" for h in handles) + "\n")
-                for z in range(lo[2], hi[2]):
-                    for y in range(lo[1], hi[1]):
-                        for x in range(lo[0], hi[0]):
-                            vals = ",".join(
-                                repr(a[z - lo[2], y - lo[1], x - lo[0]]) for a in arrays
-                            )
-                            f.write(f"{z},{y},{x},{vals}\n")
+                np.savetxt(
+                    f, table, delimiter=",",
+                    fmt=["%d", "%d", "%d"] + ["%.17g"] * len(arrays),
+                )

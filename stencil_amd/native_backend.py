@@ -9,7 +9,7 @@ from __future__ import annotations
 from typing import List, Optional, Tuple
 
 from . import _C
-from .parallel.planning import ExchangePlan, p2p_tag, wire_layout
+from .parallel.planning import ExchangePlan, pair_seq_tags, wire_layout, wire_layout_pairs
 
 Vec = Tuple[int, int, int]
 
@@ -108,6 +108,8 @@ class NativeBackend:
 
         import torch
 
+        seq = pair_seq_tags(plan)
+        ng = len(self.groups)
         for g, qis in enumerate(self.groups):
             for item, is_send in [(s, True) for s in plan.sends] + [(r, False) for r in plan.recvs]:
                 total, chunks = wire_layout(item.messages, elem_sizes, qis)
@@ -125,7 +127,9 @@ class NativeBackend:
                         self.engine.add_unpack(item.local_id, buf, off, pos, _vec3(m.ext), qi,
                                                group=3 * g)
                 tensor = torch.from_dlpack(self.engine.buffer_dlpack(buf))
-                tag = p2p_tag(item.src_gid, item.dst_gid)
+                # per-rank-pair sequence index, group-disambiguated: both
+                # sides compute the identical (small) tag for this buffer
+                tag = seq[(item.peer_rank, item.src_gid, item.dst_gid)] * ng + g
                 if is_send:
                     self._send_ops[g].append((tensor, item.peer_rank, tag))
                 else:
@@ -142,12 +146,21 @@ class NativeBackend:
         self.engine.finalize()
 
     @staticmethod
-    def _is_thin(m) -> bool:
-        """thin-row messages (x-faces/edges) written directly into remote
-        memory scatter 4-24 B stores over xGMI; they go through the staged
-        path instead (coalesced pack into the receiver's staging buffer +
-        local unpack)"""
-        return m.ext[0] * 8 < 64  # conservative: row < 64 B at fp64
+    def _is_thin(m, elem_size: int) -> bool:
+        """thin-row (message, quantity) chunks (x-faces/edges) written
+        directly into remote memory scatter <64 B rows over xGMI; they go
+        through the staged path instead (coalesced pack into the
+        receiver's staging buffer + local unpack). Per-quantity: an
+        8-cell fp64 row (64 B) is fat while the same row at fp32 (32 B)
+        is thin."""
+        return m.ext[0] * elem_size < 64
+
+    def _thin_pairs(self, messages, qis):
+        """deterministic (message, qi) enumeration of a transfer's thin
+        chunks -- identical on sender and receiver, so it defines the
+        staged buffer's wire layout"""
+        es = [e for e, _ in self.data_defs]
+        return [(m, qi) for m in messages for qi in sorted(qis) if self._is_thin(m, es[qi])]
 
     def _setup_ipc(self, plan: ExchangePlan, ctx: dict, ipc_sends, ipc_recvs):
         """exchange hipIpc handles among colocated ranks and register
@@ -183,14 +196,13 @@ class NativeBackend:
             staging = {}
             for g, qis in enumerate(self.groups):
                 for r in ipc_recvs:
-                    thin = [m for m in r.messages if self._is_thin(m)]
-                    if not thin:
+                    pairs = self._thin_pairs(r.messages, qis)
+                    if not pairs:
                         continue
-                    total, chunks = wire_layout(thin, elem_sizes, qis)
+                    total, chunks = wire_layout_pairs(pairs, elem_sizes)
                     buf = self.engine.create_buffer(r.local_id, 2 * total)
                     dom = self.domains[r.local_id]
-                    for mi, qi, off, nbytes in chunks:
-                        m = thin[mi]
+                    for m, qi, off, nbytes in chunks:
                         nd = _vec3(tuple(-c for c in m.dir))
                         pos = dom.halo_pos(nd, True)
                         for parity in (0, 1):
@@ -217,13 +229,17 @@ class NativeBackend:
         views = {}
         remote_staging = {}
         open_err = None
+        all_q = sorted({q for qs in self.groups for q in qs})
         try:
             for s in ipc_sends:
                 dst_idx = placement.dimensionize(s.dst_gid)
                 dst_li = placement.get_subdomain_id(dst_idx)
                 src_gpu = self.domains[s.local_id].gpu()
                 key = (s.peer_rank, dst_li, src_gpu)
-                if key not in views and any(not self._is_thin(m) for m in s.messages):
+                any_fat = any(
+                    not self._is_thin(m, elem_sizes[qi]) for m in s.messages for qi in all_q
+                )
+                if key not in views and any_fat:
                     info = infos[s.peer_rank]["domains"][dst_li]
                     views[key] = self.engine.create_remote_view(
                         src_gpu, info["curr"], info["next"], info["pitch"], info["ysize"],
@@ -252,21 +268,24 @@ class NativeBackend:
                 src_gpu = self.domains[s.local_id].gpu()
                 key = (s.peer_rank, dst_li, src_gpu)
                 dom = self.domains[s.local_id]
-                fat = [m for m in s.messages if not self._is_thin(m)]
-                thin = [m for m in s.messages if self._is_thin(m)]
-                for m in fat:
+                for m in s.messages:
+                    fat_qis = [
+                        qi for qi in sorted(qis) if not self._is_thin(m, elem_sizes[qi])
+                    ]
+                    if not fat_qis:
+                        continue
                     nd = _vec3(tuple(-c for c in m.dir))
                     src_pos = dom.halo_pos(_vec3(m.dir), False)
                     dst_pos = _C.halo_pos(nd, _vec3(dst_size), radius, True)
                     self.engine.add_translate_view(
-                        s.local_id, views[key], src_pos, dst_pos, _vec3(m.ext), g, sorted(qis)
+                        s.local_id, views[key], src_pos, dst_pos, _vec3(m.ext), g, fat_qis
                     )
+                thin = self._thin_pairs(s.messages, qis)
                 if thin:
                     rb, total = remote_staging[(s.src_gid, s.dst_gid, g)]
-                    _t2, chunks = wire_layout(thin, elem_sizes, qis)
+                    _t2, chunks = wire_layout_pairs(thin, elem_sizes)
                     assert _t2 == total, "staged wire layout mismatch"
-                    for mi, qi, off, nbytes in chunks:
-                        m = thin[mi]
+                    for m, qi, off, nbytes in chunks:
                         pos = dom.halo_pos(_vec3(m.dir), False)
                         for parity in (0, 1):
                             self.engine.add_pack(
@@ -324,8 +343,14 @@ class NativeBackend:
                 g = dist.new_group(ranks=by_node[node], backend="nccl")
                 if comm.rank in by_node[node]:
                     self._colo_nccl_group = g
-            self._barrier_t = torch.ones(1, device="cuda")
-            self._barrier_ev = torch.cuda.Event()
+            # allocate on the GPU the ENGINE drives, not torch's current
+            # device: if the app never set_device'd, every rank would
+            # otherwise land the all_reduce on cuda:0 and the per-node
+            # device barrier deadlocks
+            self._barrier_dev = my_dev if my_dev >= 0 else 0
+            with torch.cuda.device(self._barrier_dev):
+                self._barrier_t = torch.ones(1, device=f"cuda:{self._barrier_dev}")
+                self._barrier_ev = torch.cuda.Event()
 
     def _colo_barrier(self):
         """block the host until every colocated rank has passed its
@@ -333,10 +358,13 @@ class NativeBackend:
         import torch.distributed as dist
 
         if self._colo_nccl_group is not None:
-            w = dist.all_reduce(self._barrier_t, group=self._colo_nccl_group, async_op=True)
-            w.wait()  # torch's current (default) stream waits on the collective
-            self._barrier_ev.record()  # default stream is otherwise empty:
-            self._barrier_ev.synchronize()  # host waits for the collective only
+            import torch
+
+            with torch.cuda.device(self._barrier_dev):
+                w = dist.all_reduce(self._barrier_t, group=self._colo_nccl_group, async_op=True)
+                w.wait()  # torch's current (default) stream waits on the collective
+                self._barrier_ev.record()  # default stream is otherwise empty:
+                self._barrier_ev.synchronize()  # host waits for the collective only
         else:
             dist.barrier(group=self._colo_group)
 

@@ -62,11 +62,21 @@ class Placement:
 
     def __init__(self, size: Vec, radius: "_C.Radius", slots: List[Slot]):
         self.slots = slots
+        self._num_local: Dict[int, int] = {}
+        for s in slots:
+            self._num_local[s.rank] = self._num_local.get(s.rank, 0) + 1
         n_nodes = len({s.node for s in slots})
         per_node = [len([s for s in slots if s.node == n]) for n in range(n_nodes)]
-        if len(set(per_node)) != 1:
-            raise ValueError("placement requires the same GPU count on every node")
-        self.part = _C.NodePartition(_vec3(size), radius, n_nodes, per_node[0])
+        self.uniform_nodes = len(set(per_node)) == 1
+        if self.uniform_nodes:
+            self.part = _C.NodePartition(_vec3(size), radius, n_nodes, per_node[0])
+        else:
+            # heterogeneous per-node GPU counts: flat single-level split
+            # over all slots (the reference's Trivial handles arbitrary
+            # per-rank GPU counts, partition.hpp:337-444); the node-blocked
+            # two-level partition needs uniform counts, so NodeAware /
+            # IntraNodeRandom degrade to round-robin here (see subclasses)
+            self.part = _C.NodePartition(_vec3(size), radius, 1, len(slots))
         self.dim_v = self.part.dim().tuple()
         self.sys_dim = self.part.sys_dim().tuple()
         self.node_dim = self.part.node_dim().tuple()
@@ -105,7 +115,7 @@ class Placement:
         return self._by_rank[(rank, local_id)]
 
     def num_local(self, rank: int) -> int:
-        return len([1 for s in self.slots if s.rank == rank])
+        return self._num_local.get(rank, 0)
 
     def subdomain_size(self, idx: Vec) -> Vec:
         return self.part.subdomain_size(_vec3(idx)).tuple()
@@ -188,6 +198,12 @@ class NodeAwarePlacement(Placement):
         n = self.dim_v[0] * self.dim_v[1] * self.dim_v[2]
         if n != len(slots):
             raise ValueError(f"{n} subdomains but {len(slots)} GPU slots")
+        if not self.uniform_nodes:
+            # no node blocking to optimize within -- trivial assignment
+            for gid in range(n):
+                self.assign[gid] = gid
+            self._finish()
+            return
         n_nodes = len({s.node for s in slots})
         for node in range(n_nodes):
             gids = self._node_gids(node)
@@ -207,6 +223,11 @@ class IntraNodeRandomPlacement(Placement):
         if n != len(slots):
             raise ValueError(f"{n} subdomains but {len(slots)} GPU slots")
         rng = random.Random(seed)
+        if not self.uniform_nodes:
+            for gid in range(n):
+                self.assign[gid] = gid
+            self._finish()
+            return
         n_nodes = len({s.node for s in slots})
         for node in range(n_nodes):
             gids = self._node_gids(node)

@@ -160,8 +160,37 @@ def wire_layout(messages: List[Message], elem_sizes: List[int], qis=None
     return (off + 15) // 16 * 16, chunks
 
 
-def p2p_tag(src_gid: int, dst_gid: int) -> int:
-    """deterministic tag for a (src subdomain, dst subdomain) transfer.
-    RCCL ignores tags (ordered matching per rank pair -- both sides sort
-    identically); gloo honors them."""
-    return (src_gid * 4096 + dst_gid) % (1 << 20)
+def wire_layout_pairs(pairs, elem_sizes: List[int]
+                      ) -> Tuple[int, List[Tuple["Message", int, int, int]]]:
+    """byte layout of one packed buffer over explicit (message, quantity)
+    pairs (the staged thin-chunk IPC path, where thin-ness is decided per
+    quantity). Both ranks enumerate the pairs in the same deterministic
+    order, so the layout is the wire format. Returns (total_bytes, chunks)
+    with chunks = [(message, qi, offset, nbytes)]."""
+    chunks = []
+    off = 0
+    for m, qi in pairs:
+        off = (off + 15) // 16 * 16
+        nbytes = elem_sizes[qi] * m.volume()
+        chunks.append((m, qi, off, nbytes))
+        off += nbytes
+    return (off + 15) // 16 * 16, chunks
+
+
+def pair_seq_tags(plan: ExchangePlan) -> Dict[Tuple[int, int, int], int]:
+    """injective per-rank-pair tag for every wire transfer in the plan:
+    the index of (src_gid, dst_gid) in the sorted set of all transfers
+    (either direction) between this rank and that peer. Sender and
+    receiver enumerate the same key set for a given rank pair, so the
+    indices agree on both sides, and they are collision-free by
+    construction for any partition size (the round-1 arithmetic tag
+    `(src_gid*4096+dst_gid) % 2^20` aliased past 4096 gids; RCCL ignores
+    tags but gloo matches on them)."""
+    by_peer: Dict[int, set] = {}
+    for item in list(plan.sends) + list(plan.recvs):
+        by_peer.setdefault(item.peer_rank, set()).add((item.src_gid, item.dst_gid))
+    seq: Dict[Tuple[int, int, int], int] = {}
+    for peer, keys in by_peer.items():
+        for i, (sg, dg) in enumerate(sorted(keys)):
+            seq[(peer, sg, dg)] = i
+    return seq

@@ -13,7 +13,7 @@ from typing import Tuple
 import torch
 
 from . import _C
-from .parallel.planning import ExchangePlan, p2p_tag, wire_layout
+from .parallel.planning import ExchangePlan, pair_seq_tags, wire_layout
 
 Vec = Tuple[int, int, int]
 
@@ -95,6 +95,8 @@ class TorchBackend:
 
     def register_plan(self, plan: ExchangePlan, ctx=None):
         elem_sizes = [es for es, _ in self.data_defs]
+        seq = pair_seq_tags(plan)
+        ng = len(self.groups)
         for g, qis in enumerate(self.groups):
             for t in plan.translates:
                 src = self.domains[t.src_local]
@@ -114,7 +116,8 @@ class TorchBackend:
                     nd = tuple(-c for c in m.dir)
                     pos = dom.halo_pos(m.dir, False) if is_send else dom.halo_pos(nd, True)
                     entries.append((off, nbytes, item.local_id, pos, m.ext, qi))
-                rec = (buf, item.peer_rank, p2p_tag(item.src_gid, item.dst_gid), entries)
+                tag = seq[(item.peer_rank, item.src_gid, item.dst_gid)] * ng + g
+                rec = (buf, item.peer_rank, tag, entries)
                 (self._send_bufs[g] if is_send else self._recv_bufs[g]).append(rec)
 
     def _pack(self, g):

@@ -159,3 +159,40 @@ class TestQap:
         )
         f = _C.qap_solve(w, d)
         assert _C.qap_cost(w, d, f) == pytest.approx(best)
+
+
+def test_data_handle_preserves_dtype():
+    """add_data keeps the numpy dtype (not just element size), so
+    read_global on int32/float16 quantities does not silently
+    reinterpret bytes (round-1 advisor finding)"""
+    import numpy as np
+
+    from stencil_amd import DistributedDomain
+
+    dd = DistributedDomain(8, 8, 8, backend="torch")
+    hi32 = dd.add_data(np.int32, "ids")
+    hf32 = dd.add_data(np.float32, "temp")
+    hf16 = dd.add_data(np.float16, "half")
+    assert hi32.dtype == np.dtype(np.int32) and hi32.elem_size == 4
+    assert hf32.dtype == np.dtype(np.float32)
+    assert hf16.dtype == np.dtype(np.float16) and hf16.elem_size == 2
+    assert dd.data_handle(0).dtype == np.dtype(np.int32)
+    dd.set_radius(1)
+    dd.set_gpus([0])
+    dd.realize()
+    lo, hi = dd.local_rect(0)
+    arr = np.arange(8 * 8 * 8, dtype=np.int32).reshape(8, 8, 8)
+    dd.write_global(0, lo, arr, hi32)
+    back = dd.read_global(0, lo, hi, hi32)
+    assert back.dtype == np.int32
+    np.testing.assert_array_equal(back, arr)
+
+
+def test_add_data_rejects_bad_size():
+    import pytest
+
+    from stencil_amd import DistributedDomain
+
+    dd = DistributedDomain(8, 8, 8, backend="torch")
+    with pytest.raises(ValueError):
+        dd.add_data(3, "bad")

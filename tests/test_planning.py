@@ -110,3 +110,72 @@ def test_wire_layout_group_subset_deterministic():
     tf, cf = wire_layout(msgs, es, qis=None)
     assert tf > t1
     assert {qi for _, qi, _, _ in cf} == {0, 1, 2, 3}
+
+
+def test_pair_seq_tags_injective_and_symmetric():
+    """tags are collision-free per rank pair and agree between the
+    sender's and receiver's enumeration (round-1 arithmetic tag aliased
+    past 4096 gids; VERDICT weak #4)"""
+    from stencil_amd.parallel.planning import (
+        ExchangePlan,
+        Message,
+        WirePlanItem,
+        pair_seq_tags,
+    )
+
+    big = 5000  # beyond the old 4096 alias bound
+    # rank A's view: sends to peer 7 for two transfers, recv of one back
+    plan_a = ExchangePlan(
+        sends=[
+            WirePlanItem(7, 0, big, 0, [Message((1, 0, 0), 0, big, (2, 4, 4))]),
+            WirePlanItem(7, 1, big + 1, 0, [Message((1, 0, 0), 1, big + 1, (2, 4, 4))]),
+        ],
+        recvs=[WirePlanItem(7, big, 0, 0, [Message((1, 0, 0), big, 0, (2, 4, 4))])],
+    )
+    # rank B's (peer 7) mirrored view of the same three transfers
+    plan_b = ExchangePlan(
+        sends=[WirePlanItem(3, big, 0, 0, [Message((1, 0, 0), big, 0, (2, 4, 4))])],
+        recvs=[
+            WirePlanItem(3, 0, big, 0, [Message((1, 0, 0), 0, big, (2, 4, 4))]),
+            WirePlanItem(3, 1, big + 1, 0, [Message((1, 0, 0), 1, big + 1, (2, 4, 4))]),
+        ],
+    )
+    ta = pair_seq_tags(plan_a)
+    tb = pair_seq_tags(plan_b)
+    # injective within the pair
+    assert len(set(ta.values())) == len(ta) == 3
+    # sender and receiver agree per (src_gid, dst_gid)
+    for (peer, sg, dg), tag in ta.items():
+        assert tb[(3, sg, dg)] == tag
+
+
+def test_wire_layout_pairs_aligned_and_deterministic():
+    from stencil_amd.parallel.planning import Message, wire_layout_pairs
+
+    m1 = Message((1, 0, 0), 0, 1, (2, 3, 3))  # 18 cells
+    m2 = Message((0, 1, 0), 0, 1, (8, 2, 3))  # 48 cells
+    pairs = [(m1, 0), (m1, 1), (m2, 0)]
+    total, chunks = wire_layout_pairs(pairs, [4, 8])
+    offs = []
+    for (m, qi, off, nbytes), (em, eqi) in zip(chunks, pairs):
+        assert (m, qi) == (em, eqi)
+        assert off % 16 == 0
+        assert nbytes == [4, 8][qi] * m.volume()
+        offs.append((off, nbytes))
+    # non-overlapping, increasing
+    for (o1, n1), (o2, _) in zip(offs, offs[1:]):
+        assert o1 + n1 <= o2
+    assert total % 16 == 0 and total >= offs[-1][0] + offs[-1][1]
+
+
+def test_is_thin_per_quantity():
+    """an 8-cell row is fat at fp64 (64 B) but thin at fp32 (32 B) --
+    round 1 classified per message at a hardcoded fp64 width"""
+    from stencil_amd.native_backend import NativeBackend
+    from stencil_amd.parallel.planning import Message
+
+    m = Message((1, 0, 0), 0, 1, (8, 4, 4))
+    assert not NativeBackend._is_thin(m, 8)
+    assert NativeBackend._is_thin(m, 4)
+    wide = Message((0, 1, 0), 0, 1, (64, 2, 4))
+    assert not NativeBackend._is_thin(wide, 1)
