@@ -137,6 +137,12 @@ public:
   // make dom's compute stream wait until all exchange streams are idle
   // (used for stream-ordered overlap; v1 exchange is host-synchronous)
 
+  // the per-device pack/unpack stream, exposed so the native RCCL wire
+  // can post its grouped send/recv between the pack and unpack batches of
+  // the same stream: pack -> wire -> unpack is then fully stream-ordered
+  // with a single host sync at the end of the exchange
+  uintptr_t pack_stream_handle(int dev) { return (uintptr_t)pack_stream_(dev); }
+
   //// buffer access (for DLPack export to torch.distributed)
   uintptr_t buffer_ptr(int64_t buf) const { return (uintptr_t)buffers_[buf].ptr; }
   int64_t buffer_bytes(int64_t buf) const { return buffers_[buf].bytes; }

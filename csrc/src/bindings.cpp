@@ -16,6 +16,7 @@
 #include "stencil_amd/partition.hpp"
 #include "stencil_amd/qap.hpp"
 #include "stencil_amd/topo.hpp"
+#include "stencil_amd/wire.hpp"
 
 namespace py = pybind11;
 using namespace stencil_amd;
@@ -236,6 +237,8 @@ PYBIND11_MODULE(_C, m) {
       .def("sync_all", &ExchangeEngine::sync_all)
       .def("sync_compute", &ExchangeEngine::sync_compute)
       .def("compute_stream_handle", &ExchangeEngine::compute_stream_handle)
+      .def("pack_stream_handle", &ExchangeEngine::pack_stream_handle)
+      .def("buffer_ptr", &ExchangeEngine::buffer_ptr)
       .def("buffer_bytes", &ExchangeEngine::buffer_bytes)
       .def("buffer_device", &ExchangeEngine::buffer_device)
       .def("num_domains", &ExchangeEngine::num_domains)
@@ -293,6 +296,24 @@ PYBIND11_MODULE(_C, m) {
       .def_readonly("rms", &FieldStats::rms);
   m.def("field_stats", &field_stats, py::arg("eng"), py::arg("dom"), py::arg("qi"),
         py::arg("region"), py::arg("next_buf") = false);
+
+  // native RCCL wire (csrc/src/wire.hip): torch-free cross-rank transport
+  py::class_<RcclWire>(m, "RcclWire")
+      .def(py::init([](int device, int rank, int world, py::bytes uid) {
+             return new RcclWire(device, rank, world, std::string(uid));
+           }),
+           py::arg("device"), py::arg("rank"), py::arg("world"), py::arg("uid"))
+      .def_static("unique_id", []() { return py::bytes(RcclWire::unique_id()); })
+      .def("add_send", &RcclWire::add_send, py::arg("group"), py::arg("ptr"), py::arg("bytes"),
+           py::arg("peer"), py::arg("tag"))
+      .def("add_recv", &RcclWire::add_recv, py::arg("group"), py::arg("ptr"), py::arg("bytes"),
+           py::arg("peer"), py::arg("tag"))
+      .def("finalize", &RcclWire::finalize)
+      .def("post", &RcclWire::post, py::arg("group"), py::arg("stream"))
+      .def("barrier", &RcclWire::barrier, py::arg("stream"))
+      .def("rank", &RcclWire::rank)
+      .def("world", &RcclWire::world)
+      .def("device", &RcclWire::device);
 
   // topology utilities (csrc/src/topo.hip)
   m.def("gpu_distance", &gpu_distance);

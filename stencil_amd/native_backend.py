@@ -46,14 +46,18 @@ class NativeBackend:
         if len({c for _, _, c in domain_specs}) > 1:
             self.engine.enable_peer_all()
         ng = len(self.groups)
-        self._send_ops = [[] for _ in range(ng)]  # per group: (tensor, peer, tag)
+        self._send_items = [[] for _ in range(ng)]  # per group: (buf_id, peer, tag)
+        self._recv_items = [[] for _ in range(ng)]
+        self._send_ops = [[] for _ in range(ng)]  # torch/cpu modes: (tensor, peer, tag)
         self._recv_ops = [[] for _ in range(ng)]
         self._has_wire = [False] * ng
-        self._wire_via_cpu = False
+        self._wire_mode = "none"  # none | nccl | torch | cpu
+        self._wire = None  # _C.RcclWire (nccl mode)
+        self._wire_dev = -1
         self._cpu_mirror = {}
         self._ipc_active = False
         self._colo_group = None
-        self._colo_nccl_group = None
+        self._colo_wire = None  # _C.RcclWire over colocated ranks (device barrier)
         self._ipc_error = None
         self._colo_parity = [0] * ng
         self._staging_recv = {}
@@ -67,7 +71,7 @@ class NativeBackend:
 
         self._ipc_active = False
         self._colo_group = None
-        self._colo_nccl_group = None
+        self._colo_wire = None
         self._ipc_error = None
         self._colo_parity = [0] * len(self.groups)
         self._staging_recv = {}
@@ -106,8 +110,6 @@ class NativeBackend:
                     t.src_local, t.dst_local, src_pos, dst_pos, _vec3(t.ext), g, sorted(qis)
                 )
 
-        import torch
-
         seq = pair_seq_tags(plan)
         ng = len(self.groups)
         for g, qis in enumerate(self.groups):
@@ -126,24 +128,89 @@ class NativeBackend:
                         pos = dom.halo_pos(nd, True)
                         self.engine.add_unpack(item.local_id, buf, off, pos, _vec3(m.ext), qi,
                                                group=3 * g)
-                tensor = torch.from_dlpack(self.engine.buffer_dlpack(buf))
                 # per-rank-pair sequence index, group-disambiguated: both
                 # sides compute the identical (small) tag for this buffer
                 tag = seq[(item.peer_rank, item.src_gid, item.dst_gid)] * ng + g
                 if is_send:
-                    self._send_ops[g].append((tensor, item.peer_rank, tag))
+                    self._send_items[g].append((buf, item.peer_rank, tag))
                 else:
-                    self._recv_ops[g].append((tensor, item.peer_rank, tag))
-            self._has_wire[g] = bool(self._send_ops[g] or self._recv_ops[g])
-        if any(self._has_wire):
-            self._wire_via_cpu = self._detect_cpu_wire()
-            if self._wire_via_cpu:
-                self._cpu_mirror = {
-                    id(t): torch.empty(t.shape, dtype=torch.uint8, device="cpu")
-                    for g in range(len(self.groups))
-                    for t, _, _ in self._send_ops[g] + self._recv_ops[g]
-                }
+                    self._recv_items[g].append((buf, item.peer_rank, tag))
+            self._has_wire[g] = bool(self._send_items[g] or self._recv_items[g])
         self.engine.finalize()
+        if ctx is not None:
+            self._setup_wire(ctx)
+        elif any(self._has_wire):
+            raise RuntimeError("cross-rank wire transfers need a comm context")
+
+    def _setup_wire(self, ctx):
+        """choose the cross-rank wire transport, identically on every rank
+        (the vote is allgathered; a mixed choice would deadlock):
+
+        - 'nccl' (production default on GPUs): native RcclWire
+          (csrc/src/wire.hip) -- grouped ncclSend/ncclRecv posted on the
+          engine's per-device pack stream, so pack -> wire -> unpack is
+          entirely stream-ordered with ONE host sync per exchange (the
+          round-1 path host-blocked between each phase)
+          and libtorch is out of the hot path.
+        - 'torch': torch.distributed batch_isend_irecv on the same device
+          buffers (the round-1 implementation; retained as the test
+          oracle, forced with STENCIL_AMD_WIRE=torch).
+        - 'cpu': pinned host mirrors over gloo (CPU-only boxes/tests;
+          forced with STENCIL_AMD_WIRE=cpu)."""
+        import os
+
+        comm = ctx["comm"]
+        if comm.world_size == 1:
+            self._wire_mode = "none"  # no cross-rank traffic possible
+            return
+        env = os.environ.get("STENCIL_AMD_WIRE", "")
+        want = "cpu" if (env == "cpu" or self._detect_cpu_wire()) else (
+            "torch" if env == "torch" else "nccl"
+        )
+        # the native wire drives ONE device; multi-device ranks fall back
+        devs = {
+            self.engine.buffer_device(b)
+            for g in range(len(self.groups))
+            for b, _, _ in self._send_items[g] + self._recv_items[g]
+        }
+        if want == "nccl" and len(devs) > 1:
+            want = "torch"
+        votes = comm.allgather_object((want, any(self._has_wire)))
+        if not any(w for _, w in votes):
+            self._wire_mode = "none"
+            return
+        modes = {m for m, _ in votes}
+        mode = "cpu" if "cpu" in modes else ("torch" if "torch" in modes else "nccl")
+        self._wire_mode = mode
+        if mode == "nccl":
+            self._wire_dev = next(iter(devs)) if devs else (
+                self.domains[0].gpu() if self.domains else 0
+            )
+            # the unique id travels over the gloo control plane, setup only
+            uid = _C.RcclWire.unique_id() if comm.rank == 0 else None
+            uid = comm.allgather_object(uid)[0]
+            self._wire = _C.RcclWire(self._wire_dev, comm.rank, comm.world_size, uid)
+            for g in range(len(self.groups)):
+                for b, peer, tag in self._send_items[g]:
+                    self._wire.add_send(g, self.engine.buffer_ptr(b),
+                                        self.engine.buffer_bytes(b), peer, tag)
+                for b, peer, tag in self._recv_items[g]:
+                    self._wire.add_recv(g, self.engine.buffer_ptr(b),
+                                        self.engine.buffer_bytes(b), peer, tag)
+            self._wire.finalize()
+            return
+        import torch
+
+        for g in range(len(self.groups)):
+            for items, ops in ((self._send_items[g], self._send_ops[g]),
+                               (self._recv_items[g], self._recv_ops[g])):
+                for b, peer, tag in items:
+                    t = torch.from_dlpack(self.engine.buffer_dlpack(b))
+                    ops.append((t, peer, tag))
+                    if mode == "cpu":
+                        self._cpu_mirror[id(t)] = torch.empty(
+                            t.shape, dtype=torch.uint8, device="cpu"
+                        )
 
     @staticmethod
     def _is_thin(m, elem_size: int) -> bool:
@@ -294,17 +361,18 @@ class NativeBackend:
                             )
 
     def _make_colo_groups(self, ctx):
-        """per-node subgroups for the post-translate IPC barrier.
+        """per-node barrier machinery for the post-translate IPC path.
         new_group is collective: every rank creates every node's group.
 
         Two barrier flavors: a gloo host barrier (always created; the
         fallback and the only option when ranks share a device — RCCL
-        refuses communicators with two ranks on one GPU), and a device
-        barrier (1-element RCCL all_reduce, waited via event so the
-        engine's in-flight compute streams are NOT synced) used when every
-        colocated rank drives a distinct GPU. A host gloo barrier costs
-        O(100us..ms) per exchange at world 8, which is material against a
-        ~1.3 ms jacobi step; the RCCL barrier tracks stream completion at
+        refuses communicators with two ranks on one GPU), and a native
+        device barrier: a 1-float all-reduce on a colocated-ranks
+        RcclWire communicator (csrc/src/wire.hip), posted on the engine's
+        pack stream so the staged unpacks can be stream-ordered behind it
+        with no extra host block. A host gloo barrier costs O(100us..ms)
+        per exchange at world 8, which is material against a ~1.3 ms
+        jacobi step; the RCCL barrier tracks stream completion at
         collective cost (~20-30us). STENCIL_AMD_BARRIER=gloo forces the
         host barrier; the decision is allgathered so it is identical on
         every rank (a mixed choice would deadlock)."""
@@ -312,7 +380,6 @@ class NativeBackend:
             return
         import os
 
-        import torch
         import torch.distributed as dist
 
         comm = ctx["comm"]
@@ -323,8 +390,8 @@ class NativeBackend:
         my_dev = self.domains[0].gpu() if self.domains else -1
         want_dev = (
             os.environ.get("STENCIL_AMD_BARRIER", "auto") != "gloo"
-            and torch.cuda.is_available()
-            and "nccl" in dist.get_backend_config()
+            and _C.device_count() > 0
+            and my_dev >= 0
         )
         infos = comm.allgather_object((nodes[comm.rank], my_dev, bool(want_dev)))
         use_dev = all(i[2] for i in infos)
@@ -337,35 +404,27 @@ class NativeBackend:
             g = dist.new_group(ranks=by_node[node], backend="gloo")
             if comm.rank in by_node[node]:
                 self._colo_group = g
-        self._colo_nccl_group = None
+        self._colo_wire = None
         if use_dev:
-            for node in sorted(by_node):
-                g = dist.new_group(ranks=by_node[node], backend="nccl")
-                if comm.rank in by_node[node]:
-                    self._colo_nccl_group = g
-            # allocate on the GPU the ENGINE drives, not torch's current
-            # device: if the app never set_device'd, every rank would
-            # otherwise land the all_reduce on cuda:0 and the per-node
-            # device barrier deadlocks
-            self._barrier_dev = my_dev if my_dev >= 0 else 0
-            with torch.cuda.device(self._barrier_dev):
-                self._barrier_t = torch.ones(1, device=f"cuda:{self._barrier_dev}")
-                self._barrier_ev = torch.cuda.Event()
+            # one sub-communicator per node; each node leader mints the
+            # unique id, exchanged over the gloo control plane (setup only)
+            my_ranks = by_node[nodes[comm.rank]]
+            uid = _C.RcclWire.unique_id() if comm.rank == my_ranks[0] else None
+            uids = comm.allgather_object(uid)
+            self._colo_wire = _C.RcclWire(
+                my_dev, my_ranks.index(comm.rank), len(my_ranks), uids[my_ranks[0]]
+            )
 
     def _colo_barrier(self):
-        """block the host until every colocated rank has passed its
-        translate sync (their xGMI writes into our buffers are complete)."""
-        import torch.distributed as dist
-
-        if self._colo_nccl_group is not None:
-            import torch
-
-            with torch.cuda.device(self._barrier_dev):
-                w = dist.all_reduce(self._barrier_t, group=self._colo_nccl_group, async_op=True)
-                w.wait()  # torch's current (default) stream waits on the collective
-                self._barrier_ev.record()  # default stream is otherwise empty:
-                self._barrier_ev.synchronize()  # host waits for the collective only
+        """ensure every colocated rank has passed its translate sync
+        (their xGMI writes into our buffers are complete). Native flavor:
+        the all-reduce is ENQUEUED on the pack stream — callers that only
+        enqueue more pack-stream work after it need no host block here."""
+        if self._colo_wire is not None:
+            self._colo_wire.barrier(self.engine.pack_stream_handle(self._colo_wire.device()))
         else:
+            import torch.distributed as dist
+
             dist.barrier(group=self._colo_group)
 
     @staticmethod
@@ -401,31 +460,41 @@ class NativeBackend:
         group's halos are in place."""
         g = group
         if self._has_wire[g]:
-            import torch.distributed as dist
-
-            self.engine.sync_packs()
-            if self._wire_via_cpu:
-                ops = []
-                for t, peer, tag in self._send_ops[g]:
-                    m = self._cpu_mirror[id(t)]
-                    m.copy_(t)
-                    ops.append(dist.P2POp(dist.isend, m, peer, tag=tag))
-                for t, peer, tag in self._recv_ops[g]:
-                    ops.append(dist.P2POp(dist.irecv, self._cpu_mirror[id(t)], peer, tag=tag))
-                for w in dist.batch_isend_irecv(ops):
-                    w.wait()
-                for t, _, _ in self._recv_ops[g]:
-                    t.copy_(self._cpu_mirror[id(t)])
+            if self._wire_mode == "nccl":
+                # fully stream-ordered: the packs already sit on the pack
+                # stream (exchange_begin), the grouped send/recv is posted
+                # behind them, and the unpacks are enqueued behind the
+                # recvs -- the only host block is sync_all below
+                self._wire.post(g, self.engine.pack_stream_handle(self._wire_dev))
+                self.engine.launch_unpacks(3 * g)
             else:
-                ops = [
-                    dist.P2POp(dist.isend, t, peer, tag=tag) for t, peer, tag in self._send_ops[g]
-                ]
-                ops += [
-                    dist.P2POp(dist.irecv, t, peer, tag=tag) for t, peer, tag in self._recv_ops[g]
-                ]
-                for w in dist.batch_isend_irecv(ops):
-                    w.wait()
-            self.engine.launch_unpacks(3 * g)
+                import torch.distributed as dist
+
+                self.engine.sync_packs()
+                if self._wire_mode == "cpu":
+                    ops = []
+                    for t, peer, tag in self._send_ops[g]:
+                        m = self._cpu_mirror[id(t)]
+                        m.copy_(t)
+                        ops.append(dist.P2POp(dist.isend, m, peer, tag=tag))
+                    for t, peer, tag in self._recv_ops[g]:
+                        ops.append(dist.P2POp(dist.irecv, self._cpu_mirror[id(t)], peer, tag=tag))
+                    for w in dist.batch_isend_irecv(ops):
+                        w.wait()
+                    for t, _, _ in self._recv_ops[g]:
+                        t.copy_(self._cpu_mirror[id(t)])
+                else:
+                    ops = [
+                        dist.P2POp(dist.isend, t, peer, tag=tag)
+                        for t, peer, tag in self._send_ops[g]
+                    ]
+                    ops += [
+                        dist.P2POp(dist.irecv, t, peer, tag=tag)
+                        for t, peer, tag in self._recv_ops[g]
+                    ]
+                    for w in dist.batch_isend_irecv(ops):
+                        w.wait()
+                self.engine.launch_unpacks(3 * g)
         self.engine.sync_all()
         if self._ipc_active:
             # all colocated ranks' direct writes and staged packs are
@@ -433,6 +502,9 @@ class NativeBackend:
             # above); then unpack this exchange's staging parity locally.
             # The barrier also keeps senders at most one exchange (of this
             # group) ahead, which makes two staging parities sufficient.
+            # With the native device barrier the staged unpacks are stream-
+            # ordered behind the barrier all-reduce on the pack stream, so
+            # the host blocks once (sync_packs), not twice.
             self._colo_barrier()
             self.engine.launch_unpacks(3 * g + 1 + self._colo_parity[g])
             self.engine.sync_packs()

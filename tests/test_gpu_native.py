@@ -252,3 +252,63 @@ def test_jacobi_step_graph_matches_eager(monkeypatch):
     np.testing.assert_array_equal(outs["graph"], outs["eager"])
     ref = _run_jacobi("torch", size, 4, 1)
     np.testing.assert_allclose(outs["graph"], ref[0][1], rtol=1e-6, atol=1e-6)
+
+
+def test_rccl_wire_self_roundtrip():
+    """native RcclWire sanity on one GPU: a world-1 communicator moving a
+    device buffer to itself through grouped ncclSend/ncclRecv posted on
+    the engine pack stream (the stream-ordering contract exchange_end
+    relies on), plus the device barrier. Multi-rank matching is covered
+    by the plan-parity CPU tests + the driver's multi-GPU run (RCCL
+    refuses two ranks on one device, so N>1 cannot run here)."""
+    dom = _C.LocalDomain(_C.Vec3(8, 8, 8), _C.Vec3(0, 0, 0), 0)
+    dom.set_radius(_C.Radius.constant(0))
+    dom.add_data(4, "q")
+    dom.realize()
+    eng = _C.ExchangeEngine([dom])
+    nbytes = 4096
+    src = eng.create_buffer(0, nbytes)
+    dst = eng.create_buffer(0, nbytes)
+    payload = bytes(range(256)) * (nbytes // 256)
+    eng.buffer_from_host(src, payload)
+    eng.buffer_from_host(dst, b"\0" * nbytes)
+
+    uid = _C.RcclWire.unique_id()
+    w = _C.RcclWire(0, 0, 1, uid)
+    w.add_send(0, eng.buffer_ptr(src), nbytes, 0, 0)
+    w.add_recv(0, eng.buffer_ptr(dst), nbytes, 0, 0)
+    w.finalize()
+    stream = eng.pack_stream_handle(0)
+    w.post(0, stream)
+    w.barrier(stream)
+    eng.sync_packs()
+    assert eng.buffer_to_host(dst) == payload
+
+
+def test_rccl_wire_two_transfers_tag_order():
+    """two self-transfers in one group must match by tag order on both
+    sides (the pair_seq_tags contract): payloads land in the buffers the
+    tags name, not swapped."""
+    dom = _C.LocalDomain(_C.Vec3(8, 8, 8), _C.Vec3(0, 0, 0), 0)
+    dom.set_radius(_C.Radius.constant(0))
+    dom.add_data(4, "q")
+    dom.realize()
+    eng = _C.ExchangeEngine([dom])
+    n = 1024
+    bufs = [eng.create_buffer(0, n) for _ in range(4)]  # s0 s1 d0 d1
+    pay = [bytes([i]) * n for i in (1, 2)]
+    eng.buffer_from_host(bufs[0], pay[0])
+    eng.buffer_from_host(bufs[1], pay[1])
+    uid = _C.RcclWire.unique_id()
+    w = _C.RcclWire(0, 0, 1, uid)
+    # register recvs before sends and with shuffled tag order: finalize()
+    # must sort both sides into the same pairing
+    w.add_recv(0, eng.buffer_ptr(bufs[3]), n, 0, 1)
+    w.add_recv(0, eng.buffer_ptr(bufs[2]), n, 0, 0)
+    w.add_send(0, eng.buffer_ptr(bufs[1]), n, 0, 1)
+    w.add_send(0, eng.buffer_ptr(bufs[0]), n, 0, 0)
+    w.finalize()
+    w.post(0, eng.pack_stream_handle(0))
+    eng.sync_packs()
+    assert eng.buffer_to_host(bufs[2]) == pay[0]
+    assert eng.buffer_to_host(bufs[3]) == pay[1]

@@ -66,6 +66,7 @@ def build(verbose=True):
         cmd = [HIPCC, "-shared", "-fPIC", "-o", str(out)] + [str(o) for o in objs] + [
             "-L/opt/rocm/lib",
             "-lroctx64",
+            "-lrccl",
         ]
         if verbose:
             print("[link]", " ".join(cmd), flush=True)
@@ -87,6 +88,7 @@ def build(verbose=True):
         cmd = [HIPCC, str(exo)] + [str(o) for o in core_objs] + [
             "-L/opt/rocm/lib",
             "-lroctx64",
+            "-lrccl",
             "-o",
             str(exe),
         ]
