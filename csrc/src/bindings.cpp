@@ -7,6 +7,7 @@
 #include <pybind11/stl.h>
 
 #include <cstring>
+#include <memory>
 
 #include "stencil_amd/core.hpp"
 #include "stencil_amd/domain.hpp"
@@ -14,6 +15,8 @@
 #include "stencil_amd/hip_check.hpp"
 #include "stencil_amd/ops.hpp"
 #include "stencil_amd/partition.hpp"
+#include "stencil_amd/placement.hpp"
+#include "stencil_amd/planning.hpp"
 #include "stencil_amd/qap.hpp"
 #include "stencil_amd/topo.hpp"
 #include "stencil_amd/wire.hpp"
@@ -296,6 +299,81 @@ PYBIND11_MODULE(_C, m) {
       .def_readonly("rms", &FieldStats::rms);
   m.def("field_stats", &field_stats, py::arg("eng"), py::arg("dom"), py::arg("qi"),
         py::arg("region"), py::arg("next_buf") = false);
+
+  // C++ orchestrator introspection: run the native placement + planner
+  // (placement.hpp/planning.hpp — the ones the C++ DistributedDomain
+  // uses) and return plain structures, so tests can pin C++/Python plan
+  // parity without a GPU.
+  m.def(
+      "cpp_plan",
+      [](const Vec3 &size, const Radius &radius, int rank,
+         const std::vector<std::tuple<int, int, int, int>> &slotTuples,
+         const std::string &strategy) {
+        std::vector<Slot> slots;
+        for (auto &t : slotTuples)
+          slots.push_back({std::get<0>(t), std::get<1>(t), std::get<2>(t), std::get<3>(t)});
+        auto haloExtent = [](const Vec3 &d, const Vec3 &sz, const Radius &r) {
+          return LocalDomain::halo_extent(d, sz, r);
+        };
+        std::unique_ptr<Placement> p;
+        if (strategy == "trivial")
+          p = std::make_unique<TrivialPlacement>(size, radius, slots);
+        else if (strategy == "node_aware")
+          p = std::make_unique<NodeAwarePlacement>(size, radius, slots, haloExtent);
+        else
+          throw std::runtime_error("cpp_plan: unknown strategy " + strategy);
+        const ExchangePlan plan = plan_exchange(*p, radius, rank, haloExtent);
+        const auto seq = pair_seq_tags(plan);
+
+        auto tup3 = [](const Vec3 &v) { return py::make_tuple(v.x, v.y, v.z); };
+        py::list assign;
+        const int64_t n = p->dim().flatten();
+        for (int64_t gid = 0; gid < n; ++gid) {
+          const Vec3 idx = p->dimensionize(gid);
+          assign.append(py::make_tuple(p->get_rank(idx), p->get_subdomain_id(idx),
+                                       p->get_cuda(idx)));
+        }
+        py::list translates;
+        for (const auto &t : plan.translates)
+          translates.append(py::make_tuple(t.srcLocal, t.dstLocal, tup3(t.dir), tup3(t.ext)));
+        auto emit = [&](const std::vector<WirePlanItem> &items) {
+          py::list out;
+          for (const auto &it : items) {
+            py::list msgs;
+            for (const auto &m : it.messages)
+              msgs.append(py::make_tuple(tup3(m.dir), m.srcGid, m.dstGid, tup3(m.ext)));
+            out.append(py::make_tuple(it.peerRank, it.srcGid, it.dstGid, it.localId, msgs,
+                                      seq.at({it.peerRank, it.srcGid, it.dstGid})));
+          }
+          return out;
+        };
+        py::dict out;
+        out["dim"] = tup3(p->dim());
+        out["assign"] = assign;
+        out["translates"] = translates;
+        out["sends"] = emit(plan.sends);
+        out["recvs"] = emit(plan.recvs);
+        return out;
+      },
+      py::arg("size"), py::arg("radius"), py::arg("rank"), py::arg("slots"),
+      py::arg("strategy") = "node_aware");
+
+  m.def("cpp_wire_layout", [](const std::vector<std::tuple<py::tuple, int64_t, int64_t, py::tuple>> &msgs,
+                              const std::vector<int64_t> &elemSizes,
+                              const std::vector<int64_t> &qis) {
+    std::vector<PlanMessage> ms;
+    for (auto &t : msgs) {
+      py::tuple d = std::get<0>(t), e = std::get<3>(t);
+      ms.push_back({Vec3(d[0].cast<int64_t>(), d[1].cast<int64_t>(), d[2].cast<int64_t>()),
+                    std::get<1>(t), std::get<2>(t),
+                    Vec3(e[0].cast<int64_t>(), e[1].cast<int64_t>(), e[2].cast<int64_t>())});
+    }
+    std::vector<WireChunk> chunks;
+    const int64_t total = wire_layout(ms, elemSizes, qis, chunks);
+    py::list out;
+    for (const auto &c : chunks) out.append(py::make_tuple(c.msgIndex, c.qi, c.offset, c.nbytes));
+    return py::make_tuple(total, out);
+  });
 
   // native RCCL wire (csrc/src/wire.hip): torch-free cross-rank transport
   py::class_<RcclWire>(m, "RcclWire")
