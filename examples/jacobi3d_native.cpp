@@ -1,73 +1,67 @@
-// Pure-C++ usage example: the framework without Python. One process, one
-// (or more) GPUs; periodic self-exchange; jacobi iterations with
-// interior/exterior overlap. Build: tools/build_native.py --with-examples
-// -> build/jacobi3d_native.
+// Pure-C++ usage example: the full library without Python — partition,
+// placement, planning and transport all come from the C++
+// DistributedDomain (distributed.hpp), the same planner the Python path
+// uses (reference app shape: bin/jacobi3d.cu).
+//
+//   build/jacobi3d_native [edge=256] [iters=10] [ngpus=1]
+//
+// Multi-GPU single process: ngpus > 1 (subdomains split across devices,
+// halos move by direct-write xGMI kernels). Multi-process: launch one
+// process per GPU with STENCIL_RANK/STENCIL_WORLD/STENCIL_BOOTSTRAP_DIR
+// set (see tools/run_native_mp.sh) — cross-rank halos go over RCCL.
 #include <chrono>
 #include <cstdio>
-#include <memory>
+#include <cstdlib>
 #include <vector>
 
-#include "stencil_amd/core.hpp"
-#include "stencil_amd/domain.hpp"
-#include "stencil_amd/engine.hpp"
-#include "stencil_amd/hip_check.hpp"
+#include "stencil_amd/distributed.hpp"
 #include "stencil_amd/ops.hpp"
-#include "stencil_amd/partition.hpp"
 
 using namespace stencil_amd;
 
 int main(int argc, char **argv) {
   const int64_t n = argc > 1 ? atoll(argv[1]) : 256;
   const int iters = argc > 2 ? atoi(argv[2]) : 10;
-  const Radius radius = Radius::constant(1);
+  const int ngpus = argc > 3 ? atoi(argv[3]) : 1;
 
-  auto dom = std::make_shared<LocalDomain>(Vec3(n, n, n), Vec3(0, 0, 0), 0);
-  dom->set_radius(radius);
-  const int64_t q = dom->add_data(sizeof(float), "temp");
-  dom->realize();
+  DistributedDomain dd(n, n, n);
+  dd.set_radius(1);
+  const int64_t q = dd.add_data<float>("temp");
+  if (ngpus > 1) {
+    std::vector<int> gpus;
+    for (int i = 0; i < ngpus; ++i) gpus.push_back(i);
+    dd.set_gpus(gpus);
+  }
+  dd.realize();
 
-  ExchangeEngine eng({dom});
-  // periodic self-exchange: all 26 directions wrap onto the same domain
-  for (int dz = -1; dz <= 1; ++dz)
-    for (int dy = -1; dy <= 1; ++dy)
-      for (int dx = -1; dx <= 1; ++dx) {
-        const Vec3 d(dx, dy, dz);
-        if (d == Vec3(0, 0, 0) || radius.dir(-d) == 0) continue;
-        eng.add_translate(0, 0, dom->halo_pos(d, false), dom->halo_pos(-d, true),
-                          LocalDomain::halo_extent(-d, dom->size(), radius));
-      }
-  eng.finalize();
-
-  const Rect3 compute = dom->compute_region();
-  const Rect3 interior(compute.lo + Vec3(1, 1, 1), compute.hi - Vec3(1, 1, 1));
-  fill_f32(eng, 0, q, compute, 0.5f, false);
-  fill_f32(eng, 0, q, compute, 0.5f, true);
-  eng.sync_compute();
+  const Rect3 compute = dd.compute_region();
+  const auto interiors = dd.get_interior();
+  const auto exteriors = dd.get_exterior();
+  for (int li = 0; li < dd.num_local(); ++li) {
+    fill_f32(dd.engine(), li, q, dd.local_rect(li), 0.5f, false);
+    fill_f32(dd.engine(), li, q, dd.local_rect(li), 0.5f, true);
+  }
+  dd.engine().sync_compute();
 
   const auto t0 = std::chrono::steady_clock::now();
   for (int it = 0; it < iters; ++it) {
-    jacobi_step(eng, 0, q, interior, compute); // overlaps with...
-    eng.launch_translates();                   // ...the halo exchange
-    eng.sync_translates();
-    // exterior shells on the second compute stream: they depend only on
-    // the (synced) exchange, so they overlap the interior kernel
-    // (slide faces in; reference src/stencil.cu:927-977)
-    Rect3 c = compute;
-    for (int axis = 0; axis < 3; ++axis) {
-      Rect3 s = c;
-      s.lo[axis] = interior.hi[axis];
-      jacobi_step(eng, 0, q, s, compute, /*streamId=*/1);
-      c.hi[axis] = interior.hi[axis];
-      Rect3 t = c;
-      t.hi[axis] = interior.lo[axis];
-      jacobi_step(eng, 0, q, t, compute, /*streamId=*/1);
-      c.lo[axis] = interior.lo[axis];
-    }
-    eng.sync_compute();
-    dom->swap();
+    // interior compute overlaps the halo exchange
+    for (int li = 0; li < dd.num_local(); ++li)
+      jacobi_step(dd.engine(), li, q, interiors[li], compute);
+    dd.exchange();
+    for (int li = 0; li < dd.num_local(); ++li)
+      for (const Rect3 &box : exteriors[li])
+        jacobi_step(dd.engine(), li, q, box, compute, /*streamId=*/1);
+    dd.engine().sync_compute();
+    dd.swap();
   }
   const std::chrono::duration<double> dt = std::chrono::steady_clock::now() - t0;
-  printf("jacobi3d_native: %lld^3, %d iters, %.3f ms/iter, %.1f Gcell/s\n", (long long)n, iters,
-         dt.count() / iters * 1e3, (double)n * n * n * iters / dt.count() / 1e9);
+
+  if (dd.rank() == 0)
+    printf("jacobi3d_native: %lld^3 world=%d gpus=%d, %d iters, %.3f ms/iter, %.1f Gcell/s, "
+           "%lld translate B + %lld wire B per exchange\n",
+           (long long)n, dd.world(), ngpus, iters, dt.count() / iters * 1e3,
+           (double)n * n * n * iters / dt.count() / 1e9, (long long)dd.bytes_translate(),
+           (long long)dd.bytes_wire());
   return 0;
 }
