@@ -10,6 +10,7 @@
 #include <memory>
 
 #include "stencil_amd/core.hpp"
+#include "stencil_amd/distributed.hpp"
 #include "stencil_amd/domain.hpp"
 #include "stencil_amd/engine.hpp"
 #include "stencil_amd/hip_check.hpp"
@@ -374,6 +375,16 @@ PYBIND11_MODULE(_C, m) {
     for (const auto &c : chunks) out.append(py::make_tuple(c.msgIndex, c.qi, c.offset, c.nbytes));
     return py::make_tuple(total, out);
   });
+
+  // FileBootstrap (distributed.hpp): the C++ multi-process control plane,
+  // bound for CPU-side tests of the allgather protocol
+  py::class_<FileBootstrap>(m, "FileBootstrap")
+      .def(py::init<std::string, int, int>(), py::arg("dir"), py::arg("rank"), py::arg("world"))
+      .def("allgather", [](FileBootstrap &b, const std::string &phase, py::bytes payload) {
+        std::vector<py::bytes> out;
+        for (auto &s : b.allgather(phase, std::string(payload))) out.emplace_back(s);
+        return out;
+      });
 
   // native RCCL wire (csrc/src/wire.hip): torch-free cross-rank transport
   py::class_<RcclWire>(m, "RcclWire")

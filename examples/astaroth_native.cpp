@@ -31,7 +31,10 @@ int main(int argc, char **argv) {
   dd.set_exchange_groups({{0, 1, 2, 3, 4, 5, 6, 7}, {8, 9}});
   if (ngpus > 1) {
     std::vector<int> gpus;
-    for (int i = 0; i < ngpus; ++i) gpus.push_back(i);
+    // STENCIL_FAKE_GPUS=1: put every subdomain on device 0 (the
+    // same-GPU fake-multi-GPU test trick, reference test_exchange.cu:52)
+    const char *fake = getenv("STENCIL_FAKE_GPUS");
+    for (int i = 0; i < ngpus; ++i) gpus.push_back(fake && fake[0] == '1' ? 0 : i);
     dd.set_gpus(gpus);
   }
   dd.realize();

@@ -29,7 +29,10 @@ int main(int argc, char **argv) {
   const int64_t q = dd.add_data<float>("temp");
   if (ngpus > 1) {
     std::vector<int> gpus;
-    for (int i = 0; i < ngpus; ++i) gpus.push_back(i);
+    // STENCIL_FAKE_GPUS=1: put every subdomain on device 0 (the
+    // same-GPU fake-multi-GPU test trick, reference test_exchange.cu:52)
+    const char *fake = getenv("STENCIL_FAKE_GPUS");
+    for (int i = 0; i < ngpus; ++i) gpus.push_back(fake && fake[0] == '1' ? 0 : i);
     dd.set_gpus(gpus);
   }
   dd.realize();
