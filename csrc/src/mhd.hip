@@ -344,6 +344,105 @@ __global__ void __launch_bounds__(256, 5) mhd_momentum_kernel_w5(MhdParams p) {
   mhd_momentum_body(p);
 }
 
+// z-marching momentum (STENCIL_MHD_ZMARCH=1): each thread walks a whole
+// z-column keeping a 7-deep register pipeline of every field's center
+// column. The separable scheme (div u / div A exchanged as fields) left
+// momentum with NO cross derivatives, so its z-direction reads are
+// exactly the center-column values f(q,0,0,+-1..3) -- the 60 loads/cell
+// of z-displaced planes that thrash the 4 MB per-XCD L2 (the measured
+// 3.4x per-field fetch ratio, profiles/astaroth_256_fetch_size.csv).
+// The pipeline turns ALL of them into register shifts; what remains are
+// in-plane x/y-offset loads (served by L1/L2 plane locality) plus ONE
+// new z+3 load per field per step. Cost: ~70 fp64 pipeline registers ->
+// 1-2 waves/SIMD; an HBM-bound kernel can afford low occupancy if the
+// per-wave load stream stays deep.
+struct ZPipe {
+  double v[7]; // v[k] = f(q, 0, 0, k-3)
+  __device__ __forceinline__ void shift(double next) {
+#pragma unroll
+    for (int i = 0; i < 6; ++i) v[i] = v[i + 1];
+    v[6] = next;
+  }
+  __device__ __forceinline__ double dz(double ids) const {
+    return (D1[0] * (v[4] - v[2]) + D1[1] * (v[5] - v[1]) + D1[2] * (v[6] - v[0])) * ids;
+  }
+  __device__ __forceinline__ double dzz(double ids2) const {
+    return (D2[0] * v[3] + D2[1] * (v[4] + v[2]) + D2[2] * (v[5] + v[1]) + D2[3] * (v[6] + v[0])) *
+           ids2;
+  }
+  __device__ __forceinline__ double c() const { return v[3]; }
+};
+
+__global__ void __launch_bounds__(128) mhd_momentum_zmarch_kernel(MhdParams p) {
+  const int32_t lx = blockIdx.x * blockDim.x + threadIdx.x;
+  const int32_t ly = blockIdx.y * blockDim.y + threadIdx.y;
+  if (lx >= p.extX || ly >= p.extY) return;
+  const MhdCommon c0 = mhd_setup(p, lx, ly, 0);
+  const double ix = 1.0 / p.dsx, iy = 1.0 / p.dsy, iz = 1.0 / p.dsz;
+  const int64_t plane = p.plane;
+
+  ZPipe zp[10];
+#pragma unroll
+  for (int q = 0; q < 10; ++q)
+#pragma unroll
+    for (int k = 0; k < 7; ++k)
+      zp[q].v[k] = *(const double *)(c0.base[q] + (int64_t)(k - 3) * plane);
+
+  Stencil st; // in-plane derivative helper; base advanced per z step
+  st.pitch = p.pitch;
+  st.plane = p.plane;
+
+  for (int32_t lz = 0;;) {
+#pragma unroll
+    for (int q = 0; q < 10; ++q) st.base[q] = c0.base[q] + (int64_t)lz * plane;
+    const Vec3d uu = {zp[UUX].c(), zp[UUY].c(), zp[UUZ].c()};
+    const double rho_inv = exp(-zp[LNRHO].c());
+    const Vec3d j = {st.dx(DIVA, ix) - (st.dxx(AAX, ix * ix) + st.dyy(AAX, iy * iy) + zp[AAX].dzz(iz * iz)),
+                     st.dy(DIVA, iy) - (st.dxx(AAY, ix * ix) + st.dyy(AAY, iy * iy) + zp[AAY].dzz(iz * iz)),
+                     zp[DIVA].dz(iz) - (st.dxx(AAZ, ix * ix) + st.dyy(AAZ, iy * iy) + zp[AAZ].dzz(iz * iz))};
+    const Vec3d B = {st.dy(AAZ, iy) - zp[AAY].dz(iz), zp[AAX].dz(iz) - st.dx(AAZ, ix),
+                     st.dx(AAY, ix) - st.dy(AAX, iy)};
+    const Vec3d jxB = cross(j, B);
+    {
+      const double ugradu = uu.x * st.dx(UUX, ix) + uu.y * st.dy(UUX, iy) + uu.z * zp[UUX].dz(iz);
+      const double press = st.dx(LNRHO, ix) + p.cp_inv * st.dx(SS, ix);
+      const double visc =
+          p.nu * (st.dxx(UUX, ix * ix) + st.dyy(UUX, iy * iy) + zp[UUX].dzz(iz * iz) +
+                  st.dx(DIVU, ix) / 3.0);
+      char *out = c0.out[UUX] + (int64_t)lz * plane;
+      const double cur = zp[UUX].c(), prev = *(const double *)out;
+      *(double *)out = cur + p.beta * (p.alpha_over_beta_prev * (cur - prev) +
+                                       p.dt * (-ugradu - p.cs2 * press + rho_inv * jxB.x + visc));
+    }
+    {
+      const double ugradu = uu.x * st.dx(UUY, ix) + uu.y * st.dy(UUY, iy) + uu.z * zp[UUY].dz(iz);
+      const double press = st.dy(LNRHO, iy) + p.cp_inv * st.dy(SS, iy);
+      const double visc =
+          p.nu * (st.dxx(UUY, ix * ix) + st.dyy(UUY, iy * iy) + zp[UUY].dzz(iz * iz) +
+                  st.dy(DIVU, iy) / 3.0);
+      char *out = c0.out[UUY] + (int64_t)lz * plane;
+      const double cur = zp[UUY].c(), prev = *(const double *)out;
+      *(double *)out = cur + p.beta * (p.alpha_over_beta_prev * (cur - prev) +
+                                       p.dt * (-ugradu - p.cs2 * press + rho_inv * jxB.y + visc));
+    }
+    {
+      const double ugradu = uu.x * st.dx(UUZ, ix) + uu.y * st.dy(UUZ, iy) + uu.z * zp[UUZ].dz(iz);
+      const double press = zp[LNRHO].dz(iz) + p.cp_inv * zp[SS].dz(iz);
+      const double visc =
+          p.nu * (st.dxx(UUZ, ix * ix) + st.dyy(UUZ, iy * iy) + zp[UUZ].dzz(iz * iz) +
+                  zp[DIVU].dz(iz) / 3.0);
+      char *out = c0.out[UUZ] + (int64_t)lz * plane;
+      const double cur = zp[UUZ].c(), prev = *(const double *)out;
+      *(double *)out = cur + p.beta * (p.alpha_over_beta_prev * (cur - prev) +
+                                       p.dt * (-ugradu - p.cs2 * press + rho_inv * jxB.z + visc));
+    }
+    if (++lz >= p.extZ) break;
+#pragma unroll
+    for (int q = 0; q < 10; ++q)
+      zp[q].shift(*(const double *)(c0.base[q] + (int64_t)(lz + 3) * plane));
+  }
+}
+
 } // namespace
 
 static void mhd_fill_params(LocalDomain &d, const Rect3 &region, const MhdCoeffs &cf,
@@ -463,12 +562,18 @@ void mhd_substep_launch_on(LocalDomain &d, const Rect3 &region, int step, double
   // streams (the caller joins them)
   hipLaunchKernelGGL(mhd_scalar_kernel, grid, block, 0, sScalar, p);
   STENCIL_HIP(hipGetLastError());
-  static int mom5 = -1;
+  static int mom5 = -1, zmarch = -1;
   if (mom5 < 0) {
     const char *e = getenv("STENCIL_MHD_MOM5");
     mom5 = (e && e[0] == '1') ? 1 : 0;
+    e = getenv("STENCIL_MHD_ZMARCH");
+    zmarch = (e && e[0] == '1') ? 1 : 0;
   }
-  if (mom5)
+  if (zmarch) {
+    const dim3 zblock(64, 2, 1);
+    const dim3 zgrid((uint32_t)((ext.x + 63) / 64), (uint32_t)((ext.y + 1) / 2), 1);
+    hipLaunchKernelGGL(mhd_momentum_zmarch_kernel, zgrid, zblock, 0, sMomentum, p);
+  } else if (mom5)
     hipLaunchKernelGGL(mhd_momentum_kernel_w5, grid, block, 0, sMomentum, p);
   else
     hipLaunchKernelGGL(mhd_momentum_kernel, grid, block, 0, sMomentum, p);
