@@ -92,6 +92,9 @@ int64_t DistributedDomain::add_data(int64_t elemSize, const std::string &name) {
 }
 
 void DistributedDomain::set_exchange_groups(const std::vector<std::vector<int64_t>> &groups) {
+  // 3 engine launch slots per group x kGroups=12 (engine.hpp); RcclWire
+  // mirrors the same bound
+  if (groups.size() > 4) throw std::runtime_error("at most 4 exchange groups");
   groups_ = groups;
   for (auto &g : groups_) std::sort(g.begin(), g.end());
 }

@@ -135,10 +135,19 @@ class DistributedDomain:
         """partition quantities into independently exchangeable groups;
         exchange(group=i) then moves only groups[i]'s quantities (e.g. the
         MHD solver exchanges its 8 physics fields and its 2 div fields at
-        different points of a substep). Call before realize()."""
+        different points of a substep). Call before realize().
+
+        At most 4 groups: each group needs 3 engine launch slots (wire
+        packs + two staged-IPC parities) and the engine encodes 12
+        (engine.hpp kGroups) — raise kGroups there to lift this."""
         seen = [q for g in groups for q in g]
         if sorted(seen) != sorted(set(seen)):
             raise ValueError("exchange groups must be disjoint")
+        if len(groups) > 4:
+            raise ValueError(
+                "at most 4 exchange groups (engine launch-group encoding: "
+                "3 slots per group x kGroups=12; see csrc engine.hpp)"
+            )
         self.exchange_groups = [sorted(g) for g in groups]
 
     def set_output_prefix(self, p: str):
