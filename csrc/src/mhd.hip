@@ -227,7 +227,7 @@ __device__ __forceinline__ void write_rk3(const MhdParams &p, const Stencil &st,
 
 // kernel 1: continuity + entropy + induction (lnrho, ss, aa). First and
 // second derivatives only, no cross terms.
-__global__ void __launch_bounds__(256) mhd_scalar_kernel(MhdParams p) {
+__device__ __forceinline__ void mhd_scalar_body(const MhdParams &p) {
   int32_t bx = blockIdx.x, by = blockIdx.y, bz = blockIdx.z;
   if (p.swizzle) xcd_remap(bx, by, bz);
   if (p.ychunk) ychunk_remap(p.ychunk, bx, by, bz);
@@ -258,6 +258,12 @@ __global__ void __launch_bounds__(256) mhd_scalar_kernel(MhdParams p) {
   write_rk3(p, st, c.out[AAY], AAY, uxB.y + p.eta * st.lap(AAY, ix, iy, iz));
   write_rk3(p, st, c.out[AAZ], AAZ, uxB.z + p.eta * st.lap(AAZ, ix, iy, iz));
 }
+__global__ void __launch_bounds__(256) mhd_scalar_kernel(MhdParams p) { mhd_scalar_body(p); }
+// 512-thread variant (STENCIL_MHD_BLOCK with 512 threads, e.g. 64x4x2):
+// a taller block halves the per-cell refetch of +-3 y-rows and z-planes
+// through L1/LDS-free sharing -- the MHD kernels are L2-bandwidth-bound
+// (docs/DESIGN.md round-2 analysis), so block-footprint reuse is the lever
+__global__ void __launch_bounds__(512) mhd_scalar_kernel_b512(MhdParams p) { mhd_scalar_body(p); }
 
 // kernel 2a: Lorentz force j x B into a scratch array (j needs the cross
 // derivatives of A -- the most load-heavy part of the solver; isolating it
@@ -268,7 +274,7 @@ __global__ void __launch_bounds__(256) mhd_scalar_kernel(MhdParams p) {
 // composed cross-derivative sums (separable formulation; identical order
 // of accuracy, fp-rounding-level difference mirrored exactly in the NumPy
 // reference).
-__global__ void __launch_bounds__(256) mhd_div_kernel(MhdParams p) {
+__device__ __forceinline__ void mhd_div_body(const MhdParams &p) {
   int32_t bx = blockIdx.x, by = blockIdx.y, bz = blockIdx.z;
   if (p.swizzle) xcd_remap(bx, by, bz);
   if (p.ychunk) ychunk_remap(p.ychunk, bx, by, bz);
@@ -290,6 +296,8 @@ __global__ void __launch_bounds__(256) mhd_div_kernel(MhdParams p) {
   *(double *)(const_cast<char *>(c.base[DIVA])) =
       st.dx(AAX, ix) + st.dy(AAY, iy) + st.dz(AAZ, iz);
 }
+__global__ void __launch_bounds__(256) mhd_div_kernel(MhdParams p) { mhd_div_body(p); }
+__global__ void __launch_bounds__(512) mhd_div_kernel_b512(MhdParams p) { mhd_div_body(p); }
 
 // momentum: j_i = D_i(divA) - lap(A_i); graddiv u = grad(divA... grad(divu)
 __device__ __forceinline__ void mhd_momentum_body(const MhdParams &p) {
@@ -338,6 +346,9 @@ __device__ __forceinline__ void mhd_momentum_body(const MhdParams &p) {
 }
 
 __global__ void __launch_bounds__(256) mhd_momentum_kernel(MhdParams p) { mhd_momentum_body(p); }
+__global__ void __launch_bounds__(512) mhd_momentum_kernel_b512(MhdParams p) {
+  mhd_momentum_body(p);
+}
 // occupancy experiment (STENCIL_MHD_MOM5=1): force 5 waves/SIMD -- the
 // compiler must fit ~102 VGPR, possibly spilling; measured A/B decides
 __global__ void __launch_bounds__(256, 5) mhd_momentum_kernel_w5(MhdParams p) {
@@ -512,7 +523,8 @@ static dim3 mhd_block() {
     by = 2;
     bz = 2;
     if (const char *e = getenv("STENCIL_MHD_BLOCK"))
-      if (sscanf(e, "%dx%dx%d", &bx, &by, &bz) != 3 || bx * by * bz != 256) {
+      if (sscanf(e, "%dx%dx%d", &bx, &by, &bz) != 3 ||
+          (bx * by * bz != 256 && bx * by * bz != 512)) {
         bx = 64;
         by = 2;
         bz = 2;
@@ -538,7 +550,10 @@ void mhd_div_launch_on(LocalDomain &d, const Rect3 &region, const MhdCoeffs &cf,
   dim3 block = mhd_block();
   dim3 grid = mhd_grid(ext, block);
   p.ychunk = mhd_ychunk((int32_t)grid.y);
-  hipLaunchKernelGGL(mhd_div_kernel, grid, block, 0, stream, p);
+  if (block.x * block.y * block.z > 256)
+    hipLaunchKernelGGL(mhd_div_kernel_b512, grid, block, 0, stream, p);
+  else
+    hipLaunchKernelGGL(mhd_div_kernel, grid, block, 0, stream, p);
   STENCIL_HIP(hipGetLastError());
 }
 
@@ -560,7 +575,11 @@ void mhd_substep_launch_on(LocalDomain &d, const Rect3 &region, int step, double
   // scalar (writes lnrho/ss/aa) and momentum (writes uu) touch disjoint
   // outputs and only read shared inputs: run them CONCURRENTLY on the two
   // streams (the caller joins them)
-  hipLaunchKernelGGL(mhd_scalar_kernel, grid, block, 0, sScalar, p);
+  const bool b512 = block.x * block.y * block.z > 256;
+  if (b512)
+    hipLaunchKernelGGL(mhd_scalar_kernel_b512, grid, block, 0, sScalar, p);
+  else
+    hipLaunchKernelGGL(mhd_scalar_kernel, grid, block, 0, sScalar, p);
   STENCIL_HIP(hipGetLastError());
   static int mom5 = -1, zmarch = -1;
   if (mom5 < 0) {
@@ -573,7 +592,9 @@ void mhd_substep_launch_on(LocalDomain &d, const Rect3 &region, int step, double
     const dim3 zblock(64, 2, 1);
     const dim3 zgrid((uint32_t)((ext.x + 63) / 64), (uint32_t)((ext.y + 1) / 2), 1);
     hipLaunchKernelGGL(mhd_momentum_zmarch_kernel, zgrid, zblock, 0, sMomentum, p);
-  } else if (mom5)
+  } else if (b512)
+    hipLaunchKernelGGL(mhd_momentum_kernel_b512, grid, block, 0, sMomentum, p);
+  else if (mom5)
     hipLaunchKernelGGL(mhd_momentum_kernel_w5, grid, block, 0, sMomentum, p);
   else
     hipLaunchKernelGGL(mhd_momentum_kernel, grid, block, 0, sMomentum, p);
