@@ -14,6 +14,7 @@ sys.path.insert(0, os.path.join(os.path.dirname(__file__), ".."))
 import numpy as np
 
 import stencil_amd as sa
+from stencil_amd.parallel.machine import Machine
 from stencil_amd.parallel.planning import plan_exchange, wire_layout
 from stencil_amd.utils.statistics import Statistics
 
@@ -55,6 +56,17 @@ def main():
     ]
     total = sum(t.numel() for t, _ in sends)
 
+    # self/colocated/remote byte classification (reference bin/bench_mpi.cu
+    # via its Machine model)
+    machine = Machine.build(dd.comm, dd.gpus or [local_rank])
+    by_class = {"self": 0, "colocated": 0, "remote": 0}
+    for s_item in plan.sends:
+        by_class[machine.classify(rank, s_item.peer_rank)] += wire_layout(
+            s_item.messages, es
+        )[0]
+    for t_item in plan.translates:
+        by_class["self"] += t_item.ext[0] * t_item.ext[1] * t_item.ext[2] * es[0]
+
     stats = Statistics()
     for i in range(args.iters + 3):
         dist.barrier()
@@ -77,6 +89,15 @@ def main():
         print(
             f"transport,rccl_p2p,world={world},r={args.radius},size={args.size},"
             f"bytes={int(tot)},trimean_s={tm:.6f},GBs={tot / tm / 1e9:.2f}",
+            flush=True,
+        )
+    cls_t = torch.tensor([by_class["self"], by_class["colocated"], by_class["remote"]],
+                         dtype=torch.float64)
+    dist.all_reduce(cls_t, op=dist.ReduceOp.SUM)
+    if rank == 0:
+        print(
+            f"transport,classes,self_B={int(cls_t[0])},colocated_B={int(cls_t[1])},"
+            f"remote_B={int(cls_t[2])},nodes={machine.num_nodes()},gpus={len(machine.gpus)}",
             flush=True,
         )
     dist.destroy_process_group()

@@ -196,3 +196,19 @@ def test_add_data_rejects_bad_size():
     dd = DistributedDomain(8, 8, 8, backend="torch")
     with pytest.raises(ValueError):
         dd.add_data(3, "bad")
+
+
+def test_machine_model_single_rank():
+    """Machine.build: global GPU list with owning ranks (reference
+    src/machine.cpp:19-129); on a no-GPU box the stub entries still give
+    a consistent inventory and self/colocated/remote classification."""
+    from stencil_amd.parallel.comm import Comm
+    from stencil_amd.parallel.machine import Machine
+
+    comm = Comm()
+    m = Machine.build(comm, [0, 1])
+    assert m.num_nodes() == 1
+    assert len(m.gpus) == 2
+    assert m.gpus[0].ranks == [0] and m.gpus[0].cuda_of_rank[0] == 0
+    assert m.gpus_of_rank(0) == m.gpus
+    assert m.classify(0, 0) == "self"
