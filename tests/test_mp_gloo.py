@@ -83,3 +83,45 @@ def test_two_ranks_asymmetric():
 
 def test_four_ranks_r1():
     _run_world(4, 1, size=(16, 12, 10), port=29625)
+
+
+def _machine_worker(rank, world, port, q):
+    try:
+        os.environ["MASTER_ADDR"] = "127.0.0.1"
+        os.environ["MASTER_PORT"] = str(port)
+        import torch.distributed as dist
+
+        dist.init_process_group("gloo", rank=rank, world_size=world)
+        from stencil_amd.parallel.comm import Comm
+        from stencil_amd.parallel.machine import Machine
+
+        comm = Comm()
+        m = Machine.build(comm, [rank])  # each rank "owns" one device slot
+        assert m.num_nodes() == 1  # both ranks share the host
+        assert m.classify(0, 1) == "colocated"
+        assert m.classify(rank, rank) == "self"
+        # both ranks see the same global inventory
+        inv = [(g.host, g.pci, tuple(g.ranks)) for g in m.gpus]
+        invs = comm.allgather_object(inv)
+        assert invs[0] == invs[1]
+        dist.destroy_process_group()
+        q.put((rank, "ok"))
+    except Exception:  # pragma: no cover
+        import traceback
+
+        q.put((rank, traceback.format_exc()))
+
+
+def test_machine_world2():
+    """Machine inventory is identical and consistent across ranks"""
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    port = 29571
+    procs = [ctx.Process(target=_machine_worker, args=(r, 2, port, q)) for r in range(2)]
+    for p in procs:
+        p.start()
+    results = [q.get(timeout=120) for _ in range(2)]
+    for p in procs:
+        p.join(timeout=60)
+    for rank, status in results:
+        assert status == "ok", f"rank {rank}: {status}"
