@@ -86,7 +86,8 @@ def main():
     barrier_sync()
 
     t0 = time.perf_counter()
-    if getattr(app, "_graph", None) is not None and not args.no_overlap:
+    if (getattr(app, "_graph", None) is not None
+            or getattr(app, "_mr_graph", None) is not None) and not args.no_overlap:
         # graph mode: queue all K replays, one sync (the work is identical;
         # only the per-step host launch+wake latency is amortized)
         app.run(args.steps)

@@ -119,12 +119,21 @@ public:
 
   //// per-exchange execution (stream-ordered; host-sync via the sync_* calls)
   void launch_translates(int group = 0);
-  // launch a group's translate batches onto a caller-owned stream with
-  // plain kernel launches (used by whole-step hipGraph capture, where a
+  // launch a group's batches onto a caller-owned stream with plain
+  // kernel launches (used by whole-step hipGraph capture, where a
   // nested hipGraphLaunch would be illegal)
   void launch_translates_plain_on(uintptr_t stream, int group = 0);
+  void launch_packs_plain_on(uintptr_t stream, int group = 0);
+  void launch_unpacks_plain_on(uintptr_t stream, int group = 0);
   void launch_packs(int group = 0);
   void launch_unpacks(int group = 0);
+  // device-side parity flip of every remote view's pointer table (the
+  // in-graph analog of flip_views, paired with flip_views_host_only) --
+  // swaps devSlots <-> devSlotsAlt contents on `stream`
+  void enqueue_view_flips(uintptr_t stream);
+  void flip_views_host_only() {
+    for (auto &v : views_) v.parity ^= 1;
+  }
   void sync_translates();
   void sync_packs(); // also used after unpack
   void sync_all();
@@ -170,7 +179,8 @@ private:
     int parity = 0;
     std::vector<char *> base[2]; // [parity][qi] opened pointers
     std::vector<int64_t> pitch, ysize, elemSize, pads;
-    char **devSlots = nullptr; // local device array, refreshed on flip
+    char **devSlots = nullptr;    // local device array, refreshed on flip
+    char **devSlotsAlt = nullptr; // the other parity (for in-graph flips)
   };
   struct PackSpec {
     int dom;

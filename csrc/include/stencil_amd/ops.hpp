@@ -33,6 +33,19 @@ int64_t jacobi_graph_create(ExchangeEngine &eng, int dom, int64_t qi, const Rect
 void jacobi_graph_launch(int64_t handle, int64_t nSteps);
 void jacobi_graph_sync(int64_t handle);
 
+// Multi-rank whole-step graphs (see csrc/src/jacobi.hip): per parity,
+// A = [interior + translates + staged packs], B = [staged unpacks +
+// exterior + device swap]; the caller runs the cross-rank barrier in
+// between (RcclWire::barrier on jacobi_mr_graph_stream, or a host
+// barrier with stream syncs around it).
+int64_t jacobi_mr_graph_create(ExchangeEngine &eng, int dom, int64_t qi, const Rect3 &interior,
+                               const Rect3 &computeRegion, const std::vector<Rect3> &exteriors,
+                               int extendVec);
+uintptr_t jacobi_mr_graph_stream(int64_t handle);
+void jacobi_mr_graph_pre(int64_t handle);
+void jacobi_mr_graph_post(int64_t handle);
+void jacobi_mr_graph_sync(int64_t handle);
+
 // physical coefficients of the MHD solver (see csrc/src/mhd.hip)
 struct MhdCoeffs {
   double dsx = 1.0, dsy = 1.0, dsz = 1.0;

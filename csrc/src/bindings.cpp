@@ -274,6 +274,13 @@ PYBIND11_MODULE(_C, m) {
         py::arg("qi"), py::arg("region"), py::arg("compute_region"));
   m.def("jacobi_graph_launch", &jacobi_graph_launch, py::arg("handle"), py::arg("n_steps") = 1);
   m.def("jacobi_graph_sync", &jacobi_graph_sync);
+  m.def("jacobi_mr_graph_create", &jacobi_mr_graph_create, py::arg("eng"), py::arg("dom"),
+        py::arg("qi"), py::arg("interior"), py::arg("compute_region"), py::arg("exteriors"),
+        py::arg("extend_vec") = 2);
+  m.def("jacobi_mr_graph_stream", &jacobi_mr_graph_stream);
+  m.def("jacobi_mr_graph_pre", &jacobi_mr_graph_pre);
+  m.def("jacobi_mr_graph_post", &jacobi_mr_graph_post);
+  m.def("jacobi_mr_graph_sync", &jacobi_mr_graph_sync);
   m.def("mhd_graph_create", &mhd_graph_create, py::arg("eng"), py::arg("dom"), py::arg("region"),
         py::arg("dt"), py::arg("cf"));
   m.def("mhd_graph_iter", &mhd_graph_iter, py::arg("handle"), py::arg("n_iters") = 1);

@@ -7,6 +7,7 @@
 #include <cstring>
 #include <map>
 #include <mutex>
+#include <thread>
 #include <stdexcept>
 
 namespace stencil_amd {
@@ -185,6 +186,27 @@ PinnedBounce &pinned_bounce(int dev, int64_t need) {
   return pp;
 }
 
+// multi-threaded host memcpy: a single thread moves ~2.5 GB/s into fresh
+// (page-faulting) numpy pages -- measured as the bottleneck of the whole
+// pinned-bounce path (profiles/r2/r2_gpu7_io.log); 8 threads saturate the
+// host memory system instead
+void parallel_memcpy(void *dst, const void *src, int64_t bytes) {
+  constexpr int64_t kMinPerThread = 4ll << 20;
+  const int nt = (int)std::min<int64_t>(8, std::max<int64_t>(1, bytes / kMinPerThread));
+  if (nt <= 1) {
+    std::memcpy(dst, src, bytes);
+    return;
+  }
+  std::vector<std::thread> ts;
+  const int64_t per = (bytes + nt - 1) / nt;
+  for (int i = 0; i < nt; ++i) {
+    const int64_t off = i * per, n = std::min(per, bytes - off);
+    if (n <= 0) break;
+    ts.emplace_back([=]() { std::memcpy((char *)dst + off, (const char *)src + off, n); });
+  }
+  for (auto &t : ts) t.join();
+}
+
 // async strided copy of zs planes starting at zOff between the pitched
 // device allocation and a contiguous host buffer
 void slab_copy_async(const Pitched &p, int64_t es, const Vec3 &pos, const Vec3 &ext, int64_t zOff,
@@ -228,14 +250,14 @@ void LocalDomain::region_to_host(void *dst, const Vec3 &pos, const Vec3 &ext, in
     slab_copy_async(p, es, pos, ext, z0, zs, pp.buf[buf], true, pp.stream);
     STENCIL_HIP(hipEventRecord(pp.ev[buf], pp.stream));
     if (prev >= 0) { // overlap the DMA with draining the previous slab
-      std::memcpy(d, pp.buf[prev], prevBytes);
+      parallel_memcpy(d, pp.buf[prev], prevBytes);
       d += prevBytes;
     }
     STENCIL_HIP(hipEventSynchronize(pp.ev[buf]));
     prev = buf;
     prevBytes = zs * planeBytes;
   }
-  std::memcpy(d, pp.buf[prev], prevBytes);
+  parallel_memcpy(d, pp.buf[prev], prevBytes);
 }
 
 void LocalDomain::region_from_host(const void *src, const Vec3 &pos, const Vec3 &ext, int64_t qi,
@@ -254,7 +276,7 @@ void LocalDomain::region_from_host(const void *src, const Vec3 &pos, const Vec3 
     const int64_t bytes = zs * planeBytes;
     // the buffer's previous H2D (two slabs ago) must be complete
     STENCIL_HIP(hipEventSynchronize(pp.ev[buf]));
-    std::memcpy(pp.buf[buf], s, bytes);
+    parallel_memcpy(pp.buf[buf], s, bytes);
     s += bytes;
     slab_copy_async(p, es, pos, ext, z0, zs, pp.buf[buf], false, pp.stream);
     STENCIL_HIP(hipEventRecord(pp.ev[buf], pp.stream));

@@ -208,6 +208,7 @@ ExchangeEngine::~ExchangeEngine() {
       for (size_t qi = 0; qi < v.base[par].size(); ++qi)
         if (v.base[par][qi]) (void)hipIpcCloseMemHandle(v.base[par][qi] - v.pads[qi]);
     if (v.devSlots) (void)hipFree(v.devSlots);
+    if (v.devSlotsAlt) (void)hipFree(v.devSlotsAlt);
   }
 }
 
@@ -275,6 +276,9 @@ int64_t ExchangeEngine::create_remote_view(int openDev, const std::vector<std::s
   STENCIL_HIP(hipMalloc((void **)&v.devSlots, nq * sizeof(char *)));
   STENCIL_HIP(
       hipMemcpy(v.devSlots, v.base[0].data(), nq * sizeof(char *), hipMemcpyHostToDevice));
+  STENCIL_HIP(hipMalloc((void **)&v.devSlotsAlt, nq * sizeof(char *)));
+  STENCIL_HIP(
+      hipMemcpy(v.devSlotsAlt, v.base[1].data(), nq * sizeof(char *), hipMemcpyHostToDevice));
   views_.push_back(std::move(v));
   return (int64_t)views_.size() - 1;
 }
@@ -291,6 +295,27 @@ void ExchangeEngine::flip_views() {
     STENCIL_HIP(hipSetDevice(v.openDev));
     STENCIL_HIP(hipMemcpy(v.devSlots, v.base[v.parity].data(),
                           v.base[v.parity].size() * sizeof(char *), hipMemcpyHostToDevice));
+    STENCIL_HIP(hipMemcpy(v.devSlotsAlt, v.base[v.parity ^ 1].data(),
+                          v.base[v.parity ^ 1].size() * sizeof(char *), hipMemcpyHostToDevice));
+  }
+}
+
+namespace {
+__global__ void swap_slots_kernel(char **a, char **b, int n) {
+  const int i = threadIdx.x;
+  if (i < n) {
+    char *t = a[i];
+    a[i] = b[i];
+    b[i] = t;
+  }
+}
+} // namespace
+
+void ExchangeEngine::enqueue_view_flips(uintptr_t stream) {
+  for (auto &v : views_) {
+    hipLaunchKernelGGL(swap_slots_kernel, dim3(1), dim3(256), 0, (hipStream_t)stream, v.devSlots,
+                       v.devSlotsAlt, (int)v.base[0].size());
+    STENCIL_HIP(hipGetLastError());
   }
 }
 
@@ -490,6 +515,12 @@ void ExchangeEngine::launch_translates(int group) {
 }
 void ExchangeEngine::launch_translates_plain_on(uintptr_t stream, int group) {
   for (auto &b : translateBatches_[group]) b.launch_plain((hipStream_t)stream);
+}
+void ExchangeEngine::launch_packs_plain_on(uintptr_t stream, int group) {
+  for (auto &b : packBatches_[group]) b.launch_plain((hipStream_t)stream);
+}
+void ExchangeEngine::launch_unpacks_plain_on(uintptr_t stream, int group) {
+  for (auto &b : unpackBatches_[group]) b.launch_plain((hipStream_t)stream);
 }
 void ExchangeEngine::launch_packs(int group) {
   roctxRangePush("stencil::pack");

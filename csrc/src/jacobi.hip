@@ -524,6 +524,94 @@ void jacobi_graph_sync(int64_t handle) {
   STENCIL_HIP(hipStreamSynchronize(sg.stream));
 }
 
+// Multi-rank whole-step graphs (one rank, one domain, cross-rank halos
+// via IPC direct writes -- the 8-GPU single-node bench shape). The step
+// splits at the cross-rank barrier:
+//   A[par] = [interior jacobi -> translates (self + IPC views) ->
+//             staged thin packs (parity)]
+//   <barrier: all ranks' A complete -- RCCL all-reduce posted on the
+//    SAME stream (fully stream-ordered), or a gloo host barrier with a
+//    stream sync around it>
+//   B[par] = [staged unpacks (parity) -> exterior slabs -> device table
+//             swap -> device view flips]
+// With the device barrier, steps queue back-to-back with ONE host sync
+// per run(n): the eager path's ~0.21 ms/step of host orchestration
+// (measured 0.989 vs 0.780 ms, profiles/r2/r2_gpu7_eager.log) collapses
+// to two graph launches + one collective post per step.
+namespace {
+struct MrStepGraph {
+  hipStream_t stream = nullptr;
+  hipGraphExec_t execA[2] = {nullptr, nullptr};
+  hipGraphExec_t execB[2] = {nullptr, nullptr};
+  int parity = 0;
+  ExchangeEngine *eng = nullptr;
+  int dom = 0;
+};
+std::vector<std::unique_ptr<MrStepGraph>> g_mrGraphs;
+} // namespace
+
+int64_t jacobi_mr_graph_create(ExchangeEngine &eng, int dom, int64_t qi, const Rect3 &interior,
+                               const Rect3 &computeRegion, const std::vector<Rect3> &exteriors,
+                               int extendVec) {
+  LocalDomain &d = eng.domain(dom);
+  STENCIL_HIP(hipSetDevice(d.gpu()));
+  auto sg = std::make_unique<MrStepGraph>();
+  sg->eng = &eng;
+  sg->dom = dom;
+  STENCIL_HIP(hipStreamCreateWithFlags(&sg->stream, hipStreamNonBlocking));
+  for (int par = 0; par < 2; ++par) {
+    // A: interior + outgoing halos
+    STENCIL_HIP(hipStreamBeginCapture(sg->stream, hipStreamCaptureModeThreadLocal));
+    launch_jacobi_on(d, qi, interior, computeRegion, sg->stream, extendVec);
+    eng.launch_translates_plain_on((uintptr_t)sg->stream, 0);
+    eng.launch_packs_plain_on((uintptr_t)sg->stream, 1 + par); // staged parity
+    hipGraph_t ga = nullptr;
+    STENCIL_HIP(hipStreamEndCapture(sg->stream, &ga));
+    STENCIL_HIP(hipGraphInstantiate(&sg->execA[par], ga, nullptr, nullptr, 0));
+    STENCIL_HIP(hipGraphDestroy(ga));
+    // B: incoming halos + exterior + swap
+    STENCIL_HIP(hipStreamBeginCapture(sg->stream, hipStreamCaptureModeThreadLocal));
+    eng.launch_unpacks_plain_on((uintptr_t)sg->stream, 1 + par);
+    for (const Rect3 &box : exteriors)
+      launch_jacobi_on(d, qi, box, computeRegion, sg->stream, /*fullRectVec=*/0);
+    d.enqueue_table_swap(sg->stream);
+    eng.enqueue_view_flips((uintptr_t)sg->stream);
+    hipGraph_t gb = nullptr;
+    STENCIL_HIP(hipStreamEndCapture(sg->stream, &gb));
+    STENCIL_HIP(hipGraphInstantiate(&sg->execB[par], gb, nullptr, nullptr, 0));
+    STENCIL_HIP(hipGraphDestroy(gb));
+    d.swap(); // bake the other parity's kernarg pointers next round
+  }
+  g_mrGraphs.push_back(std::move(sg));
+  return (int64_t)g_mrGraphs.size() - 1;
+}
+
+uintptr_t jacobi_mr_graph_stream(int64_t handle) {
+  return (uintptr_t)g_mrGraphs.at(handle)->stream;
+}
+
+void jacobi_mr_graph_pre(int64_t handle) {
+  MrStepGraph &sg = *g_mrGraphs.at(handle);
+  STENCIL_HIP(hipSetDevice(sg.eng->domain(sg.dom).gpu()));
+  STENCIL_HIP(hipGraphLaunch(sg.execA[sg.parity], sg.stream));
+}
+
+void jacobi_mr_graph_post(int64_t handle) {
+  MrStepGraph &sg = *g_mrGraphs.at(handle);
+  LocalDomain &d = sg.eng->domain(sg.dom);
+  STENCIL_HIP(hipSetDevice(d.gpu()));
+  STENCIL_HIP(hipGraphLaunch(sg.execB[sg.parity], sg.stream));
+  sg.parity ^= 1;
+  d.swap_host_only(); // in-graph kernels flip the device state
+  sg.eng->flip_views_host_only();
+}
+
+void jacobi_mr_graph_sync(int64_t handle) {
+  MrStepGraph &sg = *g_mrGraphs.at(handle);
+  STENCIL_HIP(hipSetDevice(sg.eng->domain(sg.dom).gpu()));
+  STENCIL_HIP(hipStreamSynchronize(sg.stream));
+}
+
 void fill_f32(ExchangeEngine &eng, int dom, int64_t qi, const Rect3 &region, float value,
               bool nextBuf) {
   LocalDomain &d = eng.domain(dom);

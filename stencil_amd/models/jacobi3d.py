@@ -67,17 +67,64 @@ class Jacobi3D:
         import os
 
         self._graph = None
+        self._mr_graph = None
+        graphs_on = os.environ.get("STENCIL_AMD_STEP_GRAPH", "1") != "0"
         if (
             self.m == 1
             and self.dd.backend_kind == "native"
             and self.dd.comm.world_size == 1
             and self.dd.num_local() == 1
-            and os.environ.get("STENCIL_AMD_STEP_GRAPH", "1") != "0"
+            and graphs_on
         ):
             lo, hi = self.dd.local_rect(0)
             self._graph = self.dd.backend.jacobi_graph_create(
                 0, self.h.index, lo, hi, self.compute_lo, self.compute_hi
             )
+        elif (
+            self.m == 1
+            and self.dd.backend_kind == "native"
+            and self.dd.comm.world_size > 1
+            and self.dd.num_local() == 1
+            and graphs_on
+            and getattr(self.dd.backend, "_ipc_active", False)
+            and not any(getattr(self.dd.backend, "_has_wire", [True]))
+        ):
+            # multi-rank whole-step graphs (single-node 1-rank/GPU shape:
+            # every cross-rank halo is an IPC direct write or staged thin
+            # pack, so the step is two graphs around the colo barrier;
+            # see csrc/src/jacobi.hip jacobi_mr_graph_create)
+            from .. import _C
+
+            b = self.dd.backend
+            ilo, ihi = self.interiors[0]
+            self._mr_graph = _C.jacobi_mr_graph_create(
+                b.engine, 0, self.h.index,
+                _C.Rect3(_C.Vec3(*ilo), _C.Vec3(*ihi)),
+                _C.Rect3(_C.Vec3(*self.compute_lo), _C.Vec3(*self.compute_hi)),
+                [_C.Rect3(_C.Vec3(*blo), _C.Vec3(*bhi)) for blo, bhi in self.exteriors[0]],
+                extend_vec=2,
+            )
+
+    def _mr_step(self, sync: bool = True):
+        """one queued multi-rank graph step: A, barrier, B (see realize)"""
+        from .. import _C
+
+        b = self.dd.backend
+        _C.jacobi_mr_graph_pre(self._mr_graph)
+        if b._colo_wire is not None:
+            # all-reduce posted on the graph stream: fully stream-ordered
+            b._colo_wire.barrier(_C.jacobi_mr_graph_stream(self._mr_graph))
+        else:
+            # host barrier (shared-device/test configs): A must be done
+            # before signalling, B only after every rank signalled
+            import torch.distributed as dist
+
+            _C.jacobi_mr_graph_sync(self._mr_graph)
+            dist.barrier(group=b._colo_group)
+        _C.jacobi_mr_graph_post(self._mr_graph)
+        b._colo_parity[0] ^= 1  # keep the eager path's mirror in sync
+        if sync or b._colo_wire is None:
+            _C.jacobi_mr_graph_sync(self._mr_graph)
 
     def step(self, overlap: bool = True):
         dd = self.dd
@@ -86,6 +133,9 @@ class Jacobi3D:
             return
         if self._graph is not None:
             dd.backend.jacobi_graph_step(self._graph, 1)
+            return
+        if self._mr_graph is not None:
+            self._mr_step()
             return
         self._eager_step(overlap)
 
@@ -101,6 +151,14 @@ class Jacobi3D:
 
             _C.jacobi_graph_launch(self._graph, n)
             _C.jacobi_graph_sync(self._graph)
+        elif self._mr_graph is not None and self.m == 1:
+            from .. import _C
+
+            # with the device barrier the n steps queue back-to-back and
+            # only the final sync touches the host
+            for _ in range(n):
+                self._mr_step(sync=False)
+            _C.jacobi_mr_graph_sync(self._mr_graph)
         else:
             for _ in range(n):
                 self.step()

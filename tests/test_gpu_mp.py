@@ -74,15 +74,17 @@ def test_native_multi_rank_one_gpu(world, use_ipc, port):
         assert status == "ok", f"rank {rank}: {status}"
 
 
-def _jacobi_worker(rank, world, port, q, use_ipc):
+def _jacobi_worker(rank, world, port, q, use_ipc, step_graph=True):
     """3 jacobi steps across 2 ranks, every rank returns its interior for
     comparison against the single-process torch reference (validates the
-    mode-1/mode-2 fast-kernel paths under the real multi-rank transports)"""
+    mode-1/mode-2 fast-kernel paths AND the multi-rank whole-step graphs
+    under the real multi-rank transports)"""
     try:
         os.environ["MASTER_ADDR"] = "127.0.0.1"
         os.environ["MASTER_PORT"] = str(port)
         os.environ["STENCIL_AMD_WIRE"] = "cpu"
         os.environ["STENCIL_AMD_IPC"] = "1" if use_ipc else "0"
+        os.environ["STENCIL_AMD_STEP_GRAPH"] = "1" if step_graph else "0"
         import torch.distributed as dist
 
         dist.init_process_group("gloo", rank=rank, world_size=world)
@@ -98,6 +100,12 @@ def _jacobi_worker(rank, world, port, q, use_ipc):
         app.realize()
         if use_ipc:
             assert app.dd.backend._ipc_active, app.dd.backend._ipc_error
+            # the multi-rank whole-step graph path must actually engage
+            # when graphs are on (every cross-rank halo is IPC here)
+            if step_graph:
+                assert app._mr_graph is not None
+            else:
+                assert app._mr_graph is None
         fill_interiors(app.dd, app.h)
         for _ in range(3):
             app.step()
@@ -113,8 +121,11 @@ def _jacobi_worker(rank, world, port, q, use_ipc):
         q.put((rank, f"FAIL: {e}\n{traceback.format_exc()}", None))
 
 
-@pytest.mark.parametrize("use_ipc,port", [(True, 29731), (False, 29735)])
-def test_multi_rank_jacobi_matches_torch(use_ipc, port):
+@pytest.mark.parametrize(
+    "use_ipc,port,step_graph",
+    [(True, 29731, True), (True, 29739, False), (False, 29735, True)],
+)
+def test_multi_rank_jacobi_matches_torch(use_ipc, port, step_graph):
     # torch single-process global reference
     import sys
 
@@ -131,7 +142,10 @@ def test_multi_rank_jacobi_matches_torch(use_ipc, port):
 
     ctx = mp.get_context("spawn")
     q = ctx.Queue()
-    procs = [ctx.Process(target=_jacobi_worker, args=(r, 2, port, q, use_ipc)) for r in range(2)]
+    procs = [
+        ctx.Process(target=_jacobi_worker, args=(r, 2, port, q, use_ipc, step_graph))
+        for r in range(2)
+    ]
     for p in procs:
         p.start()
     results = [q.get(timeout=300) for _ in procs]
