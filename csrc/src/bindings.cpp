@@ -170,9 +170,15 @@ PYBIND11_MODULE(_C, m) {
       .def("elem_size", &LocalDomain::elem_size)
       .def("region_to_host",
            [](const LocalDomain &d, const Vec3 &pos, const Vec3 &ext, int64_t qi, bool fromNext) {
-             std::string out(ext.flatten() * d.elem_size(qi), '\0');
-             d.region_to_host(out.data(), pos, ext, qi, fromNext);
-             return py::bytes(out);
+             // uninitialized PyBytes + direct fill: the round-2 pinned
+             // bounce was bottlenecked by the std::string zero-fill and
+             // the py::bytes copy (two extra full host passes)
+             const int64_t n = ext.flatten() * d.elem_size(qi);
+             py::bytes out = py::reinterpret_steal<py::bytes>(
+                 PyBytes_FromStringAndSize(nullptr, (Py_ssize_t)n));
+             if (!out) throw std::bad_alloc();
+             d.region_to_host(PyBytes_AS_STRING(out.ptr()), pos, ext, qi, fromNext);
+             return out;
            },
            py::arg("pos"), py::arg("ext"), py::arg("qi"), py::arg("from_next") = false)
       .def("curr_pitch", [](const LocalDomain &d, int64_t qi) { return d.curr(qi).pitch; })
@@ -182,12 +188,13 @@ PYBIND11_MODULE(_C, m) {
            [](const LocalDomain &d, int64_t qi, bool next) { return py::bytes(d.ipc_handle(qi, next)); },
            py::arg("qi"), py::arg("next") = false)
       .def("region_from_host",
-           [](const LocalDomain &d, py::bytes data, const Vec3 &pos, const Vec3 &ext, int64_t qi,
+           [](const LocalDomain &d, py::buffer data, const Vec3 &pos, const Vec3 &ext, int64_t qi,
               bool toNext) {
-             std::string s = data;
-             if ((int64_t)s.size() != ext.flatten() * d.elem_size(qi))
+             // buffer protocol (bytes or a contiguous array), no copy
+             const py::buffer_info info = data.request();
+             if (info.size * info.itemsize != ext.flatten() * d.elem_size(qi))
                throw std::runtime_error("region_from_host: size mismatch");
-             d.region_from_host(s.data(), pos, ext, qi, toNext);
+             d.region_from_host(info.ptr, pos, ext, qi, toNext);
            },
            py::arg("data"), py::arg("pos"), py::arg("ext"), py::arg("qi"),
            py::arg("to_next") = false);

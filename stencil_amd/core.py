@@ -389,7 +389,13 @@ class DistributedDomain:
         flo = (o[0] - r.x(-1), o[1] - r.y(-1), o[2] - r.z(-1))
         pos = tuple(lo[i] - flo[i] for i in range(3))
         ext = (arr.shape[2], arr.shape[1], arr.shape[0])
-        self.backend.write_region(li, arr.tobytes(), pos, ext, handle.index, to_next)
+        # native backend takes any contiguous buffer zero-copy; torch
+        # backend wants bytes
+        if self.backend_kind == "native":
+            data = np.ascontiguousarray(arr)
+        else:
+            data = arr.tobytes()
+        self.backend.write_region(li, data, pos, ext, handle.index, to_next)
 
     # ---- checkpoint/restore ----
     def save_checkpoint(self, path: str):
