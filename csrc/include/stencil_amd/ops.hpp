@@ -72,6 +72,18 @@ int64_t mhd_graph_create(ExchangeEngine &eng, int dom, const Rect3 &region, doub
 void mhd_graph_iter(int64_t handle, int64_t nIters = 1);
 void mhd_graph_sync(int64_t handle);
 
+// Multi-rank substep graphs (see csrc/src/mhd.hip): per substep the
+// caller runs phase1 / barrier / phase2 / barrier / phase3 (barrier =
+// RcclWire::barrier on mhd_mr_graph_stream, or a host barrier with a
+// stream sync before it).
+int64_t mhd_mr_graph_create(ExchangeEngine &eng, int dom, const Rect3 &interior,
+                            const std::vector<Rect3> &exteriors, double dt, const MhdCoeffs &cf);
+uintptr_t mhd_mr_graph_stream(int64_t handle);
+void mhd_mr_phase1(int64_t handle);
+void mhd_mr_phase2(int64_t handle);
+void mhd_mr_phase3(int64_t handle);
+void mhd_mr_graph_sync(int64_t handle);
+
 void mhd_substep(ExchangeEngine &eng, int dom, const Rect3 &region, int step, double dt,
                  const MhdCoeffs &cf, int streamId = 0);
 

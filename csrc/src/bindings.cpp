@@ -292,6 +292,13 @@ PYBIND11_MODULE(_C, m) {
         py::arg("dt"), py::arg("cf"));
   m.def("mhd_graph_iter", &mhd_graph_iter, py::arg("handle"), py::arg("n_iters") = 1);
   m.def("mhd_graph_sync", &mhd_graph_sync);
+  m.def("mhd_mr_graph_create", &mhd_mr_graph_create, py::arg("eng"), py::arg("dom"),
+        py::arg("interior"), py::arg("exteriors"), py::arg("dt"), py::arg("cf"));
+  m.def("mhd_mr_graph_stream", &mhd_mr_graph_stream);
+  m.def("mhd_mr_phase1", &mhd_mr_phase1);
+  m.def("mhd_mr_phase2", &mhd_mr_phase2);
+  m.def("mhd_mr_phase3", &mhd_mr_phase3);
+  m.def("mhd_mr_graph_sync", &mhd_mr_graph_sync);
 
   py::class_<MhdCoeffs>(m, "MhdCoeffs")
       .def(py::init<>())

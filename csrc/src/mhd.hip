@@ -692,4 +692,126 @@ void mhd_graph_sync(int64_t handle) {
   STENCIL_HIP(hipStreamSynchronize(sg.stream));
 }
 
+// Multi-rank substep graphs (one rank, one domain, cross-rank halos via
+// IPC -- the 8-GPU single-node weak-scaling shape). Each RK3 substep
+// splits at the TWO cross-rank barriers (field halos X1, div halos X2):
+//   G1[par]    = [div(interior) || (translates g0 + staged packs g0)]
+//   <barrier: every rank's field halos written>
+//   G2[s][par] = [staged unpacks g0 -> div(exteriors) ->
+//                 (translates g1 + staged packs g1) || substep(interior)]
+//   <barrier: every rank's div halos written>
+//   G3[s][par] = [staged unpacks g1 -> substep(exteriors) ->
+//                 device table swap + view flips]
+// The skew-safety argument is the jacobi one (jacobi_mr_graph_*): each
+// barrier bounds peers to one phase, staged buffers are double-buffered
+// by parity, and direct writes land in halo rings disjoint from locally
+// written compute cells.
+namespace {
+struct MhdMrGraph {
+  hipStream_t stream = nullptr, stream2 = nullptr, stream3 = nullptr;
+  hipEvent_t evF = nullptr, evJ2 = nullptr, evJ3 = nullptr;
+  hipGraphExec_t g1[2] = {nullptr, nullptr};
+  hipGraphExec_t g2[3][2] = {};
+  hipGraphExec_t g3[3][2] = {};
+  int parity = 0, substep = 0;
+  ExchangeEngine *eng = nullptr;
+  int dom = 0;
+};
+std::vector<std::unique_ptr<MhdMrGraph>> g_mhdMrGraphs;
+} // namespace
+
+int64_t mhd_mr_graph_create(ExchangeEngine &eng, int dom, const Rect3 &interior,
+                            const std::vector<Rect3> &exteriors, double dt, const MhdCoeffs &cf) {
+  LocalDomain &d = eng.domain(dom);
+  STENCIL_HIP(hipSetDevice(d.gpu()));
+  auto sg = std::make_unique<MhdMrGraph>();
+  sg->eng = &eng;
+  sg->dom = dom;
+  STENCIL_HIP(hipStreamCreateWithFlags(&sg->stream, hipStreamNonBlocking));
+  STENCIL_HIP(hipStreamCreateWithFlags(&sg->stream2, hipStreamNonBlocking));
+  STENCIL_HIP(hipStreamCreateWithFlags(&sg->stream3, hipStreamNonBlocking));
+  STENCIL_HIP(hipEventCreateWithFlags(&sg->evF, hipEventDisableTiming));
+  STENCIL_HIP(hipEventCreateWithFlags(&sg->evJ2, hipEventDisableTiming));
+  STENCIL_HIP(hipEventCreateWithFlags(&sg->evJ3, hipEventDisableTiming));
+  auto instantiate = [&](hipGraphExec_t &exec) {
+    hipGraph_t g = nullptr;
+    STENCIL_HIP(hipStreamEndCapture(sg->stream, &g));
+    STENCIL_HIP(hipGraphInstantiate(&exec, g, nullptr, nullptr, 0));
+    STENCIL_HIP(hipGraphDestroy(g));
+  };
+  for (int par = 0; par < 2; ++par) {
+    // G1: div(interior) concurrent with the outgoing field halos
+    STENCIL_HIP(hipStreamBeginCapture(sg->stream, hipStreamCaptureModeThreadLocal));
+    STENCIL_HIP(hipEventRecord(sg->evF, sg->stream));
+    STENCIL_HIP(hipStreamWaitEvent(sg->stream2, sg->evF, 0));
+    mhd_div_launch_on(d, interior, cf, sg->stream);
+    eng.launch_translates_plain_on((uintptr_t)sg->stream2, 0);
+    eng.launch_packs_plain_on((uintptr_t)sg->stream2, 1 + par);
+    STENCIL_HIP(hipEventRecord(sg->evJ2, sg->stream2));
+    STENCIL_HIP(hipStreamWaitEvent(sg->stream, sg->evJ2, 0));
+    instantiate(sg->g1[par]);
+    for (int s = 0; s < 3; ++s) {
+      // G2: incoming field halos -> div(ext) -> div halos out || interior
+      STENCIL_HIP(hipStreamBeginCapture(sg->stream, hipStreamCaptureModeThreadLocal));
+      eng.launch_unpacks_plain_on((uintptr_t)sg->stream, 1 + par);
+      for (const Rect3 &box : exteriors) mhd_div_launch_on(d, box, cf, sg->stream);
+      STENCIL_HIP(hipEventRecord(sg->evF, sg->stream));
+      STENCIL_HIP(hipStreamWaitEvent(sg->stream2, sg->evF, 0));
+      STENCIL_HIP(hipStreamWaitEvent(sg->stream3, sg->evF, 0));
+      eng.launch_translates_plain_on((uintptr_t)sg->stream, 1);
+      eng.launch_packs_plain_on((uintptr_t)sg->stream, 4 + par); // 3*1+1+par
+      mhd_substep_launch_on(d, interior, s, dt, cf, sg->stream2, sg->stream3);
+      STENCIL_HIP(hipEventRecord(sg->evJ2, sg->stream2));
+      STENCIL_HIP(hipEventRecord(sg->evJ3, sg->stream3));
+      STENCIL_HIP(hipStreamWaitEvent(sg->stream, sg->evJ2, 0));
+      STENCIL_HIP(hipStreamWaitEvent(sg->stream, sg->evJ3, 0));
+      instantiate(sg->g2[s][par]);
+      // G3: incoming div halos -> exterior shells -> swap
+      STENCIL_HIP(hipStreamBeginCapture(sg->stream, hipStreamCaptureModeThreadLocal));
+      eng.launch_unpacks_plain_on((uintptr_t)sg->stream, 4 + par);
+      for (const Rect3 &box : exteriors)
+        mhd_substep_launch_on(d, box, s, dt, cf, sg->stream, sg->stream);
+      d.enqueue_table_swap(sg->stream);
+      eng.enqueue_view_flips((uintptr_t)sg->stream);
+      instantiate(sg->g3[s][par]);
+    }
+    d.swap(); // bake the other parity's kernarg pointers next round
+  }
+  g_mhdMrGraphs.push_back(std::move(sg));
+  return (int64_t)g_mhdMrGraphs.size() - 1;
+}
+
+uintptr_t mhd_mr_graph_stream(int64_t handle) {
+  return (uintptr_t)g_mhdMrGraphs.at(handle)->stream;
+}
+
+void mhd_mr_phase1(int64_t handle) {
+  MhdMrGraph &sg = *g_mhdMrGraphs.at(handle);
+  STENCIL_HIP(hipSetDevice(sg.eng->domain(sg.dom).gpu()));
+  STENCIL_HIP(hipGraphLaunch(sg.g1[sg.parity], sg.stream));
+}
+
+void mhd_mr_phase2(int64_t handle) {
+  MhdMrGraph &sg = *g_mhdMrGraphs.at(handle);
+  STENCIL_HIP(hipSetDevice(sg.eng->domain(sg.dom).gpu()));
+  STENCIL_HIP(hipGraphLaunch(sg.g2[sg.substep][sg.parity], sg.stream));
+}
+
+void mhd_mr_phase3(int64_t handle) {
+  MhdMrGraph &sg = *g_mhdMrGraphs.at(handle);
+  LocalDomain &d = sg.eng->domain(sg.dom);
+  STENCIL_HIP(hipSetDevice(d.gpu()));
+  STENCIL_HIP(hipGraphLaunch(sg.g3[sg.substep][sg.parity], sg.stream));
+  sg.substep = (sg.substep + 1) % 3;
+  sg.parity ^= 1;
+  d.swap_host_only(); // in-graph kernels flip the device state
+  sg.eng->flip_views_host_only();
+}
+
+void mhd_mr_graph_sync(int64_t handle) {
+  MhdMrGraph &sg = *g_mhdMrGraphs.at(handle);
+  STENCIL_HIP(hipSetDevice(sg.eng->domain(sg.dom).gpu()));
+  STENCIL_HIP(hipStreamSynchronize(sg.stream));
+}
+
 } // namespace stencil_amd

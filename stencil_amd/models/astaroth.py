@@ -116,12 +116,14 @@ class Astaroth:
         import os
 
         self._graph = None
+        self._mr_graph = None
         self._graph_dt = None
+        graphs_on = os.environ.get("STENCIL_AMD_STEP_GRAPH", "1") != "0"
         if (
             self.dd.backend_kind == "native"
             and self.dd.comm.world_size == 1
             and self.dd.num_local() == 1
-            and os.environ.get("STENCIL_AMD_STEP_GRAPH", "1") != "0"
+            and graphs_on
         ):
             lo, hi = self.dd.local_rect(0)
             rect = _C.Rect3(_C.Vec3(*lo), _C.Vec3(*hi))
@@ -129,6 +131,49 @@ class Astaroth:
             self._graph = _C.mhd_graph_create(
                 self.dd.backend.engine, 0, rect, self._graph_dt, self.cf
             )
+        elif (
+            self.dd.backend_kind == "native"
+            and self.dd.comm.world_size > 1
+            and self.dd.num_local() == 1
+            and graphs_on
+            and getattr(self.dd.backend, "_ipc_active", False)
+            and not any(getattr(self.dd.backend, "_has_wire", [True]))
+        ):
+            # multi-rank substep graphs (single-node 1-rank/GPU shape):
+            # per substep, three graphs around the two colo barriers
+            # (csrc/src/mhd.hip mhd_mr_graph_create)
+            ilo, ihi = self.interiors[0]
+            self._graph_dt = float(self.conf["dt"])
+            self._mr_graph = _C.mhd_mr_graph_create(
+                self.dd.backend.engine, 0,
+                _C.Rect3(_C.Vec3(*ilo), _C.Vec3(*ihi)),
+                [_C.Rect3(_C.Vec3(*blo), _C.Vec3(*bhi)) for blo, bhi in self.exteriors[0]],
+                self._graph_dt, self.cf,
+            )
+
+    def _mr_barrier(self, sync_first: bool):
+        b = self.dd.backend
+        if b._colo_wire is not None:
+            b._colo_wire.barrier(_C.mhd_mr_graph_stream(self._mr_graph))
+        else:
+            import torch.distributed as dist
+
+            _C.mhd_mr_graph_sync(self._mr_graph)
+            dist.barrier(group=b._colo_group)
+
+    def _mr_iter(self, sync: bool = True):
+        """one queued RK3 iteration through the multi-rank graphs"""
+        b = self.dd.backend
+        for _s in range(3):
+            _C.mhd_mr_phase1(self._mr_graph)
+            self._mr_barrier(True)
+            _C.mhd_mr_phase2(self._mr_graph)
+            self._mr_barrier(True)
+            _C.mhd_mr_phase3(self._mr_graph)
+            b._colo_parity[0] ^= 1  # keep the eager path's mirrors in sync
+            b._colo_parity[1] ^= 1
+        if sync or b._colo_wire is None:
+            _C.mhd_mr_graph_sync(self._mr_graph)
 
 
     def init_fields(self):
@@ -223,6 +268,9 @@ class Astaroth:
             _C.mhd_graph_iter(self._graph, 1)
             _C.mhd_graph_sync(self._graph)
             return
+        if self._mr_graph is not None and compute and dt == self._graph_dt:
+            self._mr_iter()
+            return
         for s in range(3):
             self._substep(s, dt, compute, overlap)
 
@@ -232,6 +280,11 @@ class Astaroth:
         if self._graph is not None and self.dd.comm.world_size == 1:
             _C.mhd_graph_iter(self._graph, n)
             _C.mhd_graph_sync(self._graph)
+            return
+        if self._mr_graph is not None:
+            for _ in range(n):
+                self._mr_iter(sync=False)
+            _C.mhd_mr_graph_sync(self._mr_graph)
             return
         for _ in range(n):
             self.step()

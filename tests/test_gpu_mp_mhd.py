@@ -11,11 +11,12 @@ import pytest
 pytestmark = pytest.mark.gpu
 
 
-def _worker(rank, world, port, q):
+def _worker(rank, world, port, q, step_graph=True):
     try:
         os.environ["MASTER_ADDR"] = "127.0.0.1"
         os.environ["MASTER_PORT"] = str(port)
         os.environ["STENCIL_AMD_WIRE"] = "cpu"
+        os.environ["STENCIL_AMD_STEP_GRAPH"] = "1" if step_graph else "0"
         import torch.distributed as dist
 
         dist.init_process_group("gloo", rank=rank, world_size=world)
@@ -29,6 +30,9 @@ def _worker(rank, world, port, q):
         size = (24, 24, 24)
         app = Astaroth(size, backend="native")
         app.realize()
+        # the multi-rank substep-graph path must engage iff graphs are on
+        # (every cross-rank halo here is IPC)
+        assert (app._mr_graph is not None) == step_graph
         app.init_fields()
 
         cf = {
@@ -58,10 +62,11 @@ def _worker(rank, world, port, q):
         q.put((rank, f"FAIL: {e}\n{traceback.format_exc()}"))
 
 
-def test_mhd_two_ranks_matches_numpy():
+@pytest.mark.parametrize("step_graph,port", [(True, 29771), (False, 29775)])
+def test_mhd_two_ranks_matches_numpy(step_graph, port):
     ctx = mp.get_context("spawn")
     q = ctx.Queue()
-    procs = [ctx.Process(target=_worker, args=(r, 2, 29771, q)) for r in range(2)]
+    procs = [ctx.Process(target=_worker, args=(r, 2, port, q, step_graph)) for r in range(2)]
     for p in procs:
         p.start()
     results = [q.get(timeout=300) for _ in procs]
