@@ -39,12 +39,26 @@ CASES = [
     ((64, 64, 64), 2, 2, 4, "node_aware"),
     ((30, 30, 30), 1, 1, 8, "node_aware"),
     ((33, 31, 29), 3, 1, 2, "trivial"),
+    ((40, 36, 28), "asym", 1, 4, "node_aware"),
+    ((40, 36, 28), "fec", 1, 8, "trivial"),
 ]
+
+
+def make_radius(r):
+    if r == "asym":  # the reference's +x=2/-x=1 asymmetric test shape
+        rad = _C.Radius.constant(1)
+        rad.set_dir(1, 0, 0, 2)
+        rad.set_dir(-1, 0, 0, 1)
+        rad.set_dir(1, 1, 0, 2)
+        return rad
+    if r == "fec":  # face/edge/corner radii 2/1/0
+        return _C.Radius.face_edge_corner(2, 1, 0)
+    return _C.Radius.constant(r)
 
 
 @pytest.mark.parametrize("size,r,n_nodes,gpn,strategy", CASES)
 def test_cpp_python_plan_parity(size, r, n_nodes, gpn, strategy):
-    radius = _C.Radius.constant(r)
+    radius = make_radius(r)
     slots = make_slots(n_nodes, gpn)
     slot_tuples = [(s.rank, s.local_id, s.cuda, s.node) for s in slots]
     cpp = _C.cpp_plan(
