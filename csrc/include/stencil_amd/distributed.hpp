@@ -95,6 +95,11 @@ public:
   int64_t bytes_translate() const { return bytesTranslate_; }
   int64_t bytes_wire() const { return bytesWire_; }
 
+  // dump each local subdomain interior as CSV 'Z,Y,X,q0,q1,...'
+  // (reference src/stencil.cu:1188-1264; same format as the Python
+  // write_paraview)
+  void write_paraview(const std::string &prefix);
+
 private:
   void gather_slots_(std::vector<Slot> &slots);
 

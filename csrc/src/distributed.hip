@@ -292,6 +292,52 @@ void DistributedDomain::swap() {
   for (auto &d : domains_) d->swap();
 }
 
+void DistributedDomain::write_paraview(const std::string &prefix) {
+  for (int li = 0; li < num_local(); ++li) {
+    LocalDomain &d = *domains_[li];
+    const Rect3 r = local_rect(li);
+    const Vec3 ext = r.extent();
+    const Rect3 full = d.full_region();
+    const Vec3 pos = r.lo - full.lo; // allocation coords
+    const int64_t nq = (int64_t)data_.size();
+    std::vector<std::vector<double>> vals(nq);
+    for (int64_t qi = 0; qi < nq; ++qi) {
+      const int64_t es = data_[qi].first;
+      std::vector<char> raw(ext.flatten() * es);
+      d.region_to_host(raw.data(), pos, ext, qi, false);
+      vals[qi].resize(ext.flatten());
+      for (int64_t i = 0; i < ext.flatten(); ++i) {
+        switch (es) { // size-canonical interpretation (fp32/fp64/int16/u8)
+        case 8: vals[qi][i] = ((const double *)raw.data())[i]; break;
+        case 4: vals[qi][i] = ((const float *)raw.data())[i]; break;
+        case 2: vals[qi][i] = ((const int16_t *)raw.data())[i]; break;
+        default: vals[qi][i] = ((const uint8_t *)raw.data())[i]; break;
+        }
+      }
+    }
+    const Vec3 idx = placement_->get_idx(rank_, li);
+    const int64_t gid = placement_->linearize(idx);
+    FILE *f = fopen((prefix + std::to_string(gid) + ".txt").c_str(), "w");
+    if (!f) throw std::runtime_error("write_paraview: cannot open output");
+    fprintf(f, "Z,Y,X");
+    for (int64_t qi = 0; qi < nq; ++qi) {
+      const std::string name =
+          data_[qi].second.empty() ? "q" + std::to_string(qi) : data_[qi].second;
+      fprintf(f, ",%s", name.c_str());
+    }
+    fprintf(f, "\n");
+    int64_t i = 0;
+    for (int64_t z = r.lo.z; z < r.hi.z; ++z)
+      for (int64_t y = r.lo.y; y < r.hi.y; ++y)
+        for (int64_t x = r.lo.x; x < r.hi.x; ++x, ++i) {
+          fprintf(f, "%lld,%lld,%lld", (long long)z, (long long)y, (long long)x);
+          for (int64_t qi = 0; qi < nq; ++qi) fprintf(f, ",%.17g", vals[qi][i]);
+          fprintf(f, "\n");
+        }
+    fclose(f);
+  }
+}
+
 Rect3 DistributedDomain::local_rect(int li) const {
   const Vec3 idx = placement_->get_idx(rank_, li);
   const Vec3 o = placement_->subdomain_origin(idx);
