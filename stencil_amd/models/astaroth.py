@@ -268,9 +268,12 @@ class Astaroth:
             _C.mhd_graph_iter(self._graph, 1)
             _C.mhd_graph_sync(self._graph)
             return
-        if self._mr_graph is not None and compute and dt == self._graph_dt:
-            self._mr_iter()
-            return
+        if self._mr_graph is not None:
+            if compute and dt == self._graph_dt:
+                self._mr_iter()
+                return
+            # parity cannot be resynced after an eager substep; degrade
+            self._mr_graph = None
         for s in range(3):
             self._substep(s, dt, compute, overlap)
 

@@ -135,8 +135,12 @@ class Jacobi3D:
             dd.backend.jacobi_graph_step(self._graph, 1)
             return
         if self._mr_graph is not None:
-            self._mr_step()
-            return
+            if overlap:
+                self._mr_step()
+                return
+            # the graph's parity cannot be resynced once an eager step
+            # interleaves; degrade one-way to the eager path
+            self._mr_graph = None
         self._eager_step(overlap)
 
     def run(self, n: int) -> float:
