@@ -7,6 +7,11 @@ import os
 import sys
 import time
 
+# exchange-only benchmark: no compute to overlap with, so let the copy
+# batches use their natural grid instead of the 512-block contention cap
+# (tools can still override)
+os.environ.setdefault("STENCIL_AMD_COPY_DIV", "1")
+
 sys.path.insert(0, os.path.join(os.path.dirname(__file__), ".."))
 
 import numpy as np
