@@ -3,6 +3,7 @@
 // (planning.hpp == planning.py, pinned by tests/test_native_plan.py).
 #include "stencil_amd/distributed.hpp"
 #include "stencil_amd/hip_check.hpp"
+#include "stencil_amd/log.hpp"
 
 #include <hip/hip_runtime.h>
 
@@ -164,6 +165,9 @@ void DistributedDomain::realize() {
     d->realize();
     domains_.push_back(std::move(d));
   }
+  LOG_INFO("realize: rank %d/%d, %d local domain(s), dim (%lld,%lld,%lld)", rank_, world_,
+           nLocal, (long long)placement_->dim().x, (long long)placement_->dim().y,
+           (long long)placement_->dim().z);
   engine_ = std::make_unique<ExchangeEngine>(domains_);
   {
     std::set<int> devs;
@@ -250,6 +254,7 @@ void DistributedDomain::realize() {
       wireDev_ = devs.empty() ? domains_[0]->gpu() : *devs.begin();
       const std::string uid = rank_ == 0 ? RcclWire::unique_id() : std::string();
       const auto uids = boot_->allgather("rccl_uid", uid);
+      LOG_INFO("wire: RCCL communicator on device %d (rank %d/%d)", wireDev_, rank_, world_);
       wire_ = std::make_unique<RcclWire>(wireDev_, rank_, world_, uids[0]);
       for (int g = 0; g < ng; ++g)
         for (const WireOp &op : wireOps[g]) {
