@@ -7,10 +7,11 @@ import os
 import sys
 import time
 
-# exchange-only benchmark: no compute to overlap with, so let the copy
-# batches use their natural grid instead of the 512-block contention cap
-# (tools can still override)
-os.environ.setdefault("STENCIL_AMD_COPY_DIV", "1")
+# NOTE: the engine's 512-block copy cap is kept even for exchange-only
+# runs — measured FASTER than the natural full grid at 1024^3 (395 vs
+# 306 GB/s faces; profiles/r2/r2_gpu11_exch.log): with one block per 512
+# copy words the per-block tail and the scattered per-job addressing
+# dominate, while 512 grid-striding blocks keep every HBM channel busy.
 
 sys.path.insert(0, os.path.join(os.path.dirname(__file__), ".."))
 
