@@ -127,6 +127,11 @@ public:
   void launch_unpacks_plain_on(uintptr_t stream, int group = 0);
   void launch_packs(int group = 0);
   void launch_unpacks(int group = 0);
+  // stream-order every device's group-unpacks after every device's
+  // group-packs (events across pack streams) -- needed when a pack on
+  // one GPU fills a staging buffer that another GPU's unpack drains
+  // (the staged-local cross-device translate path)
+  void fence_packs_unpacks(int group = 0);
   // device-side parity flip of every remote view's pointer table (the
   // in-graph analog of flip_views, paired with flip_views_host_only) --
   // swaps devSlots <-> devSlotsAlt contents on `stream`

@@ -533,6 +533,27 @@ void ExchangeEngine::launch_unpacks(int group) {
   roctxRangePop();
 }
 
+void ExchangeEngine::fence_packs_unpacks(int group) {
+  static std::map<int, hipEvent_t> evs; // per source device, reused
+  for (auto &b : packBatches_[group]) {
+    if (b.jobs.empty()) continue;
+    auto it = evs.find(b.dev);
+    if (it == evs.end()) {
+      STENCIL_HIP(hipSetDevice(b.dev));
+      hipEvent_t e;
+      STENCIL_HIP(hipEventCreateWithFlags(&e, hipEventDisableTiming));
+      it = evs.emplace(b.dev, e).first;
+    }
+    STENCIL_HIP(hipSetDevice(b.dev));
+    STENCIL_HIP(hipEventRecord(it->second, pack_stream_(b.dev)));
+    for (auto &u : unpackBatches_[group]) {
+      if (u.jobs.empty() || u.dev == b.dev) continue; // same stream: ordered
+      STENCIL_HIP(hipSetDevice(u.dev));
+      STENCIL_HIP(hipStreamWaitEvent(pack_stream_(u.dev), it->second, 0));
+    }
+  }
+}
+
 void ExchangeEngine::sync_translates() {
   roctxRangePush("stencil::sync_translates");
   for (auto &kv : commStreams_) {

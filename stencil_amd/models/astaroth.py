@@ -119,11 +119,13 @@ class Astaroth:
         self._mr_graph = None
         self._graph_dt = None
         graphs_on = os.environ.get("STENCIL_AMD_STEP_GRAPH", "1") != "0"
+        staged_local = any(getattr(self.dd.backend, "_staged_local", []))
         if (
             self.dd.backend_kind == "native"
             and self.dd.comm.world_size == 1
             and self.dd.num_local() == 1
             and graphs_on
+            and not staged_local
         ):
             lo, hi = self.dd.local_rect(0)
             rect = _C.Rect3(_C.Vec3(*lo), _C.Vec3(*hi))
@@ -138,6 +140,7 @@ class Astaroth:
             and graphs_on
             and getattr(self.dd.backend, "_ipc_active", False)
             and not any(getattr(self.dd.backend, "_has_wire", [True]))
+            and not staged_local
         ):
             # multi-rank substep graphs (single-node 1-rank/GPU shape):
             # per substep, three graphs around the two colo barriers

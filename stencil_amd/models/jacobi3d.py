@@ -69,12 +69,14 @@ class Jacobi3D:
         self._graph = None
         self._mr_graph = None
         graphs_on = os.environ.get("STENCIL_AMD_STEP_GRAPH", "1") != "0"
+        staged_local = any(getattr(self.dd.backend, "_staged_local", []))
         if (
             self.m == 1
             and self.dd.backend_kind == "native"
             and self.dd.comm.world_size == 1
             and self.dd.num_local() == 1
             and graphs_on
+            and not staged_local
         ):
             lo, hi = self.dd.local_rect(0)
             self._graph = self.dd.backend.jacobi_graph_create(
@@ -88,6 +90,7 @@ class Jacobi3D:
             and graphs_on
             and getattr(self.dd.backend, "_ipc_active", False)
             and not any(getattr(self.dd.backend, "_has_wire", [True]))
+            and not staged_local
         ):
             # multi-rank whole-step graphs (single-node 1-rank/GPU shape:
             # every cross-rank halo is an IPC direct write or staged thin

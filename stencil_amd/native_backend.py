@@ -9,7 +9,13 @@ from __future__ import annotations
 from typing import List, Optional, Tuple
 
 from . import _C
-from .parallel.planning import ExchangePlan, pair_seq_tags, wire_layout, wire_layout_pairs
+from .parallel.planning import (
+    ExchangePlan,
+    Message,
+    pair_seq_tags,
+    wire_layout,
+    wire_layout_pairs,
+)
 
 Vec = Tuple[int, int, int]
 
@@ -60,6 +66,7 @@ class NativeBackend:
         self._colo_wire = None  # _C.RcclWire over colocated ranks (device barrier)
         self._ipc_error = None
         self._colo_parity = [0] * ng
+        self._staged_local = [False] * ng
         self._staging_recv = {}
 
     # ---- plan registration ----
@@ -98,6 +105,14 @@ class NativeBackend:
                     ipc_sends = []
         self._make_colo_groups(ctx)
         elem_sizes = [es for es, _ in self.data_defs]
+        # STENCIL_AMD_STAGE_LOCAL: auto (default) stages thin-row
+        # cross-DEVICE translates through a coalesced pack into a buffer
+        # on the destination GPU + local unpack (a direct xGMI scatter of
+        # 4-24 B rows is what the IPC staged path already avoids across
+        # processes); all = stage every local translate (test hook,
+        # exercises the path on one GPU); 0 = off.
+        stage_mode = os.environ.get("STENCIL_AMD_STAGE_LOCAL", "auto")
+        self._staged_local = [False] * len(self.groups)
         for g, qis in enumerate(self.groups):
             for t in plan.translates:
                 src = self.domains[t.src_local]
@@ -106,9 +121,34 @@ class NativeBackend:
                 nd = _vec3(tuple(-c for c in t.dir))
                 src_pos = src.halo_pos(d, False)
                 dst_pos = dst.halo_pos(nd, True)
-                self.engine.add_translate(
-                    t.src_local, t.dst_local, src_pos, dst_pos, _vec3(t.ext), g, sorted(qis)
-                )
+                cross = src.gpu() != dst.gpu()
+                if stage_mode == "all":
+                    staged_qis = sorted(qis)
+                elif stage_mode == "0" or not cross:
+                    staged_qis = []
+                else:
+                    staged_qis = [
+                        qi for qi in sorted(qis) if self._is_thin(t, elem_sizes[qi])
+                    ]
+                fat_qis = [qi for qi in sorted(qis) if qi not in staged_qis]
+                if fat_qis:
+                    self.engine.add_translate(
+                        t.src_local, t.dst_local, src_pos, dst_pos, _vec3(t.ext), g, fat_qis
+                    )
+                if staged_qis:
+                    m = Message(t.dir, 0, 0, t.ext)
+                    total, chunks = wire_layout_pairs(
+                        [(m, qi) for qi in staged_qis], elem_sizes
+                    )
+                    buf = self.engine.create_buffer(t.dst_local, total)
+                    for _m, qi, off, nb in chunks:
+                        self.engine.add_pack(
+                            t.src_local, buf, off, src_pos, _vec3(t.ext), qi, group=3 * g
+                        )
+                        self.engine.add_unpack(
+                            t.dst_local, buf, off, dst_pos, _vec3(t.ext), qi, group=3 * g
+                        )
+                    self._staged_local[g] = True
 
         seq = pair_seq_tags(plan)
         ng = len(self.groups)
@@ -452,7 +492,7 @@ class NativeBackend:
             # staged thin messages: coalesced pack straight into the
             # receiver's staging buffer (parity-selected half)
             self.engine.launch_packs(3 * g + 1 + self._colo_parity[g])
-        if self._has_wire[g]:
+        if self._has_wire[g] or self._staged_local[g]:
             self.engine.launch_packs(3 * g)
 
     def exchange_end(self, group: int = 0):
@@ -464,8 +504,12 @@ class NativeBackend:
                 # fully stream-ordered: the packs already sit on the pack
                 # stream (exchange_begin), the grouped send/recv is posted
                 # behind them, and the unpacks are enqueued behind the
-                # recvs -- the only host block is sync_all below
+                # recvs -- the only host block is sync_all below. Staged-
+                # local cross-device unpacks need the event fence (their
+                # packs ran on another device's stream).
                 self._wire.post(g, self.engine.pack_stream_handle(self._wire_dev))
+                if self._staged_local[g]:
+                    self.engine.fence_packs_unpacks(3 * g)
                 self.engine.launch_unpacks(3 * g)
             else:
                 import torch.distributed as dist
@@ -495,6 +539,12 @@ class NativeBackend:
                     for w in dist.batch_isend_irecv(ops):
                         w.wait()
                 self.engine.launch_unpacks(3 * g)
+        elif self._staged_local[g]:
+            # pack (src device) -> buffer on dst device -> unpack (dst
+            # device): cross-device ordering via the event fence, one
+            # host sync below
+            self.engine.fence_packs_unpacks(3 * g)
+            self.engine.launch_unpacks(3 * g)
         self.engine.sync_all()
         if self._ipc_active:
             # all colocated ranks' direct writes and staged packs are
