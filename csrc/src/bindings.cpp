@@ -243,6 +243,7 @@ PYBIND11_MODULE(_C, m) {
       .def("launch_translates", &ExchangeEngine::launch_translates, py::arg("group") = 0)
       .def("launch_packs", &ExchangeEngine::launch_packs, py::arg("group") = 0)
       .def("launch_unpacks", &ExchangeEngine::launch_unpacks, py::arg("group") = 0)
+      .def("fence_packs_unpacks", &ExchangeEngine::fence_packs_unpacks, py::arg("group") = 0)
       .def("sync_translates", &ExchangeEngine::sync_translates)
       .def("sync_packs", &ExchangeEngine::sync_packs)
       .def("sync_all", &ExchangeEngine::sync_all)
