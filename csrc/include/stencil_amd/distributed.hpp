@@ -48,6 +48,7 @@ public:
 private:
   std::string dir_;
   int rank_, world_;
+  int seq_ = 0; // per-call suffix: repeated collectives never collide
 };
 
 class DistributedDomain {

@@ -208,6 +208,7 @@ private:
 
   std::map<int, hipStream_t> commStreams_; // per device: translate launches
   std::map<int, hipStream_t> packStreams_; // per device: pack/unpack launches
+  std::map<int, hipEvent_t> fenceEvents_;  // per device: pack->unpack fences
   std::vector<hipStream_t> computeStreams_;  // per domain, stream 0
   std::vector<hipStream_t> computeStreams2_; // per domain, stream 1
 

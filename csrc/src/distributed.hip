@@ -27,9 +27,12 @@ FileBootstrap::FileBootstrap(std::string dir, int rank, int world)
 
 std::vector<std::string> FileBootstrap::allgather(const std::string &phase,
                                                  const std::string &payload) {
-  // write <dir>/<phase>.<rank> atomically (tmp + rename), then poll-read
-  // every rank's file
-  const std::string base = dir_ + "/" + phase + ".";
+  // write <dir>/<phase>.<seq>.<rank> atomically (tmp + rename), then
+  // poll-read every rank's file. The per-call sequence number keeps
+  // repeated collectives (or two DistributedDomains constructed in the
+  // same order on every rank) from matching stale files; the bootstrap
+  // dir itself must be fresh per job (run_native_mp.sh uses mktemp -d).
+  const std::string base = dir_ + "/" + phase + "." + std::to_string(seq_++) + ".";
   const std::string tmp = base + std::to_string(rank_) + ".tmp";
   const std::string fin = base + std::to_string(rank_);
   {

@@ -189,6 +189,7 @@ ExchangeEngine::~ExchangeEngine() {
   }
   for (auto &kv : commStreams_) (void)hipStreamDestroy(kv.second);
   for (auto &kv : packStreams_) (void)hipStreamDestroy(kv.second);
+  for (auto &kv : fenceEvents_) (void)hipEventDestroy(kv.second);
   for (auto s : computeStreams_)
     if (s) (void)hipStreamDestroy(s);
   for (auto s : computeStreams2_)
@@ -534,7 +535,9 @@ void ExchangeEngine::launch_unpacks(int group) {
 }
 
 void ExchangeEngine::fence_packs_unpacks(int group) {
-  static std::map<int, hipEvent_t> evs; // per source device, reused
+  // per-ENGINE events (a process-global map could be re-recorded by a
+  // second engine between our record and wait)
+  auto &evs = fenceEvents_;
   for (auto &b : packBatches_[group]) {
     if (b.jobs.empty()) continue;
     auto it = evs.find(b.dev);
