@@ -181,6 +181,16 @@ PYBIND11_MODULE(_C, m) {
              return out;
            },
            py::arg("pos"), py::arg("ext"), py::arg("qi"), py::arg("from_next") = false)
+      .def("region_to_host_into",
+           [](const LocalDomain &d, py::buffer out, const Vec3 &pos, const Vec3 &ext, int64_t qi,
+              bool fromNext) {
+             const py::buffer_info info = out.request(true);
+             if (info.size * info.itemsize != ext.flatten() * d.elem_size(qi))
+               throw std::runtime_error("region_to_host_into: size mismatch");
+             d.region_to_host(info.ptr, pos, ext, qi, fromNext);
+           },
+           py::arg("out"), py::arg("pos"), py::arg("ext"), py::arg("qi"),
+           py::arg("from_next") = false)
       .def("curr_pitch", [](const LocalDomain &d, int64_t qi) { return d.curr(qi).pitch; })
       .def("pad_bytes", &LocalDomain::pad_bytes)
       .def("curr_ysize", [](const LocalDomain &d, int64_t qi) { return d.curr(qi).ysize; })

@@ -59,6 +59,7 @@ int main(int argc, char **argv) {
     dd.swap();
   }
   const std::chrono::duration<double> dt = std::chrono::steady_clock::now() - t0;
+  if (getenv("STENCIL_PARAVIEW")) dd.write_paraview("jacobi_pv_");
 
   if (dd.rank() == 0)
     printf("jacobi3d_native: %lld^3 world=%d gpus=%d, %d iters, %.3f ms/iter, %.1f Gcell/s, "

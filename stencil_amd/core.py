@@ -370,17 +370,26 @@ class DistributedDomain:
         return out
 
     # ---- data access helpers (global-coordinate region of a quantity) ----
-    def read_global(self, li: int, lo: Vec, hi: Vec, handle: DataHandle, from_next=False) -> np.ndarray:
-        """read a global-coordinate region of local domain li as numpy (z,y,x)"""
+    def read_global(self, li: int, lo: Vec, hi: Vec, handle: DataHandle, from_next=False,
+                    out: Optional[np.ndarray] = None) -> np.ndarray:
+        """read a global-coordinate region of local domain li as numpy
+        (z,y,x). Pass `out` (a contiguous array of the right shape/dtype)
+        to reuse a buffer -- repeated checkpoint reads then skip the
+        fresh-page faults of a new allocation (native backend only)."""
         idx = self.placement.get_idx(self.comm.rank, li)
         o = self.placement.subdomain_origin(idx)
         r = self.radius
         flo = (o[0] - r.x(-1), o[1] - r.y(-1), o[2] - r.z(-1))
         pos = tuple(lo[i] - flo[i] for i in range(3))
         ext = tuple(hi[i] - lo[i] for i in range(3))
+        shape = (ext[2], ext[1], ext[0])
+        if out is not None and self.backend_kind == "native":
+            assert out.shape == shape and out.dtype == handle.dtype and out.flags["C_CONTIGUOUS"]
+            self.backend.read_region(li, pos, ext, handle.index, from_next, out=out)
+            return out
         raw = self.backend.read_region(li, pos, ext, handle.index, from_next)
         arr = np.frombuffer(raw, dtype=handle.dtype)
-        return arr.reshape(ext[2], ext[1], ext[0])
+        return arr.reshape(*shape)
 
     def write_global(self, li: int, lo: Vec, arr: np.ndarray, handle: DataHandle, to_next=False):
         idx = self.placement.get_idx(self.comm.rank, li)

@@ -575,7 +575,10 @@ class NativeBackend:
         self.engine.sync_compute()
 
     # ---- app/test helpers (positions in allocation coords) ----
-    def read_region(self, li: int, pos: Vec, ext: Vec, qi: int, from_next=False) -> bytes:
+    def read_region(self, li: int, pos: Vec, ext: Vec, qi: int, from_next=False, out=None):
+        if out is not None:  # reuse caller's buffer: no fresh-page faults
+            self.domains[li].region_to_host_into(out, _vec3(pos), _vec3(ext), qi, from_next)
+            return out
         return self.domains[li].region_to_host(_vec3(pos), _vec3(ext), qi, from_next)
 
     def write_region(self, li: int, data: bytes, pos: Vec, ext: Vec, qi: int, to_next=False):

@@ -332,3 +332,18 @@ def test_staged_local_translates(r, monkeypatch):
     fill_interiors(dd, h, scale=3.0)
     dd.exchange()
     check_full_regions(dd, h, scale=3.0)
+
+
+def test_read_global_out_buffer_reuse():
+    """read_global(out=...) fills the caller's array in place (checkpoint
+    loops skip fresh-page allocation); contents match a fresh read."""
+    dd = make_dd((16, 12, 10), 1, 1)
+    h = dd.add_data(np.float32, "q")
+    dd.realize()
+    fill_interiors(dd, h, scale=5.0)
+    lo, hi = dd.local_rect(0)
+    fresh = dd.read_global(0, lo, hi, h)
+    out = np.zeros_like(fresh)
+    got = dd.read_global(0, lo, hi, h, out=out)
+    assert got is out
+    np.testing.assert_array_equal(out, fresh)
