@@ -346,6 +346,108 @@ __device__ __forceinline__ void mhd_momentum_body(const MhdParams &p) {
 }
 
 __global__ void __launch_bounds__(256) mhd_momentum_kernel(MhdParams p) { mhd_momentum_body(p); }
+
+// z-pair momentum (STENCIL_MHD_MOM2=1): each thread computes cells z and
+// z+1. Counter-guided experiment: the kernels are 83% SQ_WAIT stalled
+// (profiles/astaroth_256_sq_counters_r2.csv), so (a) two independent
+// dependency chains per wave double the loads in flight, and (b) the two
+// cells SHARE their z-star -- the 7-deep center columns overlap in 6 of
+// 8 values per field, cutting the L2-missing z-displaced loads ~43%.
+// In-plane derivatives still come from L1/L2 via the Stencil helper.
+struct ZPair {
+  double v[8]; // f(q, 0, 0, z-3 .. z+4)
+  __device__ __forceinline__ double dz0(double ids) const {
+    return (D1[0] * (v[4] - v[2]) + D1[1] * (v[5] - v[1]) + D1[2] * (v[6] - v[0])) * ids;
+  }
+  __device__ __forceinline__ double dz1(double ids) const {
+    return (D1[0] * (v[5] - v[3]) + D1[1] * (v[6] - v[2]) + D1[2] * (v[7] - v[1])) * ids;
+  }
+  __device__ __forceinline__ double dzz0(double ids2) const {
+    return (D2[0] * v[3] + D2[1] * (v[4] + v[2]) + D2[2] * (v[5] + v[1]) + D2[3] * (v[6] + v[0])) *
+           ids2;
+  }
+  __device__ __forceinline__ double dzz1(double ids2) const {
+    return (D2[0] * v[4] + D2[1] * (v[5] + v[3]) + D2[2] * (v[6] + v[2]) + D2[3] * (v[7] + v[1])) *
+           ids2;
+  }
+  __device__ __forceinline__ double c0() const { return v[3]; }
+  __device__ __forceinline__ double c1() const { return v[4]; }
+};
+
+__global__ void __launch_bounds__(256) mhd_momentum_pair_kernel(MhdParams p) {
+  const int32_t lx = blockIdx.x * blockDim.x + threadIdx.x;
+  const int32_t ly = blockIdx.y * blockDim.y + threadIdx.y;
+  const int32_t lz0 = 2 * (blockIdx.z * blockDim.z + threadIdx.z);
+  if (lx >= p.extX || ly >= p.extY || lz0 >= p.extZ) return;
+  const bool two = lz0 + 1 < p.extZ;
+  const MhdCommon c = mhd_setup(p, lx, ly, lz0);
+  const double ix = 1.0 / p.dsx, iy = 1.0 / p.dsy, iz = 1.0 / p.dsz;
+  const int64_t plane = p.plane;
+
+  ZPair zp[10];
+#pragma unroll
+  for (int q = 0; q < 10; ++q)
+#pragma unroll
+    for (int k = 0; k < 8; ++k)
+      zp[q].v[k] = *(const double *)(c.base[q] + (int64_t)(k - 3) * plane);
+
+  Stencil st0, st1; // in-plane helpers for the two z layers
+  st0.pitch = st1.pitch = p.pitch;
+  st0.plane = st1.plane = p.plane;
+#pragma unroll
+  for (int q = 0; q < 10; ++q) {
+    st0.base[q] = c.base[q];
+    st1.base[q] = c.base[q] + plane;
+  }
+
+#pragma unroll
+  for (int cell = 0; cell < 2; ++cell) {
+    if (cell == 1 && !two) break;
+    const Stencil &st = cell ? st1 : st0;
+    const Vec3d uu = cell ? Vec3d{zp[UUX].c1(), zp[UUY].c1(), zp[UUZ].c1()}
+                          : Vec3d{zp[UUX].c0(), zp[UUY].c0(), zp[UUZ].c0()};
+    const double rho_inv = exp(-(cell ? zp[LNRHO].c1() : zp[LNRHO].c0()));
+    auto DZ = [&](int q) { return cell ? zp[q].dz1(iz) : zp[q].dz0(iz); };
+    auto DZZ = [&](int q) { return cell ? zp[q].dzz1(iz * iz) : zp[q].dzz0(iz * iz); };
+    auto LAP = [&](int q) { return st.dxx(q, ix * ix) + st.dyy(q, iy * iy) + DZZ(q); };
+    const Vec3d j = {st.dx(DIVA, ix) - LAP(AAX), st.dy(DIVA, iy) - LAP(AAY),
+                     DZ(DIVA) - LAP(AAZ)};
+    const Vec3d B = {st.dy(AAZ, iy) - DZ(AAY), DZ(AAX) - st.dx(AAZ, ix),
+                     st.dx(AAY, ix) - st.dy(AAX, iy)};
+    const Vec3d jxB = cross(j, B);
+    const int64_t outOff = (int64_t)cell * plane;
+    {
+      const double ugradu = uu.x * st.dx(UUX, ix) + uu.y * st.dy(UUX, iy) + uu.z * DZ(UUX);
+      const double press = st.dx(LNRHO, ix) + p.cp_inv * st.dx(SS, ix);
+      const double visc = p.nu * (LAP(UUX) + st.dx(DIVU, ix) / 3.0);
+      char *out = c.out[UUX] + outOff;
+      const double cur = cell ? zp[UUX].c1() : zp[UUX].c0();
+      const double prev = *(const double *)out;
+      *(double *)out = cur + p.beta * (p.alpha_over_beta_prev * (cur - prev) +
+                                       p.dt * (-ugradu - p.cs2 * press + rho_inv * jxB.x + visc));
+    }
+    {
+      const double ugradu = uu.x * st.dx(UUY, ix) + uu.y * st.dy(UUY, iy) + uu.z * DZ(UUY);
+      const double press = st.dy(LNRHO, iy) + p.cp_inv * st.dy(SS, iy);
+      const double visc = p.nu * (LAP(UUY) + st.dy(DIVU, iy) / 3.0);
+      char *out = c.out[UUY] + outOff;
+      const double cur = cell ? zp[UUY].c1() : zp[UUY].c0();
+      const double prev = *(const double *)out;
+      *(double *)out = cur + p.beta * (p.alpha_over_beta_prev * (cur - prev) +
+                                       p.dt * (-ugradu - p.cs2 * press + rho_inv * jxB.y + visc));
+    }
+    {
+      const double ugradu = uu.x * st.dx(UUZ, ix) + uu.y * st.dy(UUZ, iy) + uu.z * DZ(UUZ);
+      const double press = DZ(LNRHO) + p.cp_inv * DZ(SS);
+      const double visc = p.nu * (LAP(UUZ) + DZ(DIVU) / 3.0);
+      char *out = c.out[UUZ] + outOff;
+      const double cur = cell ? zp[UUZ].c1() : zp[UUZ].c0();
+      const double prev = *(const double *)out;
+      *(double *)out = cur + p.beta * (p.alpha_over_beta_prev * (cur - prev) +
+                                       p.dt * (-ugradu - p.cs2 * press + rho_inv * jxB.z + visc));
+    }
+  }
+}
 __global__ void __launch_bounds__(512) mhd_momentum_kernel_b512(MhdParams p) {
   mhd_momentum_body(p);
 }
@@ -581,17 +683,22 @@ void mhd_substep_launch_on(LocalDomain &d, const Rect3 &region, int step, double
   else
     hipLaunchKernelGGL(mhd_scalar_kernel, grid, block, 0, sScalar, p);
   STENCIL_HIP(hipGetLastError());
-  static int mom5 = -1, zmarch = -1;
+  static int mom5 = -1, zmarch = -1, mom2 = -1;
   if (mom5 < 0) {
     const char *e = getenv("STENCIL_MHD_MOM5");
     mom5 = (e && e[0] == '1') ? 1 : 0;
     e = getenv("STENCIL_MHD_ZMARCH");
     zmarch = (e && e[0] == '1') ? 1 : 0;
+    e = getenv("STENCIL_MHD_MOM2");
+    mom2 = (e && e[0] == '1') ? 1 : 0;
   }
   if (zmarch) {
     const dim3 zblock(64, 2, 1);
     const dim3 zgrid((uint32_t)((ext.x + 63) / 64), (uint32_t)((ext.y + 1) / 2), 1);
     hipLaunchKernelGGL(mhd_momentum_zmarch_kernel, zgrid, zblock, 0, sMomentum, p);
+  } else if (mom2) {
+    const dim3 pgrid(grid.x, grid.y, (uint32_t)((ext.z + 2 * block.z - 1) / (2 * block.z)));
+    hipLaunchKernelGGL(mhd_momentum_pair_kernel, pgrid, block, 0, sMomentum, p);
   } else if (b512)
     hipLaunchKernelGGL(mhd_momentum_kernel_b512, grid, block, 0, sMomentum, p);
   else if (mom5)
