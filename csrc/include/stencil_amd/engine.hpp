@@ -143,8 +143,8 @@ public:
   void sync_packs(); // also used after unpack
   void sync_all();
 
-  //// compute-stream helpers for apps (two streams per domain so
-  //// exterior-shell kernels can overlap the interior kernel)
+  //// compute-stream helpers for apps (streams 0/1 for interior/exterior
+  //// overlap; 2/3 exist for concurrent kernel-split experiments)
   hipStream_t compute_stream(int dom, int which = 0);
   uintptr_t compute_stream_handle(int dom) { return (uintptr_t)compute_stream(dom); }
   void sync_compute();
@@ -211,6 +211,8 @@ private:
   std::map<int, hipEvent_t> fenceEvents_;  // per device: pack->unpack fences
   std::vector<hipStream_t> computeStreams_;  // per domain, stream 0
   std::vector<hipStream_t> computeStreams2_; // per domain, stream 1
+  std::vector<hipStream_t> computeStreams3_; // streams 2/3: concurrent
+  std::vector<hipStream_t> computeStreams4_; // kernel-split experiments
 
   // pack/unpack launch-group encoding for exchange-group g:
   //   wire = 3g, colo staging parity 0/1 = 3g+1 / 3g+2

@@ -168,6 +168,8 @@ ExchangeEngine::ExchangeEngine(std::vector<std::shared_ptr<LocalDomain>> domains
     : domains_(std::move(domains)) {
   computeStreams_.resize(domains_.size(), nullptr);
   computeStreams2_.resize(domains_.size(), nullptr);
+  computeStreams3_.resize(domains_.size(), nullptr);
+  computeStreams4_.resize(domains_.size(), nullptr);
   // spin-wait host syncs: interrupt-based stream-sync wakeups cost
   // 20-100 us each on ROCm; an HPC bench prefers burning the core.
   // Tolerant: the flag may be rejected once a device context exists.
@@ -190,10 +192,9 @@ ExchangeEngine::~ExchangeEngine() {
   for (auto &kv : commStreams_) (void)hipStreamDestroy(kv.second);
   for (auto &kv : packStreams_) (void)hipStreamDestroy(kv.second);
   for (auto &kv : fenceEvents_) (void)hipEventDestroy(kv.second);
-  for (auto s : computeStreams_)
-    if (s) (void)hipStreamDestroy(s);
-  for (auto s : computeStreams2_)
-    if (s) (void)hipStreamDestroy(s);
+  for (auto *vec : {&computeStreams_, &computeStreams2_, &computeStreams3_, &computeStreams4_})
+    for (auto s : *vec)
+      if (s) (void)hipStreamDestroy(s);
   for (auto &b : buffers_) {
     if (!b.ptr) continue;
     if (b.external) {
@@ -497,7 +498,10 @@ hipStream_t ExchangeEngine::pack_stream_(int dev) {
 }
 
 hipStream_t ExchangeEngine::compute_stream(int dom, int which) {
-  auto &vec = which ? computeStreams2_ : computeStreams_;
+  auto &vec = which == 0   ? computeStreams_
+              : which == 1 ? computeStreams2_
+              : which == 2 ? computeStreams3_
+                           : computeStreams4_;
   if (!vec[dom]) {
     STENCIL_HIP(hipSetDevice(domains_[dom]->gpu()));
     hipStream_t s;
@@ -577,7 +581,7 @@ void ExchangeEngine::sync_all() {
 }
 
 void ExchangeEngine::sync_compute() {
-  for (auto *vec : {&computeStreams_, &computeStreams2_})
+  for (auto *vec : {&computeStreams_, &computeStreams2_, &computeStreams3_, &computeStreams4_})
     for (size_t i = 0; i < vec->size(); ++i)
       if ((*vec)[i]) {
         STENCIL_HIP(hipSetDevice(domains_[i]->gpu()));

@@ -347,6 +347,54 @@ __device__ __forceinline__ void mhd_momentum_body(const MhdParams &p) {
 
 __global__ void __launch_bounds__(256) mhd_momentum_kernel(MhdParams p) { mhd_momentum_body(p); }
 
+// Per-component momentum (STENCIL_MHD_MOMSPLIT=1): three kernels, one
+// velocity component each, on three CONCURRENT streams. Each kernel
+// needs only 2 of 3 j and B components (jxB_c = j_a B_b - j_b B_a), so
+// total work rises ~2x on the shared j/B part -- but chip-wide
+// concurrent wave count triples, the one axis the per-wave experiments
+// (z-march, z-pair, 512-blocks) could not move. Counter-guided: the
+// kernels are 83% SQ_WAIT stalled, so extra independent waves fill
+// latency the way extra per-wave ILP measurably did not.
+template <int C>
+__global__ void __launch_bounds__(256) mhd_momentum_comp_kernel(MhdParams p) {
+  int32_t bx = blockIdx.x, by = blockIdx.y, bz = blockIdx.z;
+  if (p.ychunk) ychunk_remap(p.ychunk, bx, by, bz);
+  const int32_t lx = bx * blockDim.x + threadIdx.x;
+  const int32_t ly = by * blockDim.y + threadIdx.y;
+  const int32_t lz = bz * blockDim.z + threadIdx.z;
+  if (lx >= p.extX || ly >= p.extY || lz >= p.extZ) return;
+  const MhdCommon c = mhd_setup(p, lx, ly, lz);
+  Stencil st;
+  st.pitch = c.pitch;
+  st.plane = c.plane;
+#pragma unroll
+  for (int q = 0; q < 10; ++q) st.base[q] = c.base[q];
+  const double ix = 1.0 / p.dsx, iy = 1.0 / p.dsy, iz = 1.0 / p.dsz;
+  const double ids1[3] = {ix, iy, iz};
+
+  // axis-generic first derivative / laplacian (axis folds at compile time)
+  auto d1 = [&](int q, int axis) {
+    return axis == 0 ? st.dx(q, ix) : axis == 1 ? st.dy(q, iy) : st.dz(q, iz);
+  };
+  auto lap = [&](int q) { return st.lap(q, ix, iy, iz); };
+
+  constexpr int A = (C + 1) % 3, Bc = (C + 2) % 3;
+  // j_k = d_k(divA) - lap(A_k);  B_k = d_{k+1}(A_{k+2}) - d_{k+2}(A_{k+1})
+  const double jA = d1(DIVA, A) - lap(AAX + A);
+  const double jB = d1(DIVA, Bc) - lap(AAX + Bc);
+  const double BA = d1(AAX + (A + 2) % 3, (A + 1) % 3) - d1(AAX + (A + 1) % 3, (A + 2) % 3);
+  const double BB = d1(AAX + (Bc + 2) % 3, (Bc + 1) % 3) - d1(AAX + (Bc + 1) % 3, (Bc + 2) % 3);
+  const double jxB_c = jA * BB - jB * BA;
+
+  const double rho_inv = exp(-st.c(LNRHO));
+  const double ugradu =
+      st.c(UUX) * st.dx(UUX + C, ix) + st.c(UUY) * st.dy(UUX + C, iy) + st.c(UUZ) * st.dz(UUX + C, iz);
+  const double press = d1(LNRHO, C) + p.cp_inv * d1(SS, C);
+  const double visc = p.nu * (lap(UUX + C) + d1(DIVU, C) / 3.0);
+  (void)ids1;
+  write_rk3(p, st, c.out[UUX + C], UUX + C, -ugradu - p.cs2 * press + rho_inv * jxB_c + visc);
+}
+
 // z-pair momentum (STENCIL_MHD_MOM2=1): each thread computes cells z and
 // z+1. Counter-guided experiment: the kernels are 83% SQ_WAIT stalled
 // (profiles/astaroth_256_sq_counters_r2.csv), so (a) two independent
@@ -721,6 +769,36 @@ void mhd_substep(ExchangeEngine &eng, int dom, const Rect3 &region, int step, do
                  const MhdCoeffs &cf, int streamId) {
   LocalDomain &d = eng.domain(dom);
   STENCIL_HIP(hipSetDevice(d.gpu()));
+  static int momsplit = -1;
+  if (momsplit < 0) {
+    const char *e = getenv("STENCIL_MHD_MOMSPLIT");
+    momsplit = (e && e[0] == '1') ? 1 : 0;
+  }
+  if (momsplit) {
+    // scalar + three per-component momentum kernels on four concurrent
+    // streams (see mhd_momentum_comp_kernel)
+    const Vec3 ext = region.extent();
+    if (ext.flatten() <= 0) return;
+    static const double ALPHA[3] = {0.0, -5.0 / 9.0, -153.0 / 128.0};
+    static const double BETA[3] = {1.0 / 3.0, 15.0 / 16.0, 8.0 / 15.0};
+    MhdParams p{};
+    mhd_fill_params(d, region, cf, p);
+    p.dt = dt;
+    p.alpha_over_beta_prev = (step == 0) ? 0.0 : ALPHA[step] / BETA[step - 1];
+    p.beta = BETA[step];
+    dim3 block = mhd_block();
+    dim3 grid = mhd_grid(ext, block);
+    p.ychunk = mhd_ychunk((int32_t)grid.y);
+    hipLaunchKernelGGL(mhd_scalar_kernel, grid, block, 0, eng.compute_stream(dom, streamId), p);
+    hipLaunchKernelGGL(mhd_momentum_comp_kernel<0>, grid, block, 0, eng.compute_stream(dom, 1),
+                       p);
+    hipLaunchKernelGGL(mhd_momentum_comp_kernel<1>, grid, block, 0, eng.compute_stream(dom, 2),
+                       p);
+    hipLaunchKernelGGL(mhd_momentum_comp_kernel<2>, grid, block, 0, eng.compute_stream(dom, 3),
+                       p);
+    STENCIL_HIP(hipGetLastError());
+    return;
+  }
   mhd_substep_launch_on(d, region, step, dt, cf, eng.compute_stream(dom, streamId),
                         eng.compute_stream(dom, 1 - streamId));
 }
