@@ -215,12 +215,28 @@ class NativeBackend:
         }
         if want == "nccl" and len(devs) > 1:
             want = "torch"
-        votes = comm.allgather_object((want, any(self._has_wire)))
-        if not any(w for _, w in votes):
+        my_dev = next(iter(devs)) if devs else (
+            self.domains[0].gpu() if self.domains else -1
+        )
+        votes = comm.allgather_object(
+            (want, any(self._has_wire), comm.hostname, my_dev)
+        )
+        if not any(w for _, w, _, _ in votes):
             self._wire_mode = "none"
             return
-        modes = {m for m, _ in votes}
+        modes = {m for m, _, _, _ in votes}
         mode = "cpu" if "cpu" in modes else ("torch" if "torch" in modes else "nccl")
+        # NCCL (native or torch) refuses two ranks on one GPU: demote to
+        # the host-staged wire when colocated ranks share a device (the
+        # time-shared simulation shape; real nodes give each rank its own
+        # GPU and keep the fast path)
+        if mode in ("nccl", "torch"):
+            seen = set()
+            for _, _, host, dev in votes:
+                if (host, dev) in seen:
+                    mode = "cpu"
+                    break
+                seen.add((host, dev))
         self._wire_mode = mode
         if mode == "nccl":
             self._wire_dev = next(iter(devs)) if devs else (
