@@ -136,7 +136,8 @@ def main():
                 "radius": 1,
                 "halo_multiplier": args.halo_multiplier,
                 "overlap": not args.no_overlap,
-                "step_graph": getattr(app, "_graph", None) is not None,
+                "step_graph": (getattr(app, "_graph", None) is not None
+                               or getattr(app, "_mr_graph", None) is not None),
                 "exchange_bytes_per_iter": xbytes,
                 "parallelism": f"domain-decomposition {dims[0]}x{dims[1]}x{dims[2]} "
                 + (
