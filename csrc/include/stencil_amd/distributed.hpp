@@ -22,6 +22,7 @@
 //     RCCL one.)
 #pragma once
 
+#include <map>
 #include <memory>
 #include <string>
 #include <vector>
@@ -101,6 +102,10 @@ public:
   // write_paraview)
   void write_paraview(const std::string &prefix);
 
+  // setup-phase seconds (reference STENCIL_SETUP_STATS,
+  // stencil.hpp:103-112): keys topo/placement/realize/plan/create
+  const std::map<std::string, double> &setup_times() const { return setupTimes_; }
+
 private:
   void gather_slots_(std::vector<Slot> &slots);
 
@@ -120,6 +125,7 @@ private:
   int wireDev_ = -1;
   std::vector<bool> hasWire_;
   int64_t bytesTranslate_ = 0, bytesWire_ = 0;
+  std::map<std::string, double> setupTimes_;
   bool realized_ = false;
 };
 

@@ -141,8 +141,13 @@ void DistributedDomain::gather_slots_(std::vector<Slot> &slots) {
 
 void DistributedDomain::realize() {
   if (realized_) throw std::runtime_error("realize() called twice");
+  auto now = []() { return std::chrono::steady_clock::now(); };
+  auto secs = [](auto a, auto b) { return std::chrono::duration<double>(b - a).count(); };
+  auto t0 = now();
   std::vector<Slot> slots;
   gather_slots_(slots);
+  setupTimes_["topo"] = secs(t0, now());
+  t0 = now();
 
   switch (strategy_) {
   case PlacementStrategy::Trivial:
@@ -157,6 +162,8 @@ void DistributedDomain::realize() {
     break;
   }
 
+  setupTimes_["placement"] = secs(t0, now());
+  t0 = now();
   const int nLocal = placement_->num_local(rank_);
   for (int li = 0; li < nLocal; ++li) {
     const Vec3 idx = placement_->get_idx(rank_, li);
@@ -183,8 +190,12 @@ void DistributedDomain::realize() {
     for (int64_t qi = 0; qi < (int64_t)data_.size(); ++qi) groups_.back().push_back(qi);
   }
   hasWire_.assign(groups_.size(), false);
+  setupTimes_["realize"] = secs(t0, now());
+  t0 = now();
 
   const ExchangePlan plan = plan_exchange(*placement_, radius_, rank_, &halo_extent_of);
+  setupTimes_["plan"] = secs(t0, now());
+  t0 = now();
   std::vector<int64_t> elemSizes;
   int64_t esTotal = 0;
   for (auto &q : data_) {
@@ -272,6 +283,34 @@ void DistributedDomain::realize() {
     }
   } else if (anyWire) {
     throw std::runtime_error("cross-rank messages planned at world=1");
+  }
+  setupTimes_["create"] = secs(t0, now());
+  // plan files (reference src/stencil.cu:482-637; Python _write_plan_files
+  // writes the same shape)
+  if (const char *prefix = getenv("STENCIL_OUTPUT_PREFIX")) {
+    int64_t esTotal2 = 0;
+    for (auto &q : data_) esTotal2 += q.first;
+    FILE *f = fopen((std::string(prefix) + "plan_" + std::to_string(rank_) + ".txt").c_str(), "w");
+    if (f) {
+      const Vec3 dim = placement_->dim();
+      fprintf(f, "rank %d world %d dim (%lld, %lld, %lld)\n", rank_, world_, (long long)dim.x,
+              (long long)dim.y, (long long)dim.z);
+      for (const TranslatePlanItem &t : plan.translates)
+        fprintf(f, "direct_kernel dir=(%lld, %lld, %lld) src_local=%d dst_local=%d bytes=%lld\n",
+                (long long)t.dir.x, (long long)t.dir.y, (long long)t.dir.z, t.srcLocal,
+                t.dstLocal, (long long)(t.ext.flatten() * esTotal2));
+      for (const WirePlanItem &it : plan.sends)
+        for (const PlanMessage &m : it.messages)
+          fprintf(f, "rccl_send dir=(%lld, %lld, %lld) dst_rank=%d src_gid=%lld dst_gid=%lld bytes=%lld\n",
+                  (long long)m.dir.x, (long long)m.dir.y, (long long)m.dir.z, it.peerRank,
+                  (long long)m.srcGid, (long long)m.dstGid, (long long)(m.volume() * esTotal2));
+      for (const WirePlanItem &it : plan.recvs)
+        for (const PlanMessage &m : it.messages)
+          fprintf(f, "rccl_recv dir=(%lld, %lld, %lld) src_rank=%d src_gid=%lld dst_gid=%lld bytes=%lld\n",
+                  (long long)m.dir.x, (long long)m.dir.y, (long long)m.dir.z, it.peerRank,
+                  (long long)m.srcGid, (long long)m.dstGid, (long long)(m.volume() * esTotal2));
+      fclose(f);
+    }
   }
   realized_ = true;
 }
