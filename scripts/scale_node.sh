@@ -58,4 +58,5 @@ build/jacobi3d_native 750 $STEPS "$MAXN" || true
 if [ "$MAXN" -ge 2 ]; then
   echo "=== pure C++ orchestrator: $MAXN processes over FileBootstrap + RcclWire ==="
   tools/run_native_mp.sh "$MAXN" build/jacobi3d_native 1024 $STEPS || true
+  tools/run_native_mp.sh "$MAXN" build/astaroth_native 512 10 || true
 fi
