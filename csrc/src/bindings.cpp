@@ -408,6 +408,34 @@ PYBIND11_MODULE(_C, m) {
     return py::make_tuple(total, out);
   });
 
+  // The C++ orchestrator itself (distributed.hpp), bound so the GPU test
+  // tier can drive it directly (examples/*.cpp are the no-Python proof;
+  // this binding pins its behavior under pytest)
+  py::class_<DistributedDomain>(m, "CppDistributedDomain")
+      .def(py::init<int64_t, int64_t, int64_t>())
+      .def("set_radius", (void(DistributedDomain::*)(const Radius &)) & DistributedDomain::set_radius)
+      .def("add_data", (int64_t(DistributedDomain::*)(int64_t, const std::string &)) &
+                           DistributedDomain::add_data,
+           py::arg("elem_size"), py::arg("name") = "")
+      .def("set_gpus", &DistributedDomain::set_gpus)
+      .def("set_placement_trivial",
+           [](DistributedDomain &d) { d.set_placement(PlacementStrategy::Trivial); })
+      .def("set_exchange_groups", &DistributedDomain::set_exchange_groups)
+      .def("realize", &DistributedDomain::realize)
+      .def("exchange", &DistributedDomain::exchange, py::arg("group") = 0)
+      .def("swap", &DistributedDomain::swap)
+      .def("num_local", &DistributedDomain::num_local)
+      .def("local_rect", &DistributedDomain::local_rect)
+      .def("get_interior", &DistributedDomain::get_interior)
+      .def("get_exterior", &DistributedDomain::get_exterior)
+      .def("bytes_translate", &DistributedDomain::bytes_translate)
+      .def("bytes_wire", &DistributedDomain::bytes_wire)
+      .def("write_paraview", &DistributedDomain::write_paraview)
+      .def("setup_times", &DistributedDomain::setup_times)
+      .def("domain", &DistributedDomain::domain, py::return_value_policy::reference_internal)
+      .def("rank", &DistributedDomain::rank)
+      .def("world", &DistributedDomain::world);
+
   // FileBootstrap (distributed.hpp): the C++ multi-process control plane,
   // bound for CPU-side tests of the allgather protocol
   py::class_<FileBootstrap>(m, "FileBootstrap")
