@@ -26,7 +26,8 @@ def ripple_block(lo, hi, size, scale=1.0):
 def fill_interiors(dd, handle, scale=1.0):
     for li in range(dd.num_local()):
         lo, hi = dd.local_rect(li)
-        dd.write_global(li, lo, ripple_block(lo, hi, dd.size, scale), handle)
+        block = ripple_block(lo, hi, dd.size, scale).astype(handle.dtype)
+        dd.write_global(li, lo, block, handle)
 
 
 def full_region_of(dd, li):
