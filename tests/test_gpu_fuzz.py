@@ -11,7 +11,7 @@ import pytest
 import stencil_amd as sa
 from stencil_amd import _C
 
-from util import check_full_regions, fill_interiors
+from util import check_valid_regions, fill_interiors
 
 pytestmark = pytest.mark.gpu
 
@@ -44,7 +44,9 @@ def test_native_fuzz_campaign(seed):
             for g in groups:
                 dd.exchange(group=g)
             for h in hs:
-                check_full_regions(dd, h, scale=1.0 + h.index)
+                # valid-regions check: randomly sparsified radii leave
+                # radius-0 sides legitimately unexchanged
+                check_valid_regions(dd, h, scale=1.0 + h.index)
             dd.swap()
             for h in hs:
                 fill_interiors(dd, h, scale=1.0 + h.index)
