@@ -265,7 +265,8 @@ class NativeBackend:
                     ops.append((t, peer, tag))
                     if mode == "cpu":
                         self._cpu_mirror[id(t)] = torch.empty(
-                            t.shape, dtype=torch.uint8, device="cpu"
+                            t.shape, dtype=torch.uint8, device="cpu",
+                            pin_memory=torch.cuda.is_available(),
                         )
 
     @staticmethod
