@@ -188,3 +188,69 @@ def test_fuzz_mp_exchange(seed, port):
         p.join(timeout=60)
     for rank, status in results:
         assert status == "ok", f"rank {rank}: {status}"
+
+
+def _two_node_worker(rank, world, port, q):
+    try:
+        os.environ["MASTER_ADDR"] = "127.0.0.1"
+        os.environ["MASTER_PORT"] = str(port)
+        import torch.distributed as dist
+
+        dist.init_process_group("gloo", rank=rank, world_size=world)
+        import sys
+
+        sys.path.insert(0, os.path.dirname(__file__))
+        import stencil_amd as sa
+        from stencil_amd.parallel import comm as comm_mod
+        from util import check_full_regions, fill_interiors
+
+        # simulate one rank per NODE: distinct hostnames drive the
+        # two-level NodePartition and the non-colocated wire path (no IPC
+        # possible across "nodes")
+        orig_init = comm_mod.Comm.__init__
+
+        def patched(self):
+            orig_init(self)
+            self.hostname = f"fakenode{rank}"
+
+        comm_mod.Comm.__init__ = patched
+
+        dd = sa.DistributedDomain(16, 12, 10, backend="torch")
+        dd.set_radius(2)
+        dd.set_gpus([0])
+        h = dd.add_data(np.float32, "q")
+        dd.realize()
+        assert dd.comm.node_of_rank() == [0, 1]
+        assert dd.comm.colocated_ranks() == [rank]
+        assert dd.placement.part.sys_dim().tuple() != (1, 1, 1)  # node-level split
+        assert dd.bytes_by_method["rccl"] > 0  # cross-node halos on the wire
+        assert dd.bytes_by_method["ipc_kernel"] == 0
+        fill_interiors(dd, h)
+        dd.exchange()
+        check_full_regions(dd, h)
+        dd.swap()
+        fill_interiors(dd, h, scale=2.0)
+        dd.exchange()
+        check_full_regions(dd, h, scale=2.0)
+        dist.destroy_process_group()
+        q.put((rank, "ok"))
+    except Exception:  # pragma: no cover
+        import traceback
+
+        q.put((rank, traceback.format_exc()))
+
+
+def test_two_node_simulation():
+    """multi-NODE path (distinct hostnames): two-level partition, no
+    colocated ranks, every cross-rank halo over the packed wire — the
+    shape multi-node RCCL runs take, minus the fabric"""
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    procs = [ctx.Process(target=_two_node_worker, args=(r, 2, 29591, q)) for r in range(2)]
+    for p in procs:
+        p.start()
+    results = [q.get(timeout=300) for _ in range(2)]
+    for p in procs:
+        p.join(timeout=60)
+    for rank, status in results:
+        assert status == "ok", f"rank {rank}: {status}"
