@@ -254,3 +254,62 @@ def test_two_node_simulation():
         p.join(timeout=60)
     for rank, status in results:
         assert status == "ok", f"rank {rank}: {status}"
+
+
+def _hetero_worker(rank, world, port, q):
+    try:
+        os.environ["MASTER_ADDR"] = "127.0.0.1"
+        os.environ["MASTER_PORT"] = str(port)
+        import torch.distributed as dist
+
+        dist.init_process_group("gloo", rank=rank, world_size=world)
+        import sys
+
+        sys.path.insert(0, os.path.dirname(__file__))
+        import stencil_amd as sa
+        from stencil_amd.parallel import comm as comm_mod
+        from util import check_full_regions, fill_interiors
+
+        # node A holds ranks 0,1; node B holds rank 2 -> heterogeneous
+        # per-node GPU counts -> flat single-level partition fallback
+        node = "nodeA" if rank < 2 else "nodeB"
+        orig_init = comm_mod.Comm.__init__
+
+        def patched(self):
+            orig_init(self)
+            self.hostname = node
+
+        comm_mod.Comm.__init__ = patched
+
+        dd = sa.DistributedDomain(18, 12, 10, backend="torch")
+        dd.set_radius(1)
+        dd.set_gpus([0])
+        h = dd.add_data(np.float32, "q")
+        dd.realize()
+        assert dd.placement.uniform_nodes is False
+        d = dd.placement.dim()
+        assert d[0] * d[1] * d[2] == 3
+        fill_interiors(dd, h)
+        dd.exchange()
+        check_full_regions(dd, h)
+        dist.destroy_process_group()
+        q.put((rank, "ok"))
+    except Exception:  # pragma: no cover
+        import traceback
+
+        q.put((rank, traceback.format_exc()))
+
+
+def test_heterogeneous_nodes_exchange():
+    """heterogeneous per-node GPU counts (2+1) end-to-end: the flat
+    placement fallback plans and exchanges correctly across 3 ranks"""
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    procs = [ctx.Process(target=_hetero_worker, args=(r, 3, 29595, q)) for r in range(3)]
+    for p in procs:
+        p.start()
+    results = [q.get(timeout=300) for _ in range(3)]
+    for p in procs:
+        p.join(timeout=60)
+    for rank, status in results:
+        assert status == "ok", f"rank {rank}: {status}"
