@@ -148,3 +148,24 @@ def test_cpp_python_plan_parity_fuzz():
                     assert tag == tags[(it.peer_rank, it.src_gid, it.dst_gid)], ctx
                     assert [(tuple(d), m1, m2, tuple(e)) for d, m1, m2, e in msgs] == [
                         (m.dir, m.src_gid, m.dst_gid, m.ext) for m in it.messages], ctx
+
+
+def test_cpp_python_parity_heterogeneous_nodes():
+    """uneven per-node GPU counts: both planners fall back to the flat
+    single-level partition and agree exactly"""
+    size = (30, 24, 18)
+    radius = _C.Radius.constant(1)
+    slots = make_slots(2, 2)          # 2 nodes x 2 GPUs
+    slots.append(Slot(4, 0, 2, 1))    # node 1 gets a 3rd GPU
+    slot_tuples = [(s.rank, s.local_id, s.cuda, s.node) for s in slots]
+    pp = py_placement("trivial", size, radius, slots)
+    assert pp.uniform_nodes is False
+    for rank in range(5):
+        cpp = _C.cpp_plan(_C.Vec3(*size), radius, rank, slot_tuples, "trivial")
+        assert tuple(cpp["dim"]) == pp.dim()
+        plan = plan_exchange(pp, radius, rank)
+        assert len(cpp["translates"]) == len(plan.translates)
+        assert len(cpp["sends"]) == len(plan.sends)
+        assert len(cpp["recvs"]) == len(plan.recvs)
+        for (peer, sg, dg, li, msgs, tag), it in zip(cpp["sends"], plan.sends):
+            assert (peer, sg, dg, li) == (it.peer_rank, it.src_gid, it.dst_gid, it.local_id)
