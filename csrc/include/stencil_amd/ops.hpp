@@ -91,6 +91,10 @@ void mhd_substep(ExchangeEngine &eng, int dom, const Rect3 &region, int step, do
 // (deterministic smooth initial conditions, reproducible in NumPy)
 void init_harmonic_f64(ExchangeEngine &eng, int dom, int64_t qi, const Rect3 &region, double base,
                        double amp, double kx, double ky, double kz, double phase, bool nextBuf);
+// radial gaussian bump around (cx,cy,cz) (reference astaroth.cu
+// radial_explosion_init_kernel): base + amp*exp(-r^2/(2 sigma^2))
+void init_radial_f64(ExchangeEngine &eng, int dom, int64_t qi, const Rect3 &region, double base,
+                     double amp, double cx, double cy, double cz, double sigma, bool nextBuf);
 
 // min/max/RMS of a quantity over a region (reference: reductions.cuh)
 struct FieldStats {

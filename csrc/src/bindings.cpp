@@ -326,6 +326,7 @@ PYBIND11_MODULE(_C, m) {
   m.def("mhd_div_pass", &mhd_div_pass, py::arg("eng"), py::arg("dom"), py::arg("region"),
         py::arg("cf"), py::arg("stream_id") = 0);
   m.def("init_harmonic_f64", &init_harmonic_f64);
+  m.def("init_radial_f64", &init_radial_f64);
   py::class_<FieldStats>(m, "FieldStats")
       .def_readonly("min", &FieldStats::min)
       .def_readonly("max", &FieldStats::max)

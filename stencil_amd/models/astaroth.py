@@ -179,9 +179,24 @@ class Astaroth:
             _C.mhd_mr_graph_sync(self._mr_graph)
 
 
-    def init_fields(self):
-        """harmonic initial conditions on every interior (device-side)"""
+    def init_fields(self, kind: str = "harmonic"):
+        """device-side initial conditions on every interior.
+        kind='harmonic' (default): per-field sinusoidal modes;
+        kind='explosion': radial gaussian velocity bump at the domain
+        center (reference astaroth.cu radial_explosion_init_kernel),
+        other fields zero."""
         eng = self.dd.backend.engine
+        if kind == "explosion":
+            cx, cy, cz = (s / 2.0 for s in self.size)
+            sigma = min(self.size) / 8.0
+            for li in range(self.dd.num_local()):
+                lo, hi = self.dd.local_rect(li)
+                rect = _C.Rect3(_C.Vec3(*lo), _C.Vec3(*hi))
+                for qi, name in enumerate(FIELDS):
+                    amp = 0.1 if name.startswith("uu") else 0.0
+                    _C.init_radial_f64(eng, li, qi, rect, 0.0, amp, cx, cy, cz, sigma, False)
+            self.dd.backend.sync_compute()
+            return
         for li in range(self.dd.num_local()):
             lo, hi = self.dd.local_rect(li)
             for qi, (base, amp, m, phase) in enumerate(init_modes(self.size)):
