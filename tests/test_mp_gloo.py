@@ -313,3 +313,52 @@ def test_heterogeneous_nodes_exchange():
         p.join(timeout=60)
     for rank, status in results:
         assert status == "ok", f"rank {rank}: {status}"
+
+
+def _methods_worker(rank, world, port, q):
+    try:
+        os.environ["MASTER_ADDR"] = "127.0.0.1"
+        os.environ["MASTER_PORT"] = str(port)
+        import torch.distributed as dist
+
+        dist.init_process_group("gloo", rank=rank, world_size=world)
+        import sys
+
+        sys.path.insert(0, os.path.dirname(__file__))
+        import stencil_amd as sa
+        from util import check_full_regions, fill_interiors
+
+        dd = sa.DistributedDomain(14, 10, 8, backend="torch")
+        dd.set_radius(1)
+        dd.set_gpus([0])
+        # disable the colocated IPC transport via the Method API: halos
+        # between colocated ranks must fall back to the packed wire
+        dd.set_methods(sa.Method.DIRECT_KERNEL | sa.Method.RCCL)
+        h = dd.add_data(np.float32, "q")
+        dd.realize()
+        assert dd.bytes_by_method["ipc_kernel"] == 0
+        assert dd.bytes_by_method["rccl"] > 0
+        fill_interiors(dd, h)
+        dd.exchange()
+        check_full_regions(dd, h)
+        dist.destroy_process_group()
+        q.put((rank, "ok"))
+    except Exception:  # pragma: no cover
+        import traceback
+
+        q.put((rank, traceback.format_exc()))
+
+
+def test_method_flags_disable_ipc():
+    """Method API fallback chain (reference method.hpp first-match
+    dispatch): with IPC_KERNEL cleared, colocated halos ride the wire"""
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    procs = [ctx.Process(target=_methods_worker, args=(r, 2, 29599, q)) for r in range(2)]
+    for p in procs:
+        p.start()
+    results = [q.get(timeout=300) for _ in range(2)]
+    for p in procs:
+        p.join(timeout=60)
+    for rank, status in results:
+        assert status == "ok", f"rank {rank}: {status}"
