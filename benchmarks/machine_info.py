@@ -10,6 +10,7 @@ sys.path.insert(0, os.path.join(os.path.dirname(__file__), ".."))
 
 from stencil_amd import _C
 from stencil_amd.parallel.comm import Comm
+from stencil_amd.parallel.machine import Machine
 
 
 def main():
@@ -28,6 +29,7 @@ def main():
             dict(dev=d, name=gi.name, pci=gi.pci, mem_gb=gi.total_mem / 2**30, cus=gi.cu_count)
         )
     infos = comm.allgather_object({"host": comm.hostname, "gpus": gpus})
+    machine = Machine.build(comm)
     if comm.rank == 0:
         for r, info in enumerate(infos):
             print(f"rank {r} host {info['host']}")
@@ -35,6 +37,9 @@ def main():
                 print(
                     f"  gpu {g['dev']}: {g['name']} pci={g['pci']} mem={g['mem_gb']:.0f}GB cus={g['cus']}"
                 )
+        print(f"machine: {machine.num_nodes()} node(s), {len(machine.gpus)} physical GPU(s)")
+        for g in machine.gpus:
+            print(f"  global gpu {g.index}: node {g.node} {g.name} pci={g.pci} ranks={g.ranks}")
         if n > 1:
             print("link distance matrix (gpu_distance):")
             for a in range(n):

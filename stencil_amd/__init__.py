@@ -49,9 +49,11 @@ except ImportError as e:  # pragma: no cover
 
 from ._C import Radius, Rect3, Vec3, prime_factors
 from .core import DataHandle, DistributedDomain, Method
+from .parallel.machine import Machine
 from .parallel.placement import PlacementStrategy
 
 __all__ = [
+    "Machine",
     "DistributedDomain",
     "DataHandle",
     "Method",
