@@ -30,6 +30,11 @@ void fill_f32(ExchangeEngine &eng, int dom, int64_t qi, const Rect3 &region, flo
 // (alternating parities, host mirrors flipped); sync() blocks on them.
 int64_t jacobi_graph_create(ExchangeEngine &eng, int dom, int64_t qi, const Rect3 &region,
                             const Rect3 &computeRegion);
+// overlap variant: translate forked concurrent with the interior kernel
+// inside the graph (STENCIL_AMD_GRAPH_OVERLAP=1; see csrc/src/jacobi.hip)
+int64_t jacobi_graph_create_overlap(ExchangeEngine &eng, int dom, int64_t qi,
+                                    const Rect3 &interior, const Rect3 &computeRegion,
+                                    const std::vector<Rect3> &exteriors);
 void jacobi_graph_launch(int64_t handle, int64_t nSteps);
 void jacobi_graph_sync(int64_t handle);
 

@@ -290,6 +290,9 @@ PYBIND11_MODULE(_C, m) {
   m.def("fill_f32", &fill_f32);
   m.def("jacobi_graph_create", &jacobi_graph_create, py::arg("eng"), py::arg("dom"),
         py::arg("qi"), py::arg("region"), py::arg("compute_region"));
+  m.def("jacobi_graph_create_overlap", &jacobi_graph_create_overlap, py::arg("eng"),
+        py::arg("dom"), py::arg("qi"), py::arg("interior"), py::arg("compute_region"),
+        py::arg("exteriors"));
   m.def("jacobi_graph_launch", &jacobi_graph_launch, py::arg("handle"), py::arg("n_steps") = 1);
   m.def("jacobi_graph_sync", &jacobi_graph_sync);
   m.def("jacobi_mr_graph_create", &jacobi_mr_graph_create, py::arg("eng"), py::arg("dom"),

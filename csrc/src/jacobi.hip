@@ -507,6 +507,53 @@ int64_t jacobi_graph_create(ExchangeEngine &eng, int dom, int64_t qi, const Rect
   return (int64_t)g_stepGraphs.size() - 1;
 }
 
+// Overlap variant of the whole-step graph (STENCIL_AMD_GRAPH_OVERLAP=1):
+// fork the translate copy_batch onto a second stream concurrent with the
+// interior kernel, join, then exterior shells + swap. The round-1 eager
+// A/B called overlap "a wash", but that measured per-step host
+// orchestration against the SERIAL graph; in-graph the fork costs
+// nothing, so the ~60 us copy_batch can hide under the ~700 us interior.
+// Interior runs fullRectVec mode 1 (free extension); its x-overshoot
+// lands in cells the +-x exterior slabs rewrite LATER ON THE SAME
+// STREAM, which is exactly the eager-mode ordering argument.
+int64_t jacobi_graph_create_overlap(ExchangeEngine &eng, int dom, int64_t qi,
+                                    const Rect3 &interior, const Rect3 &computeRegion,
+                                    const std::vector<Rect3> &exteriors) {
+  LocalDomain &d = eng.domain(dom);
+  STENCIL_HIP(hipSetDevice(d.gpu()));
+  auto sg = std::make_unique<StepGraph>();
+  sg->eng = &eng;
+  sg->dom = dom;
+  STENCIL_HIP(hipStreamCreateWithFlags(&sg->stream, hipStreamNonBlocking));
+  hipStream_t stream2;
+  STENCIL_HIP(hipStreamCreateWithFlags(&stream2, hipStreamNonBlocking));
+  hipEvent_t evF, evJ;
+  STENCIL_HIP(hipEventCreateWithFlags(&evF, hipEventDisableTiming));
+  STENCIL_HIP(hipEventCreateWithFlags(&evJ, hipEventDisableTiming));
+  for (int par = 0; par < 2; ++par) {
+    STENCIL_HIP(hipStreamBeginCapture(sg->stream, hipStreamCaptureModeThreadLocal));
+    STENCIL_HIP(hipEventRecord(evF, sg->stream));
+    STENCIL_HIP(hipStreamWaitEvent(stream2, evF, 0));
+    eng.launch_translates_plain_on((uintptr_t)stream2, 0); // halos || interior
+    launch_jacobi_on(d, qi, interior, computeRegion, sg->stream, /*fullRectVec=*/1);
+    STENCIL_HIP(hipEventRecord(evJ, stream2));
+    STENCIL_HIP(hipStreamWaitEvent(sg->stream, evJ, 0));
+    for (const Rect3 &box : exteriors)
+      launch_jacobi_on(d, qi, box, computeRegion, sg->stream, /*fullRectVec=*/0);
+    d.enqueue_table_swap(sg->stream);
+    hipGraph_t g = nullptr;
+    STENCIL_HIP(hipStreamEndCapture(sg->stream, &g));
+    STENCIL_HIP(hipGraphInstantiate(&sg->exec[par], g, nullptr, nullptr, 0));
+    STENCIL_HIP(hipGraphDestroy(g));
+    d.swap();
+  }
+  STENCIL_HIP(hipStreamDestroy(stream2));
+  STENCIL_HIP(hipEventDestroy(evF));
+  STENCIL_HIP(hipEventDestroy(evJ));
+  g_stepGraphs.push_back(std::move(sg));
+  return (int64_t)g_stepGraphs.size() - 1;
+}
+
 void jacobi_graph_launch(int64_t handle, int64_t nSteps) {
   StepGraph &sg = *g_stepGraphs.at(handle);
   LocalDomain &d = sg.eng->domain(sg.dom);

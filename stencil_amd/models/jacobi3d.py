@@ -78,10 +78,23 @@ class Jacobi3D:
             and graphs_on
             and not staged_local
         ):
+            from .. import _C
+
             lo, hi = self.dd.local_rect(0)
-            self._graph = self.dd.backend.jacobi_graph_create(
-                0, self.h.index, lo, hi, self.compute_lo, self.compute_hi
-            )
+            if os.environ.get("STENCIL_AMD_GRAPH_OVERLAP", "0") == "1":
+                # experiment: translate forked || interior inside the graph
+                ilo, ihi = self.interiors[0]
+                self._graph = _C.jacobi_graph_create_overlap(
+                    self.dd.backend.engine, 0, self.h.index,
+                    _C.Rect3(_C.Vec3(*ilo), _C.Vec3(*ihi)),
+                    _C.Rect3(_C.Vec3(*self.compute_lo), _C.Vec3(*self.compute_hi)),
+                    [_C.Rect3(_C.Vec3(*blo), _C.Vec3(*bhi))
+                     for blo, bhi in self.exteriors[0]],
+                )
+            else:
+                self._graph = self.dd.backend.jacobi_graph_create(
+                    0, self.h.index, lo, hi, self.compute_lo, self.compute_hi
+                )
         elif (
             self.m == 1
             and self.dd.backend_kind == "native"
