@@ -38,6 +38,17 @@ echo "=== bench_exchange 1024^3 single GPU reference (BASELINE config 3 shape) =
 python benchmarks/bench_exchange.py --size 512 --iters 20 || true
 
 for N in $NS; do
+  echo "=== jacobi3d STRONG 2048^3 r=2, $N GPUs (BASELINE config 4) ==="
+  if [ "$N" = 1 ]; then
+    python benchmarks/jacobi3d.py --size 2048 --radius 2 --strong --iters 20
+  else
+    python -m torch.distributed.run --nnodes=1 --nproc-per-node $N \
+      --master-addr 127.0.0.1 --master-port 29644 \
+      benchmarks/jacobi3d.py --size 2048 --radius 2 --strong --iters 20
+  fi
+done
+
+for N in $NS; do
   echo "=== astaroth weak 256^3/GPU, $N GPUs ==="
   if [ "$N" = 1 ]; then
     python benchmarks/astaroth.py --per-gpu 256 --iters 20 --warmup 3
