@@ -479,6 +479,12 @@ PYBIND11_MODULE(_C, m) {
       .def_readonly("cu_count", &GpuInfo::cuCount);
   m.def("gpu_info", &gpu_info);
 
+  m.def("device_mem_info", [](int dev) {
+    STENCIL_HIP(hipSetDevice(dev));
+    size_t freeB = 0, totalB = 0;
+    STENCIL_HIP(hipMemGetInfo(&freeB, &totalB));
+    return py::make_tuple((int64_t)freeB, (int64_t)totalB);
+  });
   m.def("device_count", []() {
     int n = 0;
     if (hipGetDeviceCount(&n) != hipSuccess) return 0;
